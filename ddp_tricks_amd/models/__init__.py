@@ -1,0 +1,14 @@
+from .toy_net import Toy_Net  # noqa: F401
+
+_REGISTRY = {"toy_net": Toy_Net}
+
+
+def build_model(name: str, **kwargs):
+    try:
+        return _REGISTRY[name](**kwargs)
+    except KeyError:
+        raise ValueError(f"unknown model {name!r}; available: {sorted(_REGISTRY)}")
+
+
+def register_model(name: str, ctor) -> None:
+    _REGISTRY[name] = ctor

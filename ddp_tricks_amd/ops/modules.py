@@ -1,0 +1,75 @@
+"""nn.Module wrappers over the framework ops.
+
+These subclass the torch modules (identical parameter/buffer registration →
+identical state_dict keys and init) but route forward through
+ops/functional, which dispatches to the HIP kernels on GPU.  BN+ReLU pairs
+are fused by Toy_Net using ``BatchNorm2d(..., fuse_relu=True)`` style flags
+at the model level (see models/toy_net.py); the standalone modules keep
+reference semantics.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from . import functional as F_ops
+
+
+class Conv2d(nn.Conv2d):
+    def forward(self, x):
+        return F_ops.conv2d(x, self.weight, self.bias, self.stride, self.padding)
+
+
+class Linear(nn.Linear):
+    def forward(self, x):
+        return F_ops.linear(x, self.weight, self.bias)
+
+
+class _BatchNormBase:
+    def _bn_forward(self, x):
+        self._check_input_dim(x)
+        if self.training:
+            if self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+        momentum = self.momentum if self.momentum is not None else 0.1
+        fuse_relu = getattr(self, "fuse_relu", False)
+        return F_ops.batch_norm(
+            x, self.running_mean, self.running_var, self.weight, self.bias,
+            self.training or not self.track_running_stats, momentum, self.eps,
+            fuse_relu=fuse_relu)
+
+
+class BatchNorm2d(nn.BatchNorm2d, _BatchNormBase):
+    def __init__(self, *args, fuse_relu: bool = False, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.fuse_relu = fuse_relu
+
+    def forward(self, x):
+        return self._bn_forward(x)
+
+
+class BatchNorm1d(nn.BatchNorm1d, _BatchNormBase):
+    def __init__(self, *args, fuse_relu: bool = False, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.fuse_relu = fuse_relu
+
+    def forward(self, x):
+        return self._bn_forward(x)
+
+
+class ReLU(nn.ReLU):
+    def forward(self, x):
+        return F_ops.relu(x, inplace=self.inplace)
+
+
+class MaxPool2d(nn.MaxPool2d):
+    def forward(self, x):
+        return F_ops.max_pool2d(x, self.kernel_size, self.stride)
+
+
+class Flatten(nn.Flatten):
+    pass
+
+
+class Identity(nn.Identity):
+    pass

@@ -1,0 +1,175 @@
+"""Data pipeline: MNIST IDX reader, synthetic fallback, DistributedSampler,
+and a CUDA prefetching loader wrapper.
+
+Replaces the reference's torchvision.MNIST + torch DistributedSampler +
+pin_memory DataLoader stack (reference utils/train.py:24-30) without
+torchvision and without network access:
+
+  * ``MNIST`` reads the raw IDX (optionally gzipped) files if present under
+    ``{root}/MNIST/raw`` and otherwise builds a deterministic synthetic
+    look-alike (60k/10k × 1×28×28, 10 classes, class-dependent patterns) so
+    every pipeline above it behaves identically.
+  * ``DistributedSampler`` matches torch's semantics: per-epoch seeded
+    shuffle, padding to a rank-divisible length, ``set_epoch``.
+  * ``CudaPrefetcher`` stages batches pinned and copies H2D on a dedicated
+    copy stream one batch ahead (SURVEY N15 disposition).
+"""
+from __future__ import annotations
+
+import gzip
+import math
+import os
+import struct
+from typing import Iterator, Optional
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+
+def _read_idx(path: str) -> np.ndarray:
+    opener = gzip.open if path.endswith(".gz") else open
+    with opener(path, "rb") as f:
+        magic = struct.unpack(">I", f.read(4))[0]
+        ndim = magic & 0xFF
+        dims = [struct.unpack(">I", f.read(4))[0] for _ in range(ndim)]
+        data = np.frombuffer(f.read(), dtype=np.uint8)
+    return data.reshape(dims)
+
+
+def _find_idx(root: str, base: str) -> Optional[str]:
+    for sub in ("MNIST/raw", "MNIST", "raw", "."):
+        for ext in ("", ".gz"):
+            p = os.path.join(root, sub, base + ext)
+            if os.path.exists(p):
+                return p
+    return None
+
+
+class MNIST(Dataset):
+    """MNIST (or a deterministic synthetic stand-in when files are absent).
+
+    Returns (float32 [1,28,28] in [0,1], int64 label) like
+    torchvision.MNIST with ToTensor (reference utils/train.py:25).
+    """
+
+    def __init__(self, root: str = "./datasets/", train: bool = True,
+                 download: bool = False, synthetic: Optional[bool] = None,
+                 num_samples: Optional[int] = None):
+        base_img = "train-images-idx3-ubyte" if train else "t10k-images-idx3-ubyte"
+        base_lbl = "train-labels-idx1-ubyte" if train else "t10k-labels-idx1-ubyte"
+        img_path = _find_idx(root, base_img)
+        lbl_path = _find_idx(root, base_lbl)
+        n_default = 60000 if train else 10000
+        self.synthetic = synthetic if synthetic is not None else not (img_path and lbl_path)
+        if not self.synthetic:
+            images = _read_idx(img_path).astype(np.float32) / 255.0
+            labels = _read_idx(lbl_path).astype(np.int64)
+            self.images = torch.from_numpy(images).unsqueeze(1).contiguous()
+            self.labels = torch.from_numpy(labels)
+        else:
+            env_cap = os.environ.get("DDPX_SYNTH_SAMPLES")
+            if num_samples is None and env_cap:
+                num_samples = min(int(env_cap), n_default)
+            n = num_samples or n_default
+            g = torch.Generator().manual_seed(1234 if train else 4321)
+            labels = torch.randint(0, 10, (n,), generator=g)
+            # Class-dependent separable patterns + noise: a fixed random
+            # template per class so the net has something learnable.
+            templates = torch.rand(10, 1, 28, 28, generator=g)
+            noise = torch.rand(n, 1, 28, 28, generator=g)
+            self.images = (0.6 * templates[labels] + 0.4 * noise).clamp_(0, 1)
+            self.labels = labels
+
+    def __len__(self) -> int:
+        return self.labels.shape[0]
+
+    def __getitem__(self, idx: int):
+        return self.images[idx], int(self.labels[idx])
+
+
+class DistributedSampler(torch.utils.data.Sampler):
+    """torch.utils.data.distributed.DistributedSampler semantics
+    (seeded shuffle, pad to divisible, set_epoch) re-implemented."""
+
+    def __init__(self, dataset, num_replicas: Optional[int] = None,
+                 rank: Optional[int] = None, shuffle: bool = True,
+                 seed: int = 0, drop_last: bool = False):
+        import torch.distributed as dist
+        if num_replicas is None:
+            num_replicas = dist.get_world_size() if dist.is_initialized() else 1
+        if rank is None:
+            rank = dist.get_rank() if dist.is_initialized() else 0
+        self.dataset = dataset
+        self.num_replicas = num_replicas
+        self.rank = rank
+        self.epoch = 0
+        self.shuffle = shuffle
+        self.seed = seed
+        self.drop_last = drop_last
+        n = len(dataset)
+        if drop_last and n % num_replicas:
+            self.num_samples = n // num_replicas
+        else:
+            self.num_samples = math.ceil(n / num_replicas)
+        self.total_size = self.num_samples * num_replicas
+
+    def __iter__(self) -> Iterator[int]:
+        n = len(self.dataset)
+        if self.shuffle:
+            g = torch.Generator()
+            g.manual_seed(self.seed + self.epoch)
+            indices = torch.randperm(n, generator=g).tolist()
+        else:
+            indices = list(range(n))
+        if not self.drop_last:
+            padding = self.total_size - len(indices)
+            if padding > 0:
+                indices += (indices * math.ceil(padding / max(len(indices), 1)))[:padding]
+        else:
+            indices = indices[: self.total_size]
+        indices = indices[self.rank: self.total_size: self.num_replicas]
+        assert len(indices) == self.num_samples
+        return iter(indices)
+
+    def __len__(self) -> int:
+        return self.num_samples
+
+    def set_epoch(self, epoch: int) -> None:
+        self.epoch = epoch
+
+
+class CudaPrefetcher:
+    """Wrap a DataLoader: pinned H2D copies one batch ahead on a copy stream."""
+
+    def __init__(self, loader, device: torch.device):
+        self.loader = loader
+        self.device = device
+        self.stream = torch.cuda.Stream(device=device)
+
+    def __len__(self):
+        return len(self.loader)
+
+    def __iter__(self):
+        it = iter(self.loader)
+        next_batch = None
+
+        def _preload():
+            nonlocal next_batch
+            try:
+                image, target = next(it)
+            except StopIteration:
+                next_batch = None
+                return
+            with torch.cuda.stream(self.stream):
+                next_batch = (
+                    image.to(self.device, non_blocking=True),
+                    target.to(self.device, non_blocking=True),
+                )
+
+        _preload()
+        while next_batch is not None:
+            torch.cuda.current_stream(self.device).wait_stream(self.stream)
+            batch = next_batch
+            _preload()
+            yield batch

@@ -1,0 +1,150 @@
+"""train(args) — full training orchestration.
+
+Re-implements the reference trainer's exact wiring (reference
+utils/train.py:23-118) on the MI355X-native stack: our MNIST pipeline +
+DistributedSampler, our Toy_Net (HIP kernels on GPU), FusedSGD + Lookahead,
+our amp runtime, our RCCL-bucketed DDP, our tfevents writer, identical
+epoch loop / scheduler gating / early stopping / best-checkpoint rule.
+
+Behavioral quirks preserved: epoch 0 trains at LR 0;
+warmup steps while ``epoch <= warmup_epochs``; cosine branch freezes after
+warmup; plateau patience 6 / factor 0.1 on valid loss; EarlyStopping
+patience 30; unsharded validation on every rank; train metrics rank-local;
+rank-0 best-valid-acc gated ``torch.save(module.state_dict())`` to
+``{model_path}/{exp_name}.pt`` (37-key fp32 layout).
+"""
+from __future__ import annotations
+
+import math
+import os
+import time
+
+import torch
+import torch.distributed as torch_dist
+from torch.utils.data import DataLoader
+
+from .. import amp
+from ..models.toy_net import Toy_Net
+from ..ops.optim import FusedSGD
+from ..parallel.ddp import DistributedDataParallel as DDP
+from .callbacks import EarlyStopping, same_seeds
+from .data import MNIST, DistributedSampler, CudaPrefetcher
+from .engine import iterate_loader
+from .lookahead import Lookahead
+from .schedulers import ReduceLROnPlateau, WarmupLambdaLR
+from .tboard import SummaryWriter
+
+
+def _resolve_device(local_rank):
+    if torch.cuda.is_available():
+        rank = local_rank if local_rank is not None and local_rank >= 0 else 0
+        torch.cuda.set_device(rank)
+        return torch.device("cuda", rank)
+    return torch.device("cpu")
+
+
+def train(args):
+    local_rank = getattr(args, "local_rank", 0) or 0
+    device = _resolve_device(local_rank)
+
+    # Data pipeline (reference utils/train.py:24-30: sharded train loader,
+    # UNSHARDED valid loader evaluated in full on every rank)
+    train_set = MNIST(root=args.data_path, train=True, download=True)
+    train_sampler = DistributedSampler(train_set)
+    same_seeds(args.seed_num)
+    train_loader = DataLoader(train_set, batch_size=args.batch_size,
+                              shuffle=False, pin_memory=device.type == "cuda",
+                              sampler=train_sampler)
+    valid_set = MNIST(root=args.data_path, train=False, download=True)
+    valid_loader = DataLoader(valid_set, batch_size=args.batch_size,
+                              shuffle=False, pin_memory=device.type == "cuda")
+    if device.type == "cuda":
+        train_loader = CudaPrefetcher(train_loader, device)
+        valid_loader = CudaPrefetcher(valid_loader, device)
+
+    print(f"Now Training: {args.exp_name}")
+
+    # Model (seeded identically on all ranks before the DDP broadcast —
+    # reference utils/train.py:34-36, README rationale)
+    same_seeds(args.seed_num)
+    model = Toy_Net()
+    model = model.to(device)
+
+    os.makedirs(os.path.join(args.model_path, "logs"), exist_ok=True)
+    latest_model_path = os.path.join(args.model_path, args.exp_name)
+    optimizer = FusedSGD(model.parameters(), lr=args.learning_rate,
+                         momentum=0.9, nesterov=True)
+    lookahead = Lookahead(optimizer=optimizer, k=10, alpha=0.5)
+    from ..ops.functional import cross_entropy_loss as loss_function
+    best_valid_acc = 0
+
+    # LR warmup lambda (reference utils/train.py:48-51, incl. the dead
+    # cosine decay quirk — the caller only steps warmup while
+    # epoch <= warmup_epochs)
+    if args.warmup_type == "linear":
+        def warm_up(epoch):
+            return epoch / args.warmup_epochs if epoch <= args.warmup_epochs else 1
+    elif args.warmup_type == "cosine":
+        def warm_up(epoch):
+            if epoch <= args.warmup_epochs:
+                return epoch / args.warmup_epochs
+            return 0.5 * (math.cos((epoch - args.warmup_epochs)
+                                   / (args.epochs - args.warmup_epochs) * math.pi) + 1)
+    else:
+        raise ValueError(f"unknown warmup_type {args.warmup_type!r}")
+    scheduler_wu = WarmupLambdaLR(optimizer=optimizer, lr_lambda=warm_up)
+    scheduler_re = ReduceLROnPlateau(optimizer=optimizer, mode="min",
+                                     factor=0.1, patience=6, verbose=True)
+    early_stopping = EarlyStopping(patience=30, verbose=True)
+
+    # amp before DDP wrap (required order — SURVEY Appendix A.12)
+    model, apex_optimizer = amp.initialize(model, optimizers=lookahead,
+                                           opt_level="O1")
+    parallel_model = DDP(model)
+
+    if local_rank == 0:
+        tb = SummaryWriter(os.path.join(args.model_path, "logs", args.exp_name))
+
+    for epoch in range(args.epochs):
+        epoch_start_time = time.time()
+        train_sampler.set_epoch(epoch)
+
+        parallel_model.train()
+        train_loss, train_acc, curr_lr = iterate_loader(
+            loader=train_loader, model=parallel_model,
+            loss_function=loss_function, local_rank=local_rank,
+            apex_optimizer=apex_optimizer, training=True)
+
+        parallel_model.eval()
+        with torch.no_grad():
+            valid_loss, valid_acc = iterate_loader(
+                loader=valid_loader, model=parallel_model,
+                loss_function=loss_function, local_rank=local_rank,
+                apex_optimizer=None, training=False)
+
+        if local_rank == 0:
+            tb.add_scalar("LR", curr_lr, epoch)
+            tb.add_scalars("Loss", {"train": train_loss, "valid": valid_loss}, epoch)
+            tb.add_scalars("Acc", {"train": train_acc, "valid": valid_acc}, epoch)
+
+        print(f"epoch: {epoch:03d}/{args.epochs}, "
+              f"time: {time.time() - epoch_start_time:.2f}s, "
+              f"learning_rate: {curr_lr}, "
+              f"train_loss: {train_loss:.4f}, train_acc: {train_acc:.4f}, "
+              f"valid_loss: {valid_loss:.4f}, valid_acc: {valid_acc:.4f}")
+
+        # scheduler gating exactly as the reference (utils/train.py:104-106)
+        if epoch <= args.warmup_epochs:
+            scheduler_wu.step()
+        scheduler_re.step(valid_loss)
+        early_stopping(valid_loss)
+        if early_stopping.early_stop:
+            break
+
+        if local_rank == 0 and valid_acc > best_valid_acc:
+            best_valid_acc = valid_acc
+            torch.save(parallel_model.module.state_dict(),
+                       f"{latest_model_path}.pt")
+
+    if local_rank == 0:
+        tb.close()
