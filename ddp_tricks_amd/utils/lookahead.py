@@ -18,7 +18,7 @@ API- and semantics-compatible with the reference's vendored wrapper
 """
 from __future__ import annotations
 
-from collections import defaultdict
+from collections import OrderedDict, defaultdict
 
 import torch
 from torch.optim import Optimizer
@@ -29,6 +29,16 @@ class Lookahead(Optimizer):
         self.optimizer = optimizer
         self.k = k
         self.alpha = alpha
+        # We intentionally do NOT run Optimizer.__init__ (it would deep-copy
+        # param groups; the whole point is SHARING the inner optimizer's
+        # groups) — but torch>=2 Optimizer methods expect these attributes:
+        self.defaults = {"k": k, "alpha": alpha}
+        self._optimizer_step_pre_hooks = OrderedDict()
+        self._optimizer_step_post_hooks = OrderedDict()
+        self._optimizer_state_dict_pre_hooks = OrderedDict()
+        self._optimizer_state_dict_post_hooks = OrderedDict()
+        self._optimizer_load_state_dict_pre_hooks = OrderedDict()
+        self._optimizer_load_state_dict_post_hooks = OrderedDict()
         # Shared param_groups object keeps LR schedulers pointed at the inner
         # optimizer coherent with stepping through the wrapper
         # (SURVEY Appendix A.4).

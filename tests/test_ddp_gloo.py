@@ -1,0 +1,136 @@
+"""Multi-process DDP correctness on CPU/gloo, world_size=2:
+gradient all-reduce equivalence with single-process large-batch training,
+param broadcast at construction, bucket-view gradients."""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from ddp_tricks_amd import same_seeds
+
+WORLD = 2
+
+
+def _init(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+
+class TinyNet(torch.nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.fc1 = torch.nn.Linear(8, 16)
+        self.bn = torch.nn.BatchNorm1d(16)
+        self.fc2 = torch.nn.Linear(16, 4)
+
+    def forward(self, x):
+        return self.fc2(torch.relu(self.bn(self.fc1(x))))
+
+
+def _worker_grad_equiv(rank, port, results):
+    _init(rank, WORLD, port)
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    same_seeds(11)
+    model = TinyNet()
+    ddp = DDP(model, bucket_cap_mb=0.0001)  # force multiple buckets
+    torch.manual_seed(123)  # same data on both ranks' generator
+    x = torch.randn(WORLD * 4, 8)
+    t = torch.randn(WORLD * 4, 4)
+    shard_x = x[rank * 4:(rank + 1) * 4]
+    shard_t = t[rank * 4:(rank + 1) * 4]
+    model.train()
+    out = ddp(shard_x)
+    loss = ((out - shard_t) ** 2).mean()
+    loss.backward()
+    ddp.finalize_backward(average=True)
+    if rank == 0:
+        grads = {k: p.grad.detach().clone() for k, p in model.named_parameters()}
+        results.put(grads)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_ddp_grads_match_fullbatch():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29611
+    procs = [ctx.Process(target=_worker_grad_equiv, args=(r, port, results))
+             for r in range(WORLD)]
+    [p.start() for p in procs]
+    grads = results.get(timeout=110)
+    [p.join(timeout=60) for p in procs]
+    assert all(p.exitcode == 0 for p in procs)
+
+    # single-process oracle: mean of shard losses == ... careful: DDP averages
+    # gradients of per-shard MEAN losses, which equals the gradient of the
+    # mean of the two shard losses.
+    same_seeds(11)
+    model = TinyNet()
+    torch.manual_seed(123)
+    x = torch.randn(WORLD * 4, 8)
+    t = torch.randn(WORLD * 4, 4)
+    model.train()
+    loss0 = ((model(x[:4]) - t[:4]) ** 2).mean()
+    # fresh BN stats per shard on each rank — replicate rank-local BN by
+    # re-running with reset running stats
+    same_seeds(11)
+    model2 = TinyNet()
+    loss1 = ((model2(x[4:]) - t[4:]) ** 2).mean()
+    loss = (loss0 + loss1) / 2
+    loss.backward()
+    # accumulate grads from both replicas
+    for (k, p), (k2, p2) in zip(model.named_parameters(), model2.named_parameters()):
+        g = (p.grad if p.grad is not None else 0) + (p2.grad if p2.grad is not None else 0)
+        assert torch.allclose(grads[k], g, atol=1e-5), k
+
+
+def _worker_broadcast(rank, port, results):
+    _init(rank, WORLD, port)
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    torch.manual_seed(100 + rank)  # DIFFERENT init per rank
+    model = TinyNet()
+    DDP(model)
+    if rank == 1:
+        results.put({k: v.detach().clone() for k, v in model.state_dict().items()})
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_param_broadcast_from_rank0():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29612
+    procs = [ctx.Process(target=_worker_broadcast, args=(r, port, results))
+             for r in range(WORLD)]
+    [p.start() for p in procs]
+    sd1 = results.get(timeout=110)
+    [p.join(timeout=60) for p in procs]
+    assert all(p.exitcode == 0 for p in procs)
+    torch.manual_seed(100)  # rank 0 init
+    ref = TinyNet().state_dict()
+    for k in ref:
+        assert torch.allclose(sd1[k], ref[k]), k
+
+
+def test_singleproc_ddp_grad_views():
+    """world_size=1 path: grads are views into bucket flats."""
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    model = TinyNet()
+    ddp = DDP(model)
+    x = torch.randn(4, 8)
+    loss = ddp(x).sum()
+    loss.backward()
+    ddp.finalize_backward()
+    flats = ddp.bucket_flats()
+    total = sum(f.numel() for f in flats)
+    assert total == sum(p.numel() for p in model.parameters())
+    for p in model.parameters():
+        assert p.grad is not None
+        assert any(p.grad.data_ptr() >= f.data_ptr()
+                   and p.grad.data_ptr() < f.data_ptr() + f.numel() * 4
+                   for f in flats)

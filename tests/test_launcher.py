@@ -1,0 +1,58 @@
+"""Launcher contract: env exports, --local_rank injection, watchdog teardown
+(reference launch contract README.md:18-25)."""
+import os
+import subprocess
+import sys
+import textwrap
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_launch(tmp_path, script_body, nproc=2, extra=()):
+    script = tmp_path / "child.py"
+    script.write_text(textwrap.dedent(script_body))
+    cmd = [sys.executable, "-m", "ddp_tricks_amd.launch",
+           f"--nproc_per_node={nproc}", "--master_port=29713",
+           *extra, str(script)]
+    return subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                          timeout=120)
+
+
+@pytest.mark.timeout(180)
+def test_env_and_local_rank(tmp_path):
+    r = _run_launch(tmp_path, """
+        import os, sys
+        assert os.environ["MASTER_ADDR"] == "127.0.0.1"
+        assert os.environ["MASTER_PORT"] == "29713"
+        assert os.environ["WORLD_SIZE"] == "2"
+        lr = [a for a in sys.argv if a.startswith("--local_rank=")]
+        assert len(lr) == 1
+        assert lr[0].split("=")[1] == os.environ["LOCAL_RANK"]
+        print("rank-ok", os.environ["RANK"])
+    """)
+    assert r.returncode == 0, r.stderr
+    assert "rank-ok 0" in r.stdout and "rank-ok 1" in r.stdout
+
+
+@pytest.mark.timeout(180)
+def test_use_env_omits_flag(tmp_path):
+    r = _run_launch(tmp_path, """
+        import sys
+        assert not any(a.startswith("--local_rank") for a in sys.argv)
+    """, extra=("--use_env",))
+    assert r.returncode == 0, r.stderr
+
+
+@pytest.mark.timeout(180)
+def test_watchdog_kills_group_on_failure(tmp_path):
+    # rank 1 exits non-zero immediately; rank 0 would sleep forever —
+    # the watchdog must tear it down and exit non-zero well within timeout.
+    r = _run_launch(tmp_path, """
+        import os, time, sys
+        if os.environ["LOCAL_RANK"] == "1":
+            sys.exit(3)
+        time.sleep(600)
+    """)
+    assert r.returncode != 0

@@ -1,0 +1,52 @@
+"""End-to-end train(args) on CPU/gloo world 1 (BASELINE.json config 1):
+plumbing of warmup + plateau + early stopping + checkpoint + TB layout."""
+import os
+import types
+
+import pytest
+import torch
+import torch.distributed as dist
+
+from ddp_tricks_amd import amp
+
+
+@pytest.fixture(autouse=True)
+def _reset_amp():
+    amp._state.__init__()
+    yield
+    amp._state.__init__()
+
+
+def _args(tmp_path, **over):
+    a = types.SimpleNamespace(
+        exp_name="TEST_run", learning_rate=0.1, batch_size=64, epochs=3,
+        warmup_epochs=2, warmup_type="linear", seed_num=42,
+        data_path="/nonexistent_data", model_path=str(tmp_path),
+        local_rank=0)
+    for k, v in over.items():
+        setattr(a, k, v)
+    return a
+
+
+@pytest.mark.timeout(300)
+def test_train_e2e_cpu(tmp_path, monkeypatch, capsys):
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "256")
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29655")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    from ddp_tricks_amd.utils.train import train
+    train(_args(tmp_path))
+    out = capsys.readouterr().out
+    # epoch 0 trains at LR 0
+    assert "learning_rate: 0.0," in out
+    # checkpoint written with the reference naming scheme
+    assert os.path.exists(os.path.join(tmp_path, "TEST_run.pt"))
+    sd = torch.load(os.path.join(tmp_path, "TEST_run.pt"))
+    assert len(sd) == 37
+    # TB layout
+    logdir = os.path.join(tmp_path, "logs", "TEST_run")
+    subdirs = {d for d in os.listdir(logdir)
+               if os.path.isdir(os.path.join(logdir, d))}
+    assert {"Loss_train", "Loss_valid", "Acc_train", "Acc_valid"} <= subdirs
+    dist.destroy_process_group()
