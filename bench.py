@@ -1,0 +1,197 @@
+"""bench.py — flagship benchmark: Toy_Net MNIST DDP training step throughput.
+
+Measures the reference's headline metric (BASELINE.json): whole-node
+images/sec of Toy_Net MNIST DDP training (per-GPU batch 1024, amp-O1-style
+bf16 compute, FusedSGD(momentum=0.9, nesterov) + Lookahead(k=10, α=0.5)),
+on synthetic MNIST-shaped data with random-init weights (no network access
+for datasets).  Reference baseline (other hardware): ≈12.0k
+train img/s on an RTX-2080-class node.
+
+Contract: ``python bench.py --gpus N --steps K --warmup W``; for N>1 the
+driver launches it under torch.distributed.run with one rank per GPU over
+RCCL.  W untimed warmup steps, then exactly K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; elapsed time is the MAX
+over ranks; rank 0 prints one JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+os.environ.setdefault("DDPX_NO_TQDM", "1")
+
+from ddp_tricks_amd import amp, same_seeds  # noqa: E402
+from ddp_tricks_amd.models.toy_net import Toy_Net  # noqa: E402
+from ddp_tricks_amd.ops.functional import cross_entropy_loss  # noqa: E402
+from ddp_tricks_amd.ops.optim import FusedSGD  # noqa: E402
+from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP  # noqa: E402
+from ddp_tricks_amd.utils.lookahead import Lookahead  # noqa: E402
+
+BASELINE_IMG_PER_SEC = 12000.0  # BASELINE.md implied train throughput
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch-size", type=int, default=1024,
+                   help="per-GPU batch size (reference default)")
+    p.add_argument("--no-h2d", action="store_true",
+                   help="skip the per-step H2D copy (ablation only)")
+    return p.parse_args()
+
+
+class SyntheticMNIST:
+    """Pinned-host synthetic MNIST batches with one-ahead H2D prefetch on a
+    copy stream (mirrors the training pipeline's prefetch)."""
+
+    def __init__(self, batch: int, device, n_buffers: int = 4, h2d: bool = True):
+        g = torch.Generator().manual_seed(1234)
+        self.h2d = h2d and device.type == "cuda"
+        self.device = device
+        self.host_images = []
+        self.host_targets = []
+        for _ in range(n_buffers):
+            img = torch.rand(batch, 1, 28, 28, generator=g)
+            tgt = torch.randint(0, 10, (batch,), generator=g)
+            if self.h2d:
+                img = img.pin_memory()
+                tgt = tgt.pin_memory()
+            self.host_images.append(img)
+            self.host_targets.append(tgt)
+        if self.h2d:
+            self.copy_stream = torch.cuda.Stream(device=device)
+            self.dev_images = [torch.empty_like(i, device=device)
+                               for i in self.host_images[:2]]
+            self.dev_targets = [torch.empty_like(t, device=device)
+                                for t in self.host_targets[:2]]
+            self.events = [torch.cuda.Event(), torch.cuda.Event()]
+            self._i = 0
+            self._prefetch(0)
+        else:
+            self.dev_images = [i.to(device) for i in self.host_images]
+            self.dev_targets = [t.to(device) for t in self.host_targets]
+            self._i = 0
+
+    def _prefetch(self, slot: int):
+        src = self._i % len(self.host_images)
+        with torch.cuda.stream(self.copy_stream):
+            self.dev_images[slot].copy_(self.host_images[src], non_blocking=True)
+            self.dev_targets[slot].copy_(self.host_targets[src], non_blocking=True)
+            self.events[slot].record(self.copy_stream)
+        self._i += 1
+
+    def next(self):
+        if not self.h2d:
+            j = self._i % len(self.dev_images)
+            self._i += 1
+            return self.dev_images[j], self.dev_targets[j]
+        slot = (self._i - 1) % 2
+        torch.cuda.current_stream(self.device).wait_event(self.events[slot])
+        img, tgt = self.dev_images[slot], self.dev_targets[slot]
+        self._prefetch(slot ^ 1)
+        return img, tgt
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29501")
+    os.environ.setdefault("RANK", str(rank))
+    os.environ.setdefault("WORLD_SIZE", str(world))
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        backend = "nccl"
+    else:
+        device = torch.device("cpu")
+        backend = "gloo"
+    dist.init_process_group(backend=backend, init_method="env://")
+
+    same_seeds(42)
+    model = Toy_Net().to(device)
+    optimizer = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, nesterov=True)
+    lookahead = Lookahead(optimizer, k=10, alpha=0.5)
+    model, apex_optimizer = amp.initialize(model, lookahead, opt_level="O1")
+    ddp = DDP(model)
+    ddp.train()
+
+    data = SyntheticMNIST(args.batch_size, device, h2d=not args.no_h2d)
+
+    def step():
+        apex_optimizer.zero_grad()
+        image, target = data.next()
+        outputs = ddp(image)
+        batch_loss = cross_entropy_loss(outputs, target)
+        batch_loss = batch_loss / outputs.shape[0]
+        with amp.scale_loss(batch_loss, apex_optimizer) as scaled_loss:
+            scaled_loss.backward()
+        apex_optimizer.step()
+        return batch_loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if use_cuda:
+        torch.cuda.synchronize()
+    dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if backend == "nccl" else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        total_images = args.steps * args.batch_size * world
+        value = total_images / elapsed
+        print(json.dumps({
+            "metric": "images/sec",
+            "value": value,
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / BASELINE_IMG_PER_SEC,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "Toy_Net",
+                "dataset": "MNIST(synthetic)",
+                "global_batch": args.batch_size * world,
+                "per_gpu_batch": args.batch_size,
+                "seq_len": None,
+                "parallelism": f"dp{world}",
+                "optimizer": "FusedSGD(momentum=0.9,nesterov)+Lookahead(k=10,a=0.5)",
+            },
+        }))
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

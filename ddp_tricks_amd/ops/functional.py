@@ -1,18 +1,16 @@
 """Autograd-integrated ops: HIP/CDNA4 kernels on GPU, torch reference on CPU.
 
-Each op is a torch.autograd.Function whose forward/backward call the
-_hip_ops extension (bf16 compute, fp32 accumulate — the framework's amp-O1
-policy: conv/linear whitelisted to bf16, BN stats/CE in fp32; reference
-behavior via apex amp.initialize at utils/train.py:58).  On CPU (the
-no-GPU test tier) the same Python entry points run the equivalent torch
-ops in fp32 so they double as the numerics oracle.
+GPU data layout: activations are bf16 **channels_last** (NHWC — channels
+fastest, the MFMA implicit-GEMM-friendly layout on CDNA4); weights are
+cached per-parameter-version as bf16 channels_last (KRSC) plus the
+transformed WT2[C, R·S·K] operand for backward-data.  Statistics, biases,
+losses and weight gradients are fp32 (the framework's amp-O1 policy —
+reference behavior via apex at utils/train.py:58).
 
-Weight bf16 casts are cached per parameter version (one cast per optimizer
-step, like apex's per-iteration cast cache).
+On CPU (the no-GPU test tier) the same entry points run equivalent torch
+ops in fp32, doubling as the numerics oracle the GPU tests compare against.
 """
 from __future__ import annotations
-
-from typing import Optional, Tuple
 
 import torch
 import torch.nn.functional as F
@@ -20,30 +18,62 @@ import torch.nn.functional as F
 from . import require_ext_for
 from .. import amp as amp_mod
 
+CL = torch.channels_last
+
 
 # --------------------------------------------------------------- casting ---
 
-_BF16_CACHE = {}  # param -> (version, bf16 tensor)
+_WCACHE = {}  # param -> {"version": v, variant: tensor}
 
 
-def bf16_weight(w: torch.Tensor) -> torch.Tensor:
-    """Cached fp32->bf16 cast of a master weight (re-cast after each step)."""
-    if w.dtype == torch.bfloat16:
-        return w
-    ent = _BF16_CACHE.get(w)
-    if ent is not None and ent[0] == w._version:
-        return ent[1]
-    wb = w.detach().to(torch.bfloat16)
-    _BF16_CACHE[w] = (w._version, wb)
-    return wb
+def weight_variant(w: torch.Tensor, variant: str) -> torch.Tensor:
+    """Per-step cached transforms of fp32 master weights.
+
+    variants: "flat"  — bf16, native contiguous (linear [N,K])
+              "nhwc"  — bf16 channels_last (conv KRSC memory)
+              "wt2"   — bf16 [C, R*S*K] (conv dgrad operand)
+    """
+    ent = _WCACHE.get(w)
+    if ent is None or ent["version"] != w._version:
+        ent = {"version": w._version}
+        _WCACHE[w] = ent
+    if variant not in ent:
+        wb = w.detach().to(torch.bfloat16)
+        if variant == "flat":
+            ent[variant] = wb.contiguous()
+        elif variant == "nhwc":
+            ent[variant] = wb.contiguous(memory_format=CL)
+        elif variant == "wt2":
+            K, C, R, S = w.shape
+            ent[variant] = wb.permute(1, 2, 3, 0).reshape(C, R * S * K).contiguous()
+        else:
+            raise KeyError(variant)
+    return ent[variant]
 
 
 def clear_weight_cache() -> None:
-    _BF16_CACHE.clear()
+    _WCACHE.clear()
+
+
+def bf16_weight(w: torch.Tensor) -> torch.Tensor:
+    if w.dtype == torch.bfloat16:
+        return w
+    return weight_variant(w, "flat")
 
 
 def _to_bf16(x: torch.Tensor) -> torch.Tensor:
     return x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
+
+
+def _chlast(x: torch.Tensor) -> torch.Tensor:
+    if x.dim() == 4:
+        return x.contiguous(memory_format=CL)
+    return x.contiguous()
+
+
+def _nhwc_2d(x: torch.Tensor) -> torch.Tensor:
+    """Zero-copy [N*H*W, C] view of a channels_last 4D tensor."""
+    return x.permute(0, 2, 3, 1).reshape(-1, x.shape[1])
 
 
 # ----------------------------------------------------------------- conv2d ---
@@ -52,30 +82,33 @@ class _HIPConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
         ext = require_ext_for(x)
-        xb = _to_bf16(x.contiguous())
-        wb = bf16_weight(weight)
-        y = ext.conv2d_fwd(xb, wb, stride, padding)
-        if bias is not None:
-            y += bias.to(y.dtype).view(1, -1, 1, 1)
-        ctx.save_for_backward(xb, wb)
+        xb = _chlast(_to_bf16(x))
+        wb = weight_variant(weight, "nhwc")
+        bias_f = bias if bias is None else bias.detach().float()
+        y = ext.conv2d_fwd(xb, wb, bias_f, stride, padding)
+        ctx.save_for_backward(xb, weight)
         ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        xb, wb = ctx.saved_tensors
+        xb, weight = ctx.saved_tensors
         stride, padding, x_dtype, w_shape, has_bias = ctx.meta
         ext = require_ext_for(dy)
-        dyb = _to_bf16(dy.contiguous())
+        dyb = _chlast(_to_bf16(dy))
+        K, C, R, S = w_shape
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = ext.conv2d_dgrad(dyb, wb, xb.shape[2], xb.shape[3], stride, padding)
-            if x_dtype != dx.dtype:
-                dx = dx.to(x_dtype)
+            wt2 = weight_variant(weight, "wt2")
+            dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], C,
+                                  xb.shape[2], xb.shape[3], R, S,
+                                  stride, padding)
+            if x_dtype == torch.float32:
+                dx = dx.float()
         if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(dyb, xb, w_shape[2], w_shape[3], stride, padding)
+            dw = ext.conv2d_wgrad(dyb, xb, R, S, stride, padding)
         if has_bias and ctx.needs_input_grad[2]:
-            db = dyb.float().sum(dim=(0, 2, 3))
+            db = ext.col_sum(_nhwc_2d(dyb))
         return dx, dw, db, None, None
 
 
@@ -85,7 +118,7 @@ def conv2d(x, weight, bias=None, stride=1, padding=0):
     if x.is_cuda and require_ext_for(x) is not None:
         return _HIPConv2d.apply(x, weight, bias, stride, padding)
     if x.is_cuda and amp_mod.is_enabled():  # explicit torch-fallback bring-up path
-        return F.conv2d(_to_bf16(x), bf16_weight(weight),
+        return F.conv2d(_to_bf16(x), bf16_weight(weight).view_as(weight),
                         bias.to(torch.bfloat16) if bias is not None else None,
                         stride, padding)
     return F.conv2d(x, weight, bias, stride, padding)
@@ -98,8 +131,9 @@ class _HIPLinear(torch.autograd.Function):
     def forward(ctx, x, weight, bias):
         ext = require_ext_for(x)
         xb = _to_bf16(x.contiguous())
-        wb = bf16_weight(weight)
-        y = ext.linear_fwd(xb, wb, bias if bias is not None else None)
+        wb = weight_variant(weight, "flat")
+        bias_f = bias if bias is None else bias.detach().float()
+        y = ext.linear_fwd(xb, wb, bias_f)
         ctx.save_for_backward(xb, wb)
         ctx.meta = (x.dtype, bias is not None)
         return y
@@ -112,13 +146,13 @@ class _HIPLinear(torch.autograd.Function):
         dyb = _to_bf16(dy.contiguous())
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = ext.linear_dgrad(dyb, wb)          # dy [M,N] @ W [N,K] -> [M,K]
-            if dx.dtype != x_dtype and x_dtype == torch.float32:
-                pass  # keep bf16 grads flowing between bf16 layers
+            dx = ext.linear_dgrad(dyb, wb)
+            if x_dtype == torch.float32:
+                dx = dx.float()
         if ctx.needs_input_grad[1]:
-            dw = ext.linear_wgrad(dyb, xb)          # dy^T [N,M] @ X [M,K] -> fp32
+            dw = ext.linear_wgrad(dyb, xb)
         if has_bias and ctx.needs_input_grad[2]:
-            db = dyb.float().sum(dim=0)
+            db = ext.col_sum(dyb)
         return dx, dw, db
 
 
@@ -134,26 +168,25 @@ def linear(x, weight, bias=None):
 # ------------------------------------------------------------- batch norm ---
 
 class _HIPBatchNorm(torch.autograd.Function):
-    """BatchNorm (2d NCHW or 1d NC) with optional fused ReLU.
+    """BatchNorm (2d channels_last or 1d) with optional fused ReLU.
 
-    Stats in fp32 over bf16 activations; running stats updated in-place
-    (rank-local, broadcast from rank 0 each forward by the DDP wrapper —
-    reference DDP broadcast_buffers semantics).
-    """
+    fp32 stats over bf16 activations; running stats updated in-place
+    (rank-local; broadcast from rank 0 each forward by the DDP wrapper —
+    reference broadcast_buffers semantics)."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
                 training, momentum, eps, fuse_relu):
         ext = require_ext_for(x)
-        xb = _to_bf16(x.contiguous())
+        xb = _chlast(_to_bf16(x))
         if training:
             y, save_mean, save_invstd = ext.bn_fwd_train(
-                xb, weight, bias, running_mean, running_var,
+                xb, weight.detach(), bias.detach(), running_mean, running_var,
                 momentum, eps, fuse_relu)
             ctx.save_for_backward(xb, weight, save_mean, save_invstd, y)
         else:
-            y = ext.bn_fwd_eval(xb, weight, bias, running_mean, running_var,
-                                eps, fuse_relu)
+            y = ext.bn_fwd_eval(xb, weight.detach(), bias.detach(),
+                                running_mean, running_var, eps, fuse_relu)
         ctx.fuse_relu = fuse_relu
         ctx.training = training
         ctx.x_dtype = x.dtype
@@ -164,10 +197,10 @@ class _HIPBatchNorm(torch.autograd.Function):
         assert ctx.training, "backward through eval-mode BN is unsupported"
         xb, weight, save_mean, save_invstd, y = ctx.saved_tensors
         ext = require_ext_for(dy)
-        dyb = _to_bf16(dy.contiguous())
-        dx, dweight, dbias = ext.bn_bwd(xb, dyb, weight, save_mean,
+        dyb = _chlast(_to_bf16(dy))
+        dx, dweight, dbias = ext.bn_bwd(xb, dyb, weight.detach(), save_mean,
                                         save_invstd, y, ctx.fuse_relu)
-        if dx.dtype != ctx.x_dtype and ctx.x_dtype == torch.float32:
+        if ctx.x_dtype == torch.float32:
             dx = dx.float()
         return dx, dweight, dbias, None, None, None, None, None, None
 
@@ -191,7 +224,7 @@ class _HIPMaxPool2x2(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
         ext = require_ext_for(x)
-        xb = _to_bf16(x.contiguous())
+        xb = _chlast(_to_bf16(x))
         y, idx = ext.maxpool2x2_fwd(xb)
         ctx.save_for_backward(idx)
         ctx.x_shape = xb.shape
@@ -201,7 +234,7 @@ class _HIPMaxPool2x2(torch.autograd.Function):
     def backward(ctx, dy):
         (idx,) = ctx.saved_tensors
         ext = require_ext_for(dy)
-        dx = ext.maxpool2x2_bwd(_to_bf16(dy.contiguous()), idx,
+        dx = ext.maxpool2x2_bwd(_chlast(_to_bf16(dy)), idx,
                                 ctx.x_shape[2], ctx.x_shape[3])
         return dx
 
@@ -235,7 +268,7 @@ class _HIPCrossEntropy(torch.autograd.Function):
     def backward(ctx, dloss):
         logits, target, lse = ctx.saved_tensors
         ext = require_ext_for(logits)
-        dlogits = ext.ce_bwd(logits, target, lse, dloss)
+        dlogits = ext.ce_bwd(logits, target, lse, dloss.contiguous())
         return dlogits, None
 
 

@@ -1,0 +1,520 @@
+#include "hip/hip_runtime.h"
+// conv.hip — NHWC implicit-GEMM convolutions on MFMA.
+//
+// Layout: activations bf16 channels_last (NHWC), weights bf16 KRSC.
+// Forward treats conv as C[M=N·P·Q, Ko] = im2col(x)[M, R·S·C] · W[Ko, RSC]^T
+// with the im2col gather done on the fly into LDS — channels are the
+// fastest dim so the gathers are 16-byte vectors whenever C % 8 == 0.
+// Backward-data is the same structure over dy with transformed weights
+// WT2[C, R·S·Ko] (built once per step from the bf16 weight cache);
+// backward-weight contracts over M with LDS-transposed staging and a
+// fixed-order split-K slab reduction (deterministic, no atomics).
+// conv1-style Cin<8 layers take a direct VALU path (MFMA starved at K=9 —
+// SURVEY §2.4 note).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+
+constexpr int CBM = 128, CBN = 128, CBK = 64;
+constexpr int CLDK = CBK + 8;
+
+struct ConvShape {
+    int N, H, W, C;     // input
+    int Ko, R, S;       // weights
+    int P, Q;           // output
+    int stride, pad;
+};
+
+// ---------------------------------------------------------------- forward ---
+
+// A-gather: element (gm, gk): gm -> (n,p,q); gk -> (rs, c)
+// addr = x[((n*H + p*st + r - pad)*W + q*st + s - pad)*C + c]
+__global__ __launch_bounds__(256)
+void k_conv_fwd(const bf16* __restrict__ x, const bf16* __restrict__ w,
+                const float* __restrict__ bias, bf16* __restrict__ y,
+                ConvShape cs, int M, int Kgemm) {
+    __shared__ bf16 lds_a[CBM][CLDK];
+    __shared__ bf16 lds_b[CBN][CLDK];
+    const int m0 = blockIdx.x * CBM;
+    const int n0 = blockIdx.y * CBN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[4][4] = {};
+    const int ld_row = tid >> 3;
+    const int ld_col = (tid & 7) * 8;
+
+    for (int kt = 0; kt < Kgemm; kt += CBK) {
+        #pragma unroll
+        for (int p4 = 0; p4 < 4; ++p4) {
+            int row = p4 * 32 + ld_row;
+            int gm = m0 + row, gk = kt + ld_col;
+            bf16x8_t va = {};
+            if (gm < M && gk < Kgemm) {
+                int q = gm % cs.Q, rem = gm / cs.Q;
+                int p = rem % cs.P, n = rem / cs.P;
+                int c = gk % cs.C, rs = gk / cs.C;
+                int r = rs / cs.S, s = rs % cs.S;
+                int h = p * cs.stride + r - cs.pad;
+                int wcol = q * cs.stride + s - cs.pad;
+                if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W) {
+                    va = *reinterpret_cast<const bf16x8_t*>(
+                        &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+            }
+            *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
+
+            int gn = n0 + row;
+            bf16x8_t vb = {};
+            if (gn < cs.Ko && gk < Kgemm)
+                vb = *reinterpret_cast<const bf16x8_t*>(
+                    &w[(long)gn * Kgemm + gk]);
+            *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
+        }
+        __syncthreads();
+        #pragma unroll
+        for (int ks = 0; ks < CBK; ks += 32) {
+            bf16x8_t af[4], bfr[4];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = n0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= cs.Ko) continue;
+            float badd = bias ? bias[col] : 0.f;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                y[(long)row * cs.Ko + col] = f2bf(acc[mi][ni][r] + badd);
+            }
+        }
+}
+
+// conv with Cin < 8 (e.g. the MNIST stem, Cin=1): direct VALU kernel,
+// weights staged in LDS, thread computes 8 output channels of one (n,p,q).
+__global__ void k_conv_small_cin(const bf16* __restrict__ x,
+                                 const bf16* __restrict__ w,
+                                 const float* __restrict__ bias,
+                                 bf16* __restrict__ y, ConvShape cs, long M) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    bf16* wlds = reinterpret_cast<bf16*>(smem);   // [Ko][R*S*C]
+    int rsc = cs.R * cs.S * cs.C;
+    for (int i = threadIdx.x; i < cs.Ko * rsc; i += blockDim.x)
+        wlds[i] = w[i];
+    __syncthreads();
+    long total = M * (cs.Ko / 8);
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long gstride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += gstride) {
+        int ko8 = (i % (cs.Ko / 8)) * 8;
+        long gm = i / (cs.Ko / 8);
+        int q = gm % cs.Q; long rem = gm / cs.Q;
+        int p = rem % cs.P; int n = rem / cs.P;
+        float acc[8] = {};
+        for (int r = 0; r < cs.R; ++r) {
+            int h = p * cs.stride + r - cs.pad;
+            if (h < 0 || h >= cs.H) continue;
+            for (int s = 0; s < cs.S; ++s) {
+                int wc = q * cs.stride + s - cs.pad;
+                if (wc < 0 || wc >= cs.W) continue;
+                for (int c = 0; c < cs.C; ++c) {
+                    float xv = bf2f(x[(((long)n * cs.H + h) * cs.W + wc) * cs.C + c]);
+                    int kidx = (r * cs.S + s) * cs.C + c;
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        acc[j] = fmaf(xv, bf2f(wlds[(ko8 + j) * rsc + kidx]), acc[j]);
+                }
+            }
+        }
+        bf16x8_t o;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            o[j] = (short)f2us(acc[j] + (bias ? bias[ko8 + j] : 0.f));
+        *reinterpret_cast<bf16x8_t*>(&y[gm * cs.Ko + ko8]) = o;
+    }
+}
+
+// ---------------------------------------------------------- backward data ---
+
+// dx[M=N·H·W, C] = Σ_{r,s,ko} dy[n, (h+pad-r)/st, (w+pad-s)/st, ko] ·
+//                  WT2[c, (r·S+s)·Ko+ko]        (stride 1 only for now)
+__global__ __launch_bounds__(256)
+void k_conv_dgrad(const bf16* __restrict__ dy, const bf16* __restrict__ wt2,
+                  bf16* __restrict__ dx, ConvShape cs, int M, int Kgemm) {
+    __shared__ bf16 lds_a[CBM][CLDK];
+    __shared__ bf16 lds_b[CBN][CLDK];
+    const int m0 = blockIdx.x * CBM;
+    const int n0 = blockIdx.y * CBN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[4][4] = {};
+    const int ld_row = tid >> 3;
+    const int ld_col = (tid & 7) * 8;
+
+    for (int kt = 0; kt < Kgemm; kt += CBK) {
+        #pragma unroll
+        for (int p4 = 0; p4 < 4; ++p4) {
+            int row = p4 * 32 + ld_row;
+            int gm = m0 + row, gk = kt + ld_col;
+            bf16x8_t va = {};
+            if (gm < M && gk < Kgemm) {
+                int wcol = gm % cs.W; long rem = gm / cs.W;
+                int h = rem % cs.H; int n = rem / cs.H;
+                int ko = gk % cs.Ko, rs = gk / cs.Ko;
+                int r = rs / cs.S, s = rs % cs.S;
+                int p = h + cs.pad - r;
+                int q = wcol + cs.pad - s;
+                if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q) {
+                    va = *reinterpret_cast<const bf16x8_t*>(
+                        &dy[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko]);
+                }
+            }
+            *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
+
+            int gn = n0 + row;
+            bf16x8_t vb = {};
+            if (gn < cs.C && gk < Kgemm)
+                vb = *reinterpret_cast<const bf16x8_t*>(
+                    &wt2[(long)gn * Kgemm + gk]);
+            *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
+        }
+        __syncthreads();
+        #pragma unroll
+        for (int ks = 0; ks < CBK; ks += 32) {
+            bf16x8_t af[4], bfr[4];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = n0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= cs.C) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                dx[(long)row * cs.C + col] = f2bf(acc[mi][ni][r]);
+            }
+        }
+}
+
+// -------------------------------------------------------- backward weight ---
+
+// slab[s][ko, rsc] = Σ_{m in split s} dy[m][ko] · im2col(x)[m][rsc]
+// 64×64 output tile per block (4 waves, 32×32 each), contraction staged
+// 32-deep with LDS-transposed writes (sources are m-major).
+constexpr int WBM = 64, WBN = 64, WBK = 32;
+constexpr int WLDK = WBK + 8;
+
+__global__ __launch_bounds__(256)
+void k_conv_wgrad(const bf16* __restrict__ dy, const bf16* __restrict__ x,
+                  float* __restrict__ slab, ConvShape cs, long M, int Kgemm,
+                  int S) {
+    __shared__ bf16 lds_a[WBM][WLDK];   // [ko][m]
+    __shared__ bf16 lds_b[WBN][WLDK];   // [rsc][m]
+    const int ko0 = blockIdx.x * WBM;
+    const int rc0 = blockIdx.y * WBN;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[2][2] = {};
+    // staging assignment: thread -> (mloc = tid%32, j8 = (tid/32)*8)
+    const int mloc = tid & 31;
+    const int j8 = (tid >> 5) * 8;
+
+    const long m_begin = (long)split * WBK;
+    for (long mt = m_begin; mt < M; mt += (long)S * WBK) {
+        long gm = mt + mloc;
+        // A: dy[m][ko0+j8..+7] -> lds_a[j][m]
+        bf16x8_t va = {};
+        if (gm < M && ko0 + j8 < cs.Ko)
+            va = *reinterpret_cast<const bf16x8_t*>(
+                &dy[gm * cs.Ko + ko0 + j8]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) lds_a[j8 + j][mloc] = ((bf16*)&va)[j];
+        // B: im2col(x)[m][rc0+j8..+7] -> lds_b[j][m]
+        bf16x8_t vb = {};
+        int gk = rc0 + j8;
+        if (gm < M && gk < Kgemm) {
+            int q = gm % cs.Q; long rem = gm / cs.Q;
+            int p = rem % cs.P; int n = rem / cs.P;
+            int c = gk % cs.C, rs = gk / cs.C;
+            int r = rs / cs.S, s = rs % cs.S;
+            int h = p * cs.stride + r - cs.pad;
+            int wcol = q * cs.stride + s - cs.pad;
+            if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                vb = *reinterpret_cast<const bf16x8_t*>(
+                    &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) lds_b[j8 + j][mloc] = ((bf16*)&vb)[j];
+        __syncthreads();
+
+        // MFMA: contraction over the 32 staged m's
+        bf16x8_t af[2], bfr[2];
+        const int kcol = (lane >> 4) * 8;
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+            int col = rc0 + wc * 32 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
+// combine slab -> dw fp32 in torch KCRS layout (fixed order, deterministic)
+__global__ void k_wgrad_combine(const float* __restrict__ slab, int S,
+                                ConvShape cs, int Kgemm,
+                                float* __restrict__ dw) {
+    long total = (long)cs.Ko * Kgemm;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    int RS = cs.R * cs.S;
+    for (; i < total; i += stride) {
+        float v = 0.f;
+        for (int s = 0; s < S; ++s) v += slab[(long)s * total + i];
+        int ko = i / Kgemm;
+        int rc = i % Kgemm;
+        int c = rc % cs.C, rs = rc / cs.C;
+        dw[((long)ko * cs.C + c) * RS + rs] = v;
+    }
+}
+
+// conv1-style wgrad (Cin < 8): per-(rs, split) block, lanes along Ko
+__global__ void k_wgrad_small_cin(const bf16* __restrict__ dy,
+                                  const bf16* __restrict__ x,
+                                  float* __restrict__ slab, ConvShape cs,
+                                  long M, int S) {
+    int rsc = blockIdx.x;    // (r*S_ + s)*C + c  flattened small index
+    int split = blockIdx.y;
+    int c = rsc % cs.C, rsq = rsc / cs.C;
+    int r = rsq / cs.S, s = rsq % cs.S;
+    int ko = threadIdx.x & 63;
+    int walker = threadIdx.x >> 6;    // 4 walkers
+    __shared__ float red[4][64];
+    float acc = 0.f;
+    if (ko < cs.Ko) {
+        for (long gm = (long)split * 4 + walker; gm < M; gm += (long)S * 4) {
+            int q = gm % cs.Q; long rem = gm / cs.Q;
+            int p = rem % cs.P; int n = rem / cs.P;
+            int h = p * cs.stride + r - cs.pad;
+            int wcol = q * cs.stride + s - cs.pad;
+            if (h < 0 || h >= cs.H || wcol < 0 || wcol >= cs.W) continue;
+            float xv = bf2f(x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+            acc = fmaf(bf2f(dy[gm * cs.Ko + ko]), xv, acc);
+        }
+    }
+    red[walker][ko] = acc;
+    __syncthreads();
+    if (walker == 0 && ko < cs.Ko) {
+        float t = 0.f;
+        #pragma unroll
+        for (int w = 0; w < 4; ++w) t += red[w][ko];
+        // slab layout [S][rsc_total][Ko]
+        slab[((long)split * gridDim.x + rsc) * cs.Ko + ko] = t;
+    }
+}
+
+__global__ void k_wgrad_small_combine(const float* __restrict__ slab, int S,
+                                      ConvShape cs, int rsc_total,
+                                      float* __restrict__ dw) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= rsc_total * cs.Ko) return;
+    int ko = i % cs.Ko;
+    int rc = i / cs.Ko;
+    float v = 0.f;
+    for (int s = 0; s < S; ++s)
+        v += slab[((long)s * rsc_total + rc) * cs.Ko + ko];
+    int c = rc % cs.C, rs = rc / cs.C;
+    dw[((long)ko * cs.C + c) * (cs.R * cs.S) + rs] = v;
+}
+
+// ------------------------------------------------------------------ hosts ---
+
+static ConvShape make_shape(const at::Tensor& x, int Ko, int R, int S,
+                            int stride, int pad) {
+    ConvShape cs;
+    cs.N = x.size(0); cs.C = x.size(1); cs.H = x.size(2); cs.W = x.size(3);
+    cs.Ko = Ko; cs.R = R; cs.S = S; cs.stride = stride; cs.pad = pad;
+    cs.P = (cs.H + 2 * pad - R) / stride + 1;
+    cs.Q = (cs.W + 2 * pad - S) / stride + 1;
+    return cs;
+}
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
+                      c10::optional<at::Tensor> bias, long stride, long pad) {
+    // x: NCHW logical / channels_last physical bf16; w: KCRS logical /
+    // channels_last physical (= KRSC memory) bf16
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+    ConvShape cs = make_shape(x, w.size(0), w.size(2), w.size(3),
+                              (int)stride, (int)pad);
+    int Kgemm = cs.R * cs.S * cs.C;
+    long M = (long)cs.N * cs.P * cs.Q;
+    auto y = at::empty({cs.N, cs.Ko, cs.P, cs.Q},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = at::hip::getCurrentHIPStream();
+    const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+    const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+    const bf16* wp = reinterpret_cast<const bf16*>(w.data_ptr());
+    bf16* yp = reinterpret_cast<bf16*>(y.data_ptr());
+
+    if (cs.C < 8) {
+        TORCH_CHECK(cs.Ko % 8 == 0);
+        int lds = cs.Ko * Kgemm * 2;
+        long total = M * (cs.Ko / 8);
+        int blocks = std::min<long>(8192, ceil_div_i(total, 256));
+        hipLaunchKernelGGL(k_conv_small_cin, dim3(blocks), dim3(256), lds,
+                           stream.stream(), xp, wp, bp, yp, cs, M);
+    } else {
+        TORCH_CHECK(cs.C % 8 == 0, "conv fwd needs C % 8 == 0");
+        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, CBN));
+        hipLaunchKernelGGL(k_conv_fwd, grid, dim3(256), 0, stream.stream(),
+                           xp, wp, bp, yp, cs, (int)M, Kgemm);
+    }
+    HIP_CHECK_LAST();
+    return y;
+}
+
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
+                        long H, long W, long R, long S, long stride, long pad) {
+    // dy: NCHW logical / channels_last bf16 [N,Ko,P,Q]; wt2: [C, R*S*Ko] bf16
+    TORCH_CHECK(stride == 1, "dgrad stride>1 not implemented yet");
+    ConvShape cs;
+    cs.N = N; cs.C = C; cs.H = H; cs.W = W;
+    cs.Ko = dy.size(1); cs.P = dy.size(2); cs.Q = dy.size(3);
+    cs.R = R; cs.S = S; cs.stride = stride; cs.pad = pad;
+    TORCH_CHECK(cs.Ko % 8 == 0);
+    int Kgemm = cs.R * cs.S * cs.Ko;
+    long M = (long)cs.N * cs.H * cs.W;
+    auto dx = at::empty({(long)cs.N, (long)cs.C, (long)cs.H, (long)cs.W},
+                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, CBN));
+    hipLaunchKernelGGL(k_conv_dgrad, grid, dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const bf16*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16*>(wt2.data_ptr()),
+                       reinterpret_cast<bf16*>(dx.data_ptr()), cs, (int)M,
+                       Kgemm);
+    HIP_CHECK_LAST();
+    return dx;
+}
+
+at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
+                        long stride, long pad) {
+    ConvShape cs = make_shape(x, dy.size(1), (int)R, (int)S, (int)stride,
+                              (int)pad);
+    TORCH_CHECK(cs.P == dy.size(2) && cs.Q == dy.size(3));
+    long M = (long)cs.N * cs.P * cs.Q;
+    int Kgemm = cs.R * cs.S * cs.C;
+    auto stream = at::hip::getCurrentHIPStream();
+    auto dw = at::empty({(long)cs.Ko, (long)cs.C, (long)cs.R, (long)cs.S},
+                        x.options().dtype(at::kFloat));
+    const bf16* dyp = reinterpret_cast<const bf16*>(dy.data_ptr());
+    const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+
+    if (cs.C < 8) {
+        int rsc_total = Kgemm;
+        int S_ = 64;
+        auto slab = at::empty({S_, rsc_total, cs.Ko},
+                              x.options().dtype(at::kFloat));
+        hipLaunchKernelGGL(k_wgrad_small_cin, dim3(rsc_total, S_), dim3(256),
+                           0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, S_);
+        HIP_CHECK_LAST();
+        int total = rsc_total * cs.Ko;
+        hipLaunchKernelGGL(k_wgrad_small_combine,
+                           dim3(ceil_div_i(total, 256)), dim3(256), 0,
+                           stream.stream(), slab.data_ptr<float>(), S_, cs,
+                           rsc_total, dw.data_ptr<float>());
+        HIP_CHECK_LAST();
+        return dw;
+    }
+
+    TORCH_CHECK(cs.C % 8 == 0 && cs.Ko % 8 == 0);
+    int gk = ceil_div_i(cs.Ko, WBM), gr = ceil_div_i(Kgemm, WBN);
+    int S_ = 1;
+    while (gk * gr * S_ < 1024 && S_ < 64 && (M / (S_ * 2)) >= WBK) S_ *= 2;
+    auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
+                          x.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(k_conv_wgrad, dim3(gk, gr, S_), dim3(256), 0,
+                       stream.stream(), dyp, xp, slab.data_ptr<float>(), cs,
+                       M, Kgemm, S_);
+    HIP_CHECK_LAST();
+    long total = (long)cs.Ko * Kgemm;
+    int blocks = std::min<long>(4096, ceil_div_i(total, 256));
+    hipLaunchKernelGGL(k_wgrad_combine, dim3(blocks), dim3(256), 0,
+                       stream.stream(), slab.data_ptr<float>(), S_, cs, Kgemm,
+                       dw.data_ptr<float>());
+    HIP_CHECK_LAST();
+    return dw;
+}

@@ -1,0 +1,222 @@
+#include "hip/hip_runtime.h"
+// elementwise.hip — multi-tensor optimizer/amp kernels + casts.
+//
+// Implements the implicit native surface of apex amp_C + the fused SGD step
+// (SURVEY N4/N12/N13): multi-tensor unscale+inf-check over the DDP bucket
+// flats, fused SGD(momentum, nesterov) [+ Lookahead interpolation] over the
+// whole parameter list in one launch, bf16<->fp32 casts.
+//
+// Multi-tensor scheme: host packs up to MT_MAX tensor pointers + a prefix
+// of chunk offsets into the kernarg struct; blocks binary-map to
+// (tensor, chunk) so one launch covers the whole list.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+#include <vector>
+
+constexpr int MT_MAX = 32;
+constexpr int MT_CHUNK = 1 << 16;  // elements per block-chunk
+constexpr int MT_BLOCK = 256;
+
+struct MTMeta {
+    const float* __restrict__ a[MT_MAX];
+    float* __restrict__ b[MT_MAX];
+    float* __restrict__ c[MT_MAX];
+    long sizes[MT_MAX];
+    int chunk_start[MT_MAX + 1];  // prefix sum of per-tensor chunk counts
+    int ntensors;
+};
+
+DEV_INLINE int find_tensor(const MTMeta& m, int chunk, int& local_chunk) {
+    // linear scan (ntensors <= 32): cheap relative to the memory work
+    int t = 0;
+    while (t + 1 < m.ntensors && m.chunk_start[t + 1] <= chunk) ++t;
+    local_chunk = chunk - m.chunk_start[t];
+    return t;
+}
+
+// ---------------------------------------------------------------- unscale ---
+
+__global__ void k_mt_unscale(MTMeta m, float inv_scale, float* found_inf) {
+    int local;
+    int t = find_tensor(m, blockIdx.x, local);
+    float* g = m.b[t];
+    long n = m.sizes[t];
+    long base = (long)local * MT_CHUNK;
+    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
+    bool bad = false;
+    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+        float v = g[i] * inv_scale;
+        if (!isfinite(v)) bad = true;
+        g[i] = v;
+    }
+    if (bad) *found_inf = 1.0f;  // racy same-value store: any writer sets it
+}
+
+void multi_tensor_unscale(std::vector<at::Tensor> grads, at::Tensor found_inf,
+                          double inv_scale) {
+    auto stream = at::hip::getCurrentHIPStream();
+    for (size_t start = 0; start < grads.size(); start += MT_MAX) {
+        MTMeta m{};
+        int chunks = 0;
+        int nt = 0;
+        for (size_t i = start; i < std::min(grads.size(), start + MT_MAX); ++i) {
+            auto& g = grads[i];
+            TORCH_CHECK(g.is_cuda() && g.scalar_type() == at::kFloat &&
+                        g.is_contiguous(), "unscale expects contiguous fp32");
+            m.b[nt] = g.data_ptr<float>();
+            m.sizes[nt] = g.numel();
+            m.chunk_start[nt] = chunks;
+            chunks += ceil_div_i(g.numel(), MT_CHUNK);
+            ++nt;
+        }
+        m.chunk_start[nt] = chunks;
+        m.ntensors = nt;
+        if (chunks == 0) continue;
+        hipLaunchKernelGGL(k_mt_unscale, dim3(chunks), dim3(MT_BLOCK), 0,
+                           stream.stream(), m, (float)inv_scale,
+                           found_inf.data_ptr<float>());
+        HIP_CHECK_LAST();
+    }
+}
+
+// -------------------------------------------------------------- fused SGD ---
+
+// buf = mu*buf + g ; d = nesterov ? g + mu*buf : buf ; p -= lr*d
+// If found_inf != nullptr and *found_inf != 0, the step is a device-side
+// no-op (amp overflow skip without a host sync).
+__global__ void k_mt_sgd(MTMeta m, float lr, float mu, float wd, float nesterov,
+                         const float* __restrict__ found_inf) {
+    if (found_inf && *found_inf != 0.0f) return;
+    int local;
+    int t = find_tensor(m, blockIdx.x, local);
+    const float* __restrict__ g = m.a[t];
+    float* __restrict__ p = m.b[t];
+    float* __restrict__ buf = m.c[t];
+    long n = m.sizes[t];
+    long base = (long)local * MT_CHUNK;
+    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
+    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+        float gi = g[i];
+        if (wd != 0.0f) gi = fmaf(wd, p[i], gi);
+        float d = gi;
+        if (mu != 0.0f) {
+            float b = fmaf(mu, buf[i], gi);
+            buf[i] = b;
+            d = (nesterov != 0.0f) ? fmaf(mu, b, gi) : b;
+        }
+        p[i] = fmaf(-lr, d, p[i]);
+    }
+}
+
+void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> bufs, double lr, double momentum,
+               double wd, double nesterov,
+               c10::optional<at::Tensor> found_inf) {
+    TORCH_CHECK(params.size() == grads.size());
+    bool has_mu = momentum != 0.0;
+    TORCH_CHECK(!has_mu || bufs.size() == params.size());
+    auto stream = at::hip::getCurrentHIPStream();
+    const float* fi = found_inf.has_value()
+        ? found_inf->data_ptr<float>() : nullptr;
+    for (size_t start = 0; start < params.size(); start += MT_MAX) {
+        MTMeta m{};
+        int chunks = 0, nt = 0;
+        for (size_t i = start; i < std::min(params.size(), start + MT_MAX); ++i) {
+            m.a[nt] = grads[i].data_ptr<float>();
+            m.b[nt] = params[i].data_ptr<float>();
+            m.c[nt] = has_mu ? bufs[i].data_ptr<float>() : nullptr;
+            m.sizes[nt] = params[i].numel();
+            m.chunk_start[nt] = chunks;
+            chunks += ceil_div_i(params[i].numel(), MT_CHUNK);
+            ++nt;
+        }
+        m.chunk_start[nt] = chunks;
+        m.ntensors = nt;
+        if (chunks == 0) continue;
+        hipLaunchKernelGGL(k_mt_sgd, dim3(chunks), dim3(MT_BLOCK), 0,
+                           stream.stream(), m, (float)lr, (float)momentum,
+                           (float)wd, (float)nesterov, fi);
+        HIP_CHECK_LAST();
+    }
+}
+
+// --------------------------------------------------------------- lookahead ---
+
+// slow += alpha * (fast - slow); fast = slow     (reference lookahead.py:19-27)
+__global__ void k_mt_lookahead(MTMeta m, float alpha,
+                               const float* __restrict__ found_inf) {
+    if (found_inf && *found_inf != 0.0f) return;
+    int local;
+    int t = find_tensor(m, blockIdx.x, local);
+    float* __restrict__ fast = m.b[t];
+    float* __restrict__ slow = m.c[t];
+    long n = m.sizes[t];
+    long base = (long)local * MT_CHUNK;
+    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
+    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+        float s = fmaf(alpha, fast[i] - slow[i], slow[i]);
+        slow[i] = s;
+        fast[i] = s;
+    }
+}
+
+void fused_lookahead(std::vector<at::Tensor> fast, std::vector<at::Tensor> slow,
+                     double alpha, c10::optional<at::Tensor> found_inf) {
+    TORCH_CHECK(fast.size() == slow.size());
+    auto stream = at::hip::getCurrentHIPStream();
+    const float* fi = found_inf.has_value()
+        ? found_inf->data_ptr<float>() : nullptr;
+    for (size_t start = 0; start < fast.size(); start += MT_MAX) {
+        MTMeta m{};
+        int chunks = 0, nt = 0;
+        for (size_t i = start; i < std::min(fast.size(), start + MT_MAX); ++i) {
+            m.b[nt] = fast[i].data_ptr<float>();
+            m.c[nt] = slow[i].data_ptr<float>();
+            m.sizes[nt] = fast[i].numel();
+            m.chunk_start[nt] = chunks;
+            chunks += ceil_div_i(fast[i].numel(), MT_CHUNK);
+            ++nt;
+        }
+        m.chunk_start[nt] = chunks;
+        m.ntensors = nt;
+        if (chunks == 0) continue;
+        hipLaunchKernelGGL(k_mt_lookahead, dim3(chunks), dim3(MT_BLOCK), 0,
+                           stream.stream(), m, (float)alpha, fi);
+        HIP_CHECK_LAST();
+    }
+}
+
+// ------------------------------------------------------------------- casts ---
+
+__global__ void k_f32_to_bf16(const float* __restrict__ in,
+                              unsigned short* __restrict__ out, long n) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < n / 4; i += stride) {
+        f32x4 v = reinterpret_cast<const f32x4*>(in)[i];
+        s16x4 o;
+        o[0] = f2us(v[0]); o[1] = f2us(v[1]); o[2] = f2us(v[2]); o[3] = f2us(v[3]);
+        reinterpret_cast<s16x4*>(out)[i] = o;
+    }
+    // tail
+    long tail = n & ~3L;
+    for (long j = tail + (blockIdx.x * blockDim.x + threadIdx.x);
+         j < n; j += stride)
+        out[j] = f2us(in[j]);
+}
+
+at::Tensor cast_to_bf16(at::Tensor x) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat && x.is_contiguous());
+    auto y = at::empty_like(x, x.options().dtype(at::kBFloat16));
+    long n = x.numel();
+    int blocks = std::min<long>(2048, (n / 4 + 255) / 256 + 1);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(k_f32_to_bf16, dim3(blocks), dim3(256), 0, stream.stream(),
+                       x.data_ptr<float>(),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()), n);
+    HIP_CHECK_LAST();
+    return y;
+}

@@ -1,0 +1,454 @@
+// gemm.hip — bf16 MFMA GEMM family for the linear layers plus
+// transpose / column-sum / slab-combine utilities shared with conv.
+//
+// Core kernel: C[M,N] = A[M,K] · B[N,K]^T ("tn": both operands row-major,
+// K fastest — the natural layout for X·W^T with torch Linear weights and,
+// after a one-off per-step weight transpose, for the dgrad/wgrad products).
+//
+// CDNA4 mapping: v_mfma_f32_16x16x32_bf16 per-wave tiles; 256-thread block
+// = 4 waves in a 2×2 grid, each wave owns a 64×64 output sub-tile as 4×4
+// fragments with fp32 accumulators in AGPRs.  LDS tiles are row-padded by
+// 8 bf16 (16 B) so the 64-lane ds_read_b128 fragment reads are
+// bank-conflict-free (row stride 144 B = 36 dwords; r*36 mod 64 covers
+// distinct multiples of 4 — see cdna_hip_programming.md §6 Guideline 4).
+// Split-K (grid.z) writes fp32 partial slabs, combined by a deterministic
+// fixed-order reduction (no fp atomics anywhere).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int LDK = BK + 8;          // padded row length (bf16 elements)
+constexpr int GEMM_THREADS = 256;
+
+// ------------------------------------------------------------ core kernel ---
+
+template <bool OUT_F32, bool SPLITK>
+__global__ __launch_bounds__(GEMM_THREADS)
+void k_gemm_tn(const bf16* __restrict__ A, const bf16* __restrict__ B,
+               const float* __restrict__ bias, void* __restrict__ Cout,
+               int M, int N, int K, int k_per_split) {
+    __shared__ bf16 lds_a[BM][LDK];
+    __shared__ bf16 lds_b[BN][LDK];
+
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int kz0 = SPLITK ? blockIdx.z * k_per_split : 0;
+    const int kz1 = SPLITK ? min(K, kz0 + k_per_split) : K;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;           // 4 waves: 2x2
+    const int wr = wid >> 1;            // wave row (0..1) -> 64 rows each
+    const int wc = wid & 1;             // wave col (0..1) -> 64 cols each
+
+    f32x4 acc[4][4] = {};
+
+    const int ld_row = tid >> 3;        // 32 rows per pass
+    const int ld_col = (tid & 7) * 8;   // 8 segments of 8 bf16 = 16B
+
+    for (int kt = kz0; kt < kz1; kt += BK) {
+        // ---- stage A and B tiles (zero-fill out of range) ----
+        #pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            int row = p * 32 + ld_row;
+            int gm = m0 + row, gk = kt + ld_col;
+            bf16x8_t va = {};
+            if (gm < M && gk + 7 < kz1) {
+                va = *reinterpret_cast<const bf16x8_t*>(&A[(long)gm * K + gk]);
+            } else if (gm < M) {
+                for (int j = 0; j < 8; ++j) {
+                    short v = 0;
+                    if (gk + j < kz1) v = ((const short*)A)[(long)gm * K + gk + j];
+                    va[j] = v;
+                }
+            }
+            *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
+
+            int gn = n0 + row;
+            bf16x8_t vb = {};
+            if (gn < N && gk + 7 < kz1) {
+                vb = *reinterpret_cast<const bf16x8_t*>(&B[(long)gn * K + gk]);
+            } else if (gn < N) {
+                for (int j = 0; j < 8; ++j) {
+                    short v = 0;
+                    if (gk + j < kz1) v = ((const short*)B)[(long)gn * K + gk + j];
+                    vb[j] = v;
+                }
+            }
+            *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
+        }
+        __syncthreads();
+
+        // ---- MFMA over the staged BK in two 32-deep steps ----
+        #pragma unroll
+        for (int ks = 0; ks < BK; ks += 32) {
+            bf16x8_t af[4], bf[4];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bf[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: D lane map col=lane&15, row=(lane>>4)*4+r ----
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = n0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= N) continue;
+            float badd = (bias && !SPLITK) ? bias[col] : 0.f;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                float v = acc[mi][ni][r] + badd;
+                if (SPLITK) {
+                    float* slab = reinterpret_cast<float*>(Cout);
+                    slab[((long)blockIdx.z * M + row) * N + col] = v;
+                } else if (OUT_F32) {
+                    reinterpret_cast<float*>(Cout)[(long)row * N + col] = v;
+                } else {
+                    reinterpret_cast<bf16*>(Cout)[(long)row * N + col] =
+                        f2bf(v);
+                }
+            }
+        }
+    }
+}
+
+// deterministic split-K combine: out[i] = bias-free fixed-order sum of slabs
+template <bool OUT_F32>
+__global__ void k_splitk_combine(const float* __restrict__ slab, int S,
+                                 long MN, int N, const float* __restrict__ bias,
+                                 void* __restrict__ out) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < MN; i += stride) {
+        float v = 0.f;
+        for (int s = 0; s < S; ++s) v += slab[(long)s * MN + i];
+        if (bias) v += bias[i % N];
+        if (OUT_F32) reinterpret_cast<float*>(out)[i] = v;
+        else reinterpret_cast<bf16*>(out)[i] = f2bf(v);
+    }
+}
+
+// ----------------------------------------------- small-N / small-K kernels ---
+
+// C[M,N] = A[M,K] · B[N,K]^T for tiny N (<=16): thread per (m,n), vec8 K loop
+__global__ void k_gemm_tn_smalln(const bf16* __restrict__ A,
+                                 const bf16* __restrict__ B,
+                                 const float* __restrict__ bias,
+                                 bf16* __restrict__ C, int M, int N, int K) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= (long)M * N) return;
+    int n = i % N;
+    long m = i / N;
+    const bf16* a = A + m * K;
+    const bf16* b = B + (long)n * K;
+    float acc = 0.f;
+    int k = 0;
+    for (; k + 8 <= K; k += 8) {
+        bf16x8_t va = *reinterpret_cast<const bf16x8_t*>(&a[k]);
+        bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(&b[k]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            acc = fmaf(us2f((unsigned short)va[j]), us2f((unsigned short)vb[j]), acc);
+    }
+    for (; k < K; ++k) acc = fmaf(bf2f(a[k]), bf2f(b[k]), acc);
+    if (bias) acc += bias[n];
+    C[i] = f2bf(acc);
+}
+
+// C[M,N] = A[M,K] · B[K,N] for tiny K: thread per (m, 8-wide n chunk)
+__global__ void k_gemm_nn_smallk(const bf16* __restrict__ A,
+                                 const bf16* __restrict__ B,
+                                 bf16* __restrict__ C, int M, int N, int K) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long nv = N / 8;
+    if (i >= (long)M * nv) return;
+    int nb = (i % nv) * 8;
+    long m = i / nv;
+    float acc[8] = {};
+    for (int k = 0; k < K; ++k) {
+        float a = bf2f(A[m * K + k]);
+        bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(&B[(long)k * N + nb]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            acc[j] = fmaf(a, us2f((unsigned short)vb[j]), acc[j]);
+    }
+    bf16x8_t o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (short)f2us(acc[j]);
+    *reinterpret_cast<bf16x8_t*>(&C[m * N + nb]) = o;
+}
+
+// dW[N,K] += dY^T·X for tiny N: grid.z splits over M into fp32 slab
+__global__ void k_wgrad_smalln(const bf16* __restrict__ dY,
+                               const bf16* __restrict__ X,
+                               float* __restrict__ slab,
+                               int M, int N, int K, int S) {
+    int s = blockIdx.z;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long kv = K / 8;
+    if (i >= (long)N * kv) return;
+    int kb = (i % kv) * 8;
+    int n = i / kv;
+    float acc[8] = {};
+    for (long m = s; m < M; m += S) {
+        float g = bf2f(dY[m * N + n]);
+        bf16x8_t vx = *reinterpret_cast<const bf16x8_t*>(&X[m * K + kb]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            acc[j] = fmaf(g, us2f((unsigned short)vx[j]), acc[j]);
+    }
+    float* out = slab + ((long)s * N + n) * K + kb;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = acc[j];
+}
+
+// ------------------------------------------------------------- transposes ---
+
+// bf16 2D transpose: out[j,i] = in[i,j]; 32x32 LDS tiles (+1 pad)
+__global__ void k_transpose_bf16(const bf16* __restrict__ in,
+                                 bf16* __restrict__ out, int R, int Ccols) {
+    __shared__ bf16 tile[32][33];
+    int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
+    int tc = threadIdx.x % 32, tr = threadIdx.x / 32;  // 256 thr: 32x8
+    #pragma unroll
+    for (int rr = 0; rr < 32; rr += 8) {
+        int r = r0 + tr + rr, c = c0 + tc;
+        tile[tr + rr][tc] = (r < R && c < Ccols) ? in[(long)r * Ccols + c]
+                                                 : f2bf(0.f);
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int rr = 0; rr < 32; rr += 8) {
+        int r = r0 + tc, c = c0 + tr + rr;   // write transposed
+        if (r < R && c < Ccols)
+            out[(long)c * R + r] = tile[tc][tr + rr];
+    }
+}
+
+// -------------------------------------------------------------- column sum ---
+
+// colsum[c] = sum_m in[m][c] (bf16 in, fp32 out), deterministic slab scheme
+__global__ void k_colsum_partial(const bf16* __restrict__ x, long M, int C,
+                                 int S, float* __restrict__ slab) {
+    int cb = blockIdx.x * 64 + (threadIdx.x % 64);
+    int s = blockIdx.y;
+    int walker = threadIdx.x / 64;
+    __shared__ float red[4][64];
+    float sum = 0.f;
+    if (cb < C) {
+        for (long r = s * 4 + walker; r < M; r += (long)S * 4)
+            sum += bf2f(x[r * C + cb]);
+    }
+    red[walker][threadIdx.x % 64] = sum;
+    __syncthreads();
+    if (walker == 0 && cb < C) {
+        float t = 0.f;
+        #pragma unroll
+        for (int w = 0; w < 4; ++w) t += red[w][threadIdx.x % 64];
+        slab[(long)s * C + cb] = t;
+    }
+}
+
+__global__ void k_colsum_combine(const float* __restrict__ slab, int S, int C,
+                                 float* __restrict__ out) {
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float v = 0.f;
+    for (int s = 0; s < S; ++s) v += slab[(long)s * C + c];
+    out[c] = v;
+}
+
+// ------------------------------------------------------------------ hosts ---
+
+static void launch_gemm_tn(const at::Tensor& A, const at::Tensor& B,
+                           const c10::optional<at::Tensor>& bias,
+                           at::Tensor& C, bool out_f32) {
+    int M = A.size(0), K = A.size(1), N = B.size(0);
+    auto stream = at::hip::getCurrentHIPStream();
+    const bf16* a = reinterpret_cast<const bf16*>(A.data_ptr());
+    const bf16* b = reinterpret_cast<const bf16*>(B.data_ptr());
+    const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+
+    int gm = ceil_div_i(M, BM), gn = ceil_div_i(N, BN);
+    // split-K when the tile grid underfills the 256-CU chip
+    int S = 1;
+    while (gm * gn * S < 512 && S < 16 && (K / (S * 2)) >= BK) S *= 2;
+    if (S > 1) {
+        int k_per_split = ceil_div_i(K, S);
+        k_per_split = ((k_per_split + BK - 1) / BK) * BK;
+        S = ceil_div_i(K, k_per_split);  // recompute actual
+        auto slab = at::empty({S, M, N}, A.options().dtype(at::kFloat));
+        hipLaunchKernelGGL((k_gemm_tn<false, true>), dim3(gm, gn, S),
+                           dim3(GEMM_THREADS), 0, stream.stream(), a, b,
+                           nullptr, slab.data_ptr<float>(), M, N, K,
+                           k_per_split);
+        HIP_CHECK_LAST();
+        long MN = (long)M * N;
+        int blocks = std::min<long>(4096, ceil_div_i(MN, 256));
+        if (out_f32)
+            hipLaunchKernelGGL(k_splitk_combine<true>, dim3(blocks), dim3(256),
+                               0, stream.stream(), slab.data_ptr<float>(), S,
+                               MN, N, bp, C.data_ptr());
+        else
+            hipLaunchKernelGGL(k_splitk_combine<false>, dim3(blocks), dim3(256),
+                               0, stream.stream(), slab.data_ptr<float>(), S,
+                               MN, N, bp, C.data_ptr());
+        HIP_CHECK_LAST();
+    } else {
+        if (out_f32)
+            hipLaunchKernelGGL((k_gemm_tn<true, false>), dim3(gm, gn, 1),
+                               dim3(GEMM_THREADS), 0, stream.stream(), a, b,
+                               bp, C.data_ptr(), M, N, K, K);
+        else
+            hipLaunchKernelGGL((k_gemm_tn<false, false>), dim3(gm, gn, 1),
+                               dim3(GEMM_THREADS), 0, stream.stream(), a, b,
+                               bp, C.data_ptr(), M, N, K, K);
+        HIP_CHECK_LAST();
+    }
+}
+
+at::Tensor gemm_tn(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> bias,
+                   bool out_f32) {
+    TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+                A.is_contiguous());
+    TORCH_CHECK(B.is_cuda() && B.scalar_type() == at::kBFloat16 &&
+                B.is_contiguous());
+    TORCH_CHECK(A.size(1) == B.size(1), "K mismatch");
+    int M = A.size(0), N = B.size(0);
+    auto C = at::empty({M, N}, A.options().dtype(out_f32 ? at::kFloat
+                                                         : at::kBFloat16));
+    launch_gemm_tn(A, B, bias, C, out_f32);
+    return C;
+}
+
+at::Tensor transpose_bf16(at::Tensor x) {
+    TORCH_CHECK(x.dim() == 2 && x.scalar_type() == at::kBFloat16 &&
+                x.is_contiguous());
+    int R = x.size(0), C = x.size(1);
+    auto out = at::empty({C, R}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(k_transpose_bf16,
+                       dim3(ceil_div_i(C, 32), ceil_div_i(R, 32)), dim3(256),
+                       0, stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<bf16*>(out.data_ptr()), R, C);
+    HIP_CHECK_LAST();
+    return out;
+}
+
+at::Tensor col_sum(at::Tensor x) {
+    // x viewed as [M, C] bf16 (rows contiguous); returns fp32 [C]
+    TORCH_CHECK(x.dim() == 2 && x.scalar_type() == at::kBFloat16 &&
+                x.is_contiguous());
+    long M = x.size(0);
+    int C = x.size(1);
+    int S = (int)std::max<long>(1, std::min<long>(64, M / 4));
+    auto slab = at::empty({S, C}, x.options().dtype(at::kFloat));
+    auto out = at::empty({C}, x.options().dtype(at::kFloat));
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(k_colsum_partial, dim3(ceil_div_i(C, 64), S), dim3(256),
+                       0, stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()), M, C, S,
+                       slab.data_ptr<float>());
+    HIP_CHECK_LAST();
+    hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(C, 256)), dim3(256),
+                       0, stream.stream(), slab.data_ptr<float>(), S, C,
+                       out.data_ptr<float>());
+    HIP_CHECK_LAST();
+    return out;
+}
+
+// --------------------------------------------------------- linear wrappers ---
+
+at::Tensor linear_fwd(at::Tensor x, at::Tensor w,
+                      c10::optional<at::Tensor> bias) {
+    // x [M,K] bf16, w [N,K] bf16, bias fp32 -> y [M,N] bf16
+    int M = x.size(0), N = w.size(0), K = x.size(1);
+    auto stream = at::hip::getCurrentHIPStream();
+    if (N <= 16) {
+        auto y = at::empty({M, N}, x.options());
+        long total = (long)M * N;
+        int blocks = ceil_div_i(total, 256);
+        const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+        hipLaunchKernelGGL(k_gemm_tn_smalln, dim3(blocks), dim3(256), 0,
+                           stream.stream(),
+                           reinterpret_cast<const bf16*>(x.data_ptr()),
+                           reinterpret_cast<const bf16*>(w.data_ptr()), bp,
+                           reinterpret_cast<bf16*>(y.data_ptr()), M, N, K);
+        HIP_CHECK_LAST();
+        return y;
+    }
+    return gemm_tn(x, w, bias, /*out_f32=*/false);
+}
+
+at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w) {
+    // dy [M,N] bf16, w [N,K] -> dx [M,K] = dy·w
+    int M = dy.size(0), N = dy.size(1), K = w.size(1);
+    auto stream = at::hip::getCurrentHIPStream();
+    if (N <= 16) {           // tiny contraction: nn-smallk over w directly
+        TORCH_CHECK(K % 8 == 0);
+        auto dx = at::empty({M, K}, dy.options());
+        long total = (long)M * (K / 8);
+        int blocks = ceil_div_i(total, 256);
+        hipLaunchKernelGGL(k_gemm_nn_smallk, dim3(blocks), dim3(256), 0,
+                           stream.stream(),
+                           reinterpret_cast<const bf16*>(dy.data_ptr()),
+                           reinterpret_cast<const bf16*>(w.data_ptr()),
+                           reinterpret_cast<bf16*>(dx.data_ptr()), M, K, N);
+        HIP_CHECK_LAST();
+        return dx;
+    }
+    auto wt = transpose_bf16(w);                  // [K,N]
+    return gemm_tn(dy, wt, c10::nullopt, false);  // [M,K]
+}
+
+at::Tensor linear_wgrad(at::Tensor dy, at::Tensor x) {
+    // dy [M,N], x [M,K] -> dw fp32 [N,K] = dy^T·x
+    int M = dy.size(0), N = dy.size(1), K = x.size(1);
+    auto stream = at::hip::getCurrentHIPStream();
+    if (N <= 16) {
+        TORCH_CHECK(K % 8 == 0);
+        int S = 32;
+        auto slab = at::empty({S, N, K}, dy.options().dtype(at::kFloat));
+        long total = (long)N * (K / 8);
+        int blocks = ceil_div_i(total, 256);
+        hipLaunchKernelGGL(k_wgrad_smalln, dim3(blocks, 1, S), dim3(256), 0,
+                           stream.stream(),
+                           reinterpret_cast<const bf16*>(dy.data_ptr()),
+                           reinterpret_cast<const bf16*>(x.data_ptr()),
+                           slab.data_ptr<float>(), M, N, K, S);
+        HIP_CHECK_LAST();
+        auto dw = at::empty({N, K}, dy.options().dtype(at::kFloat));
+        long MN = (long)N * K;
+        int cblocks = std::min<long>(4096, ceil_div_i(MN, 256));
+        hipLaunchKernelGGL(k_splitk_combine<true>, dim3(cblocks), dim3(256), 0,
+                           stream.stream(), slab.data_ptr<float>(), S, MN, K,
+                           nullptr, dw.data_ptr());
+        HIP_CHECK_LAST();
+        return dw;
+    }
+    auto dyt = transpose_bf16(dy);  // [N,M]
+    auto xt = transpose_bf16(x);    // [K,M]
+    return gemm_tn(dyt, xt, c10::nullopt, /*out_f32=*/true);  // [N,K] fp32
+}

@@ -59,6 +59,12 @@ class Lookahead(Optimizer):
             slows.append(param_state["slow_param"])
         if not fasts:
             return
+        if fasts[0].is_cuda:
+            from ..ops import load_extension
+            ext = load_extension(required=False)
+            if ext is not None:
+                ext.fused_lookahead(fasts, slows, self.alpha, None)
+                return
         # slow = slow + alpha*(fast - slow) == lerp(slow, fast, alpha)
         torch._foreach_lerp_(slows, fasts, self.alpha)
         torch._foreach_copy_(fasts, slows)

@@ -1,0 +1,315 @@
+"""GPU kernel numerics: every HIP kernel vs a plain PyTorch fp32 reference
+computed on the same bf16-quantized inputs (removes input-quantization
+noise; what remains is accumulation order + output rounding).
+All tests @pytest.mark.gpu — they need an MI355X."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from ddp_tricks_amd.ops import load_extension
+    ext = load_extension(required=True)
+    DEV = torch.device("cuda:0")
+CL = torch.channels_last
+
+
+def _close(a, b, rel=2e-2, atol=1e-2, name=""):
+    a = a.float()
+    b = b.float()
+    err = (a - b).abs().max().item()
+    scale = b.abs().max().item()
+    assert err <= atol + rel * scale, f"{name}: err={err} scale={scale}"
+
+
+# ------------------------------------------------------------------ GEMM ---
+
+def test_gemm_tn_correct():
+    torch.manual_seed(0)
+    for M, N, K in [(256, 128, 512), (1024, 512, 8192), (100, 70, 130)]:
+        A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+        B = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+        C = ext.gemm_tn(A, B, None, False)
+        ref = A.float() @ B.float().t()
+        _close(C, ref, name=f"gemm {M}x{N}x{K}")
+
+
+def test_gemm_tn_asymmetric_detects_transpose():
+    # asymmetric pattern: catches operand/output transposition (guide G9)
+    M, N, K = 128, 128, 64
+    A = torch.zeros(M, K, device=DEV)
+    A[3, :] = 1.0
+    B = torch.arange(N, device=DEV, dtype=torch.float32).unsqueeze(1).repeat(1, K) / N
+    C = ext.gemm_tn(A.to(torch.bfloat16), B.to(torch.bfloat16), None, False)
+    ref = A @ B.t()
+    _close(C, ref, name="gemm asym")
+    assert C.float()[3, 77].item() != 0 and abs(
+        C.float()[3, 77].item() - ref[3, 77].item()) < 0.5
+
+
+def test_gemm_bias_and_f32_out():
+    M, N, K = 256, 256, 256
+    A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    B = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    bias = torch.randn(N, device=DEV)
+    C = ext.gemm_tn(A, B, bias, True)
+    assert C.dtype == torch.float32
+    ref = A.float() @ B.float().t() + bias
+    _close(C, ref, name="gemm bias f32")
+
+
+def test_gemm_splitk_path():
+    # shapes that trigger split-K (tile grid underfills)
+    M, N, K = 1024, 512, 8192
+    A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    B = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    C = ext.gemm_tn(A, B, None, False)
+    _close(C, A.float() @ B.float().t(), name="gemm splitk")
+
+
+def test_transpose_and_colsum():
+    x = torch.randn(300, 200, device=DEV).to(torch.bfloat16)
+    xt = ext.transpose_bf16(x)
+    assert torch.equal(xt.float(), x.float().t())
+    cs = ext.col_sum(x.contiguous())
+    _close(cs, x.float().sum(0), rel=1e-3, atol=0.5, name="colsum")
+
+
+def test_linear_fwd_small_n():
+    M, N, K = 1024, 10, 512
+    x = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    w = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    b = torch.randn(N, device=DEV)
+    y = ext.linear_fwd(x, w, b)
+    _close(y, x.float() @ w.float().t() + b, name="linear smalln")
+
+
+def test_linear_grads():
+    M, N, K = 1024, 512, 8192
+    dy = torch.randn(M, N, device=DEV).to(torch.bfloat16)
+    w = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    x = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    dx = ext.linear_dgrad(dy, w)
+    _close(dx, dy.float() @ w.float(), name="linear dgrad")
+    dw = ext.linear_wgrad(dy, x)
+    assert dw.dtype == torch.float32
+    _close(dw, dy.float().t() @ x.float(), rel=2e-2, atol=0.5, name="linear wgrad")
+
+
+# ------------------------------------------------------------------ conv ---
+
+def _conv_ref(x, w, b, stride, pad):
+    return torch.nn.functional.conv2d(x.float(), w.float(),
+                                      b.float() if b is not None else None,
+                                      stride, pad)
+
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 26, 26, 128, 3, 1, 0),    # conv2-like
+    (8, 128, 12, 12, 256, 3, 1, 0),   # conv3-like
+    (8, 256, 10, 10, 512, 3, 1, 0),   # conv4-like
+    (4, 64, 16, 16, 128, 3, 1, 1),    # padding path
+])
+def test_conv_fwd(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(1)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    b = torch.randn(K, device=DEV)
+    y = ext.conv2d_fwd(x, w, b, stride, pad)
+    ref = _conv_ref(x, w, b, stride, pad)
+    _close(y, ref, name=f"conv fwd {shape}")
+
+
+def test_conv_small_cin_fwd():
+    N, C, H, W, K, R = 8, 1, 28, 28, 64, 3
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    b = torch.randn(K, device=DEV)
+    y = ext.conv2d_fwd(x, w, b, 1, 0)
+    _close(y, _conv_ref(x, w, b, 1, 0), name="conv1 fwd")
+
+
+def test_conv_dgrad():
+    N, C, H, W, K, R = 8, 64, 26, 26, 128, 3
+    dy = torch.randn(N, K, H - 2, W - 2, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16)
+    wt2 = w.permute(1, 2, 3, 0).reshape(C, R * R * K).contiguous()
+    dx = ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R, 1, 0)
+    ref = torch.nn.grad.conv2d_input((N, C, H, W), w.float(), dy.float(),
+                                     stride=1, padding=0)
+    _close(dx, ref, name="conv dgrad")
+
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 26, 26, 128, 3, 0),
+    (8, 1, 28, 28, 64, 3, 0),         # small-Cin path
+])
+def test_conv_wgrad(shape):
+    N, C, H, W, K, R, pad = shape
+    torch.manual_seed(2)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    P = H + 2 * pad - R + 1
+    dy = torch.randn(N, K, P, P, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    dw = ext.conv2d_wgrad(dy, x, R, R, 1, pad)
+    assert dw.dtype == torch.float32 and dw.shape == (K, C, R, R)
+    ref = torch.nn.grad.conv2d_weight(x.float(), (K, C, R, R), dy.float(),
+                                      stride=1, padding=pad)
+    _close(dw, ref, rel=2e-2, atol=1.0, name=f"conv wgrad {shape}")
+
+
+# -------------------------------------------------------------------- BN ---
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_bn2d_fwd_bwd(relu):
+    N, C, H, W = 16, 64, 13, 13
+    torch.manual_seed(3)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    g = torch.randn(C, device=DEV).abs() + 0.5
+    b = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, relu)
+
+    xf = x.float()
+    ref_rm = torch.zeros(C)
+    ref_rv = torch.ones(C)
+    ref = torch.nn.functional.batch_norm(
+        xf, ref_rm.to(DEV), ref_rv.to(DEV), g, b, True, 0.1, 1e-5)
+    if relu:
+        ref = ref.relu()
+    _close(y, ref, name="bn fwd")
+    _close(rm, xf.mean(dim=(0, 2, 3)) * 0.1, rel=1e-2, atol=1e-3, name="bn rmean")
+
+    # backward vs autograd
+    xf2 = x.float().detach().requires_grad_(True)
+    g2 = g.detach().requires_grad_(True)
+    b2 = b.detach().requires_grad_(True)
+    ref2 = torch.nn.functional.batch_norm(
+        xf2, torch.zeros(C, device=DEV), torch.ones(C, device=DEV),
+        g2, b2, True, 0.1, 1e-5)
+    if relu:
+        ref2 = ref2.relu()
+    dy = torch.randn_like(ref2).to(torch.bfloat16)
+    ref2.backward(dy.float())
+    dx, dg, db = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si, y, relu)
+    _close(dg, g2.grad, rel=2e-2, atol=0.1, name="bn dgamma")
+    _close(db, b2.grad, rel=2e-2, atol=0.1, name="bn dbeta")
+    _close(dx, xf2.grad, rel=5e-2, atol=2e-2, name="bn dx")
+
+
+def test_bn1d_fwd():
+    N, C = 512, 512
+    x = torch.randn(N, C, device=DEV).to(torch.bfloat16)
+    g = torch.randn(C, device=DEV).abs() + 0.5
+    b = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True)
+    ref = torch.nn.functional.batch_norm(
+        x.float(), torch.zeros(C, device=DEV), torch.ones(C, device=DEV),
+        g, b, True, 0.1, 1e-5).relu()
+    _close(y, ref, name="bn1d fwd")
+
+
+def test_bn_eval_mode():
+    N, C, H, W = 8, 64, 12, 12
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    g = torch.randn(C, device=DEV)
+    b = torch.randn(C, device=DEV)
+    rm = torch.randn(C, device=DEV)
+    rv = torch.rand(C, device=DEV) + 0.5
+    y = ext.bn_fwd_eval(x, g, b, rm, rv, 1e-5, False)
+    ref = torch.nn.functional.batch_norm(x.float(), rm, rv, g, b, False, 0.1, 1e-5)
+    _close(y, ref, name="bn eval")
+
+
+# ------------------------------------------------------------------ pool ---
+
+def test_maxpool_fwd_bwd():
+    N, C, H, W = 8, 128, 24, 24
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    y, idx = ext.maxpool2x2_fwd(x)
+    ref = torch.nn.functional.max_pool2d(x.float(), 2)
+    _close(y, ref, rel=0, atol=1e-6, name="pool fwd")
+    dy = torch.randn_like(y).contiguous(memory_format=CL)
+    dx = ext.maxpool2x2_bwd(dy, idx, H, W)
+    xf = x.float().requires_grad_(True)
+    torch.nn.functional.max_pool2d(xf, 2).backward(dy.float())
+    _close(dx, xf.grad, rel=0, atol=1e-6, name="pool bwd")
+
+
+# -------------------------------------------------------------------- CE ---
+
+def test_ce_fwd_bwd():
+    B, C = 1024, 10
+    logits = torch.randn(B, C, device=DEV).to(torch.bfloat16)
+    target = torch.randint(0, C, (B,), device=DEV)
+    loss, lse = ext.ce_fwd(logits, target)
+    ref = torch.nn.functional.cross_entropy(logits.float(), target)
+    _close(loss, ref, rel=1e-3, atol=1e-3, name="ce fwd")
+    dloss = torch.tensor(0.37, device=DEV)
+    dl = ext.ce_bwd(logits, target, lse, dloss)
+    lf = logits.float().requires_grad_(True)
+    torch.nn.functional.cross_entropy(lf, target).backward(dloss)
+    _close(dl, lf.grad, rel=2e-2, atol=1e-4, name="ce bwd")
+
+
+def test_argmax_correct():
+    B, C = 2048, 10
+    logits = torch.randn(B, C, device=DEV).to(torch.bfloat16)
+    target = torch.randint(0, C, (B,), device=DEV)
+    n = ext.argmax_correct(logits, target)
+    ref = (logits.float().argmax(dim=1) == target).sum()
+    assert n.item() == ref.item()
+
+
+# --------------------------------------------------------------- optimizer ---
+
+def test_fused_sgd_matches_torch():
+    torch.manual_seed(4)
+    shapes = [(1000,), (64, 33), (7,)]
+    pa = [torch.randn(s, device=DEV) for s in shapes]
+    pb = [p.clone() for p in pa]
+    ga = [torch.randn(s, device=DEV) for s in shapes]
+    bufa = [torch.zeros(s, device=DEV) for s in shapes]
+    bufb = [torch.zeros(s, device=DEV) for s in shapes]
+    for step in range(5):
+        g2 = [g * (step + 1) for g in ga]
+        ext.fused_sgd(pa, g2, bufa, 0.1, 0.9, 0.0, 1.0, None)
+        # torch reference
+        for p, g, buf in zip(pb, g2, bufb):
+            buf.mul_(0.9).add_(g)
+            d = g + 0.9 * buf
+            p.add_(d, alpha=-0.1)
+    for a, b in zip(pa, pb):
+        assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_fused_sgd_skips_on_inf_flag():
+    p = [torch.ones(16, device=DEV)]
+    g = [torch.ones(16, device=DEV)]
+    buf = [torch.zeros(16, device=DEV)]
+    flag = torch.ones(1, device=DEV)
+    ext.fused_sgd(p, g, buf, 0.1, 0.9, 0.0, 1.0, flag)
+    assert torch.equal(p[0], torch.ones(16, device=DEV))
+
+
+def test_fused_lookahead():
+    fast = [torch.full((32,), 3.0, device=DEV)]
+    slow = [torch.full((32,), 1.0, device=DEV)]
+    ext.fused_lookahead(fast, slow, 0.5, None)
+    assert torch.allclose(slow[0], torch.full((32,), 2.0, device=DEV))
+    assert torch.allclose(fast[0], torch.full((32,), 2.0, device=DEV))
+
+
+def test_multi_tensor_unscale():
+    g = [torch.full((1000,), 8.0, device=DEV), torch.full((65537 * 2,), 4.0, device=DEV)]
+    fi = torch.zeros(1, device=DEV)
+    ext.multi_tensor_unscale(g, fi, 0.25)
+    assert torch.allclose(g[0], torch.full((1000,), 2.0, device=DEV))
+    assert torch.allclose(g[1], torch.full((65537 * 2,), 1.0, device=DEV))
+    assert fi.item() == 0.0
+    g[0][17] = float("inf")
+    ext.multi_tensor_unscale(g, fi, 1.0)
+    assert fi.item() == 1.0
