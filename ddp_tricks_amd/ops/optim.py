@@ -64,6 +64,11 @@ class FusedSGD(Optimizer):
             if ext is not None:
                 ext.fused_sgd(params, grads, bufs, lr, momentum, wd,
                               1.0 if nesterov else 0.0)
+                # raw-pointer writes don't bump Tensor._version — drop the
+                # per-version bf16 weight cast cache so the next forward
+                # re-casts the updated masters
+                from .functional import clear_weight_cache
+                clear_weight_cache()
                 return
         if wd != 0:
             grads = torch._foreach_add(grads, params, alpha=wd)

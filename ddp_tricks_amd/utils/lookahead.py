@@ -64,6 +64,8 @@ class Lookahead(Optimizer):
             ext = load_extension(required=False)
             if ext is not None:
                 ext.fused_lookahead(fasts, slows, self.alpha, None)
+                from ..ops.functional import clear_weight_cache
+                clear_weight_cache()  # raw writes don't bump _version
                 return
         # slow = slow + alpha*(fast - slow) == lerp(slow, fast, alpha)
         torch._foreach_lerp_(slows, fasts, self.alpha)
