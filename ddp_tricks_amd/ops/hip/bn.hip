@@ -2,58 +2,62 @@
 // epilogue, fp32 statistics over bf16 activations.
 //
 // Both cases reduce over M rows × C channels where M = N*H*W (2d,
-// channels_last) or N (1d): lanes run along C → fully coalesced.
-// Reductions are deterministic: S fixed split partials into a slab,
-// fixed-order combine (no fp atomics) — bit-reproducible under
-// same_seeds like the whole gradient path.
+// channels_last) or N (1d).  All passes are bf16x8-vectorized (G13: hipcc
+// does not auto-vectorize bf16 — scalar loads are 2-2.5× slower) and
+// C % 8 == 0 is required on the HIP path (the torch fallback covers the
+// rest).  Reductions are deterministic: per-split partials into a
+// [2][C][S] fp32 slab (combine waves read each channel's splits
+// coalesced), fixed-order combine — no fp atomics, bit-reproducible.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
 
-constexpr int BN_CB = 64;    // channels per block
-constexpr int BN_RW = 4;     // row-walkers per channel (block = 256 threads)
-
-static inline int bn_splits(long M, int C) {
-    // target >=1024 blocks for the partial pass, capped by row count
-    int cb = ceil_div_i(C, BN_CB);
-    int s = std::max(1, 1024 / std::max(cb, 1));
-    s = std::min<long>(s, (M + BN_RW - 1) / BN_RW);
-    return std::max(1, s);
+static inline int bn_splits(long M, int nw) {
+    long s = M / (nw * 4);
+    if (s < 1) s = 1;
+    if (s > 1024) s = 1024;
+    return (int)s;
 }
 
-// partial sums: slab[s][c] = {sum, sumsq} over rows s::S
+// partial sums: slab[0][c][s] = sum, slab[1][c][s] = sumsq over rows s::S
 __global__ void k_bn_partial(const bf16* __restrict__ x, long M, int C,
                              int S, float* __restrict__ slab) {
-    __shared__ float red[2][BN_RW][BN_CB];
-    int cb = blockIdx.x;           // channel block
-    int s = blockIdx.y;            // split
-    int c = cb * BN_CB + (threadIdx.x % BN_CB);
-    int walker = threadIdx.x / BN_CB;
-    float sum = 0.f, sq = 0.f;
-    if (c < C) {
-        for (long r = s * BN_RW + walker; r < M; r += (long)S * BN_RW) {
-            float v = bf2f(x[r * C + c]);
-            sum += v;
-            sq = fmaf(v, v, sq);
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* red = reinterpret_cast<float*>(smem);   // [2][nw][C]
+    const int cpg = C >> 3;
+    const int c8 = threadIdx.x % cpg;
+    const int walker = threadIdx.x / cpg;
+    const int nw = blockDim.x / cpg;
+    const int s = blockIdx.x;
+    float sum[8] = {}, sq[8] = {};
+    for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
+        s16x8 v = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = us2f((unsigned short)v[j]);
+            sum[j] += f;
+            sq[j] = fmaf(f, f, sq[j]);
         }
     }
-    red[0][walker][threadIdx.x % BN_CB] = sum;
-    red[1][walker][threadIdx.x % BN_CB] = sq;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red[walker * C + c8 * 8 + j] = sum[j];
+        red[nw * C + walker * C + c8 * 8 + j] = sq[j];
+    }
     __syncthreads();
-    if (walker == 0 && c < C) {
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
         float ts = 0.f, tq = 0.f;
-        #pragma unroll
-        for (int w = 0; w < BN_RW; ++w) {
-            ts += red[0][w][threadIdx.x % BN_CB];
-            tq += red[1][w][threadIdx.x % BN_CB];
+        for (int w = 0; w < nw; ++w) {
+            ts += red[w * C + c];
+            tq += red[nw * C + w * C + c];
         }
-        slab[((long)s * C + c) * 2 + 0] = ts;
-        slab[((long)s * C + c) * 2 + 1] = tq;
+        slab[(long)c * S + s] = ts;
+        slab[(long)C * S + (long)c * S + s] = tq;
     }
 }
 
-// combine: per-channel stats, scale/shift, running-stat update
+// combine: wave per channel; lanes sweep the S splits coalesced
 __global__ void k_bn_combine(const float* __restrict__ slab, int S, int C,
                              long M, const float* __restrict__ gamma,
                              const float* __restrict__ beta,
@@ -64,15 +68,19 @@ __global__ void k_bn_combine(const float* __restrict__ slab, int S, int C,
                              float* __restrict__ save_invstd,
                              float* __restrict__ scale,
                              float* __restrict__ shift) {
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
     if (c >= C) return;
     float sum = 0.f, sq = 0.f;
-    for (int s = 0; s < S; ++s) {          // fixed order: deterministic
-        sum += slab[((long)s * C + c) * 2 + 0];
-        sq += slab[((long)s * C + c) * 2 + 1];
+    for (int s = lane; s < S; s += 64) {
+        sum += slab[(long)c * S + s];
+        sq += slab[(long)C * S + (long)c * S + s];
     }
+    sum = wave_reduce_sum(sum);
+    sq = wave_reduce_sum(sq);
+    if (lane != 0) return;
     float mean = sum / M;
-    float var = fmaxf(sq / M - mean * mean, 0.f);   // biased (normalization)
+    float var = fmaxf(sq / M - mean * mean, 0.f);
     float invstd = rsqrtf(var + eps);
     save_mean[c] = mean;
     save_invstd[c] = invstd;
@@ -88,7 +96,6 @@ __global__ void k_bn_combine(const float* __restrict__ slab, int S, int C,
     }
 }
 
-// eval-mode scale/shift from running stats
 __global__ void k_bn_eval_coeffs(const float* __restrict__ gamma,
                                  const float* __restrict__ beta,
                                  const float* __restrict__ running_mean,
@@ -106,7 +113,7 @@ __global__ void k_bn_eval_coeffs(const float* __restrict__ gamma,
     shift[c] = b - running_mean[c] * sc;
 }
 
-// apply: y = relu?(x*scale[c] + shift[c]) — vectorized bf16x8, C % 8 == 0 path
+// apply: y = relu?(x*scale[c] + shift[c]) — bf16x8
 __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
                               const float* __restrict__ scale,
                               const float* __restrict__ shift,
@@ -127,76 +134,78 @@ __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
     }
 }
 
-__global__ void k_bn_apply_scalar(const bf16* __restrict__ x, bf16* __restrict__ y,
-                                  const float* __restrict__ scale,
-                                  const float* __restrict__ shift,
-                                  long total, int C, bool relu) {
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total; i += stride) {
-        int c = i % C;
-        float f = fmaf(bf2f(x[i]), scale[c], shift[c]);
-        if (relu) f = fmaxf(f, 0.f);
-        y[i] = f2bf(f);
-    }
-}
-
-// backward partials: per-channel {sum(dy_eff), sum(dy_eff * xhat)}
+// backward partials: slab[0][c][s] = sum(dy_eff), slab[1][c][s] = sum(dy_eff*xhat)
 __global__ void k_bn_bwd_partial(const bf16* __restrict__ x,
                                  const bf16* __restrict__ dy,
-                                 const bf16* __restrict__ y,  // post-relu (mask)
+                                 const bf16* __restrict__ y,  // post-relu mask
                                  const float* __restrict__ save_mean,
                                  const float* __restrict__ save_invstd,
                                  long M, int C, int S, bool relu,
                                  float* __restrict__ slab) {
-    __shared__ float red[2][BN_RW][BN_CB];
-    int cb = blockIdx.x;
-    int s = blockIdx.y;
-    int c = cb * BN_CB + (threadIdx.x % BN_CB);
-    int walker = threadIdx.x / BN_CB;
-    float sum_dy = 0.f, sum_dyx = 0.f;
-    if (c < C) {
-        float mean = save_mean[c], invstd = save_invstd[c];
-        for (long r = s * BN_RW + walker; r < M; r += (long)S * BN_RW) {
-            float g = bf2f(dy[r * C + c]);
-            if (relu && bf2f(y[r * C + c]) <= 0.f) g = 0.f;
-            float xh = (bf2f(x[r * C + c]) - mean) * invstd;
-            sum_dy += g;
-            sum_dyx = fmaf(g, xh, sum_dyx);
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* red = reinterpret_cast<float*>(smem);
+    const int cpg = C >> 3;
+    const int c8 = threadIdx.x % cpg;
+    const int walker = threadIdx.x / cpg;
+    const int nw = blockDim.x / cpg;
+    const int s = blockIdx.x;
+    float mean[8], invstd[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        mean[j] = save_mean[c8 * 8 + j];
+        invstd[j] = save_invstd[c8 * 8 + j];
+    }
+    float sum_dy[8] = {}, sum_dyx[8] = {};
+    for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
+        s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y + r * C)[c8]
+                        : vg;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = us2f((unsigned short)vg[j]);
+            if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
+            float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
+            sum_dy[j] += g;
+            sum_dyx[j] = fmaf(g, xh, sum_dyx[j]);
         }
     }
-    red[0][walker][threadIdx.x % BN_CB] = sum_dy;
-    red[1][walker][threadIdx.x % BN_CB] = sum_dyx;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red[walker * C + c8 * 8 + j] = sum_dy[j];
+        red[nw * C + walker * C + c8 * 8 + j] = sum_dyx[j];
+    }
     __syncthreads();
-    if (walker == 0 && c < C) {
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
         float a = 0.f, b = 0.f;
-        #pragma unroll
-        for (int w = 0; w < BN_RW; ++w) {
-            a += red[0][w][threadIdx.x % BN_CB];
-            b += red[1][w][threadIdx.x % BN_CB];
+        for (int w = 0; w < nw; ++w) {
+            a += red[w * C + c];
+            b += red[nw * C + w * C + c];
         }
-        slab[((long)s * C + c) * 2 + 0] = a;
-        slab[((long)s * C + c) * 2 + 1] = b;
+        slab[(long)c * S + s] = a;
+        slab[(long)C * S + (long)c * S + s] = b;
     }
 }
 
-// combine backward: dgamma, dbeta + dx coefficients
 __global__ void k_bn_bwd_combine(const float* __restrict__ slab, int S, int C,
                                  long M, const float* __restrict__ gamma,
                                  const float* __restrict__ save_invstd,
                                  float* __restrict__ dgamma,
                                  float* __restrict__ dbeta,
-                                 float* __restrict__ coef_a,  // gamma*invstd
-                                 float* __restrict__ coef_b,  // dbeta/M
-                                 float* __restrict__ coef_c)  // sum_dyx/M
-{
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
+                                 float* __restrict__ coef_a,
+                                 float* __restrict__ coef_b,
+                                 float* __restrict__ coef_c) {
+    int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
     if (c >= C) return;
     float sum_dy = 0.f, sum_dyx = 0.f;
-    for (int s = 0; s < S; ++s) {
-        sum_dy += slab[((long)s * C + c) * 2 + 0];
-        sum_dyx += slab[((long)s * C + c) * 2 + 1];
+    for (int s = lane; s < S; s += 64) {
+        sum_dy += slab[(long)c * S + s];
+        sum_dyx += slab[(long)C * S + (long)c * S + s];
     }
+    sum_dy = wave_reduce_sum(sum_dy);
+    sum_dyx = wave_reduce_sum(sum_dyx);
+    if (lane != 0) return;
     dgamma[c] = sum_dyx;
     dbeta[c] = sum_dy;
     float g = gamma ? gamma[c] : 1.f;
@@ -205,7 +214,7 @@ __global__ void k_bn_bwd_combine(const float* __restrict__ slab, int S, int C,
     coef_c[c] = sum_dyx / M;
 }
 
-// dx = a[c] * (dy_eff - b[c] - xhat * c[c])
+// dx = a[c] * (dy_eff - b[c] - xhat * c[c]) — bf16x8
 __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const bf16* __restrict__ dy,
                             const bf16* __restrict__ y,
@@ -215,15 +224,24 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const float* __restrict__ coef_b,
                             const float* __restrict__ coef_c,
                             bf16* __restrict__ dx,
-                            long total, int C, bool relu) {
+                            long total_v, int Cv, bool relu) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total; i += stride) {
-        int c = i % C;
-        float g = bf2f(dy[i]);
-        if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
-        float xh = (bf2f(x[i]) - save_mean[c]) * save_invstd[c];
-        dx[i] = f2bf(coef_a[c] * (g - coef_b[c] - xh * coef_c[c]));
+    for (; i < total_v; i += stride) {
+        int cv = (i % Cv) * 8;
+        s16x8 vx = reinterpret_cast<const s16x8*>(x)[i];
+        s16x8 vg = reinterpret_cast<const s16x8*>(dy)[i];
+        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y)[i] : vg;
+        s16x8 o;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            int c = cv + j;
+            float g = us2f((unsigned short)vg[j]);
+            if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
+            float xh = (us2f((unsigned short)vx[j]) - save_mean[c]) * save_invstd[c];
+            o[j] = (short)f2us(coef_a[c] * (g - coef_b[c] - xh * coef_c[c]));
+        }
+        reinterpret_cast<s16x8*>(dx)[i] = o;
     }
 }
 
@@ -240,16 +258,25 @@ static void shape_mc(const at::Tensor& x, long& M, int& C) {
     }
 }
 
+static int pick_block(int C) {
+    // threads = largest multiple of (C/8) channel-groups <= 256
+    int cpg = C / 8;
+    return (256 / cpg) * cpg;
+}
+
 std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
                                      double eps, bool fuse_relu) {
     long M; int C;
     shape_mc(x, M, C);
+    TORCH_CHECK(C % 8 == 0 && C / 8 <= 256, "bn HIP path needs C%8==0");
     auto stream = at::hip::getCurrentHIPStream();
     auto fopts = gamma.options().dtype(at::kFloat);
-    int S = bn_splits(M, C);
-    auto slab = at::empty({S, C, 2}, fopts);
+    int block = pick_block(C);
+    int nw = block / (C / 8);
+    int S = bn_splits(M, nw);
+    auto slab = at::empty({2, C, S}, fopts);
     auto save_mean = at::empty({C}, fopts);
     auto save_invstd = at::empty({C}, fopts);
     auto scale = at::empty({C}, fopts);
@@ -258,12 +285,12 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
         ? at::empty_like(x, x.options().memory_format(at::MemoryFormat::ChannelsLast))
         : at::empty_like(x);
     const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+    int lds = 2 * nw * C * 4;
 
-    dim3 pgrid(ceil_div_i(C, BN_CB), S);
-    hipLaunchKernelGGL(k_bn_partial, pgrid, dim3(BN_CB * BN_RW), 0,
+    hipLaunchKernelGGL(k_bn_partial, dim3(S), dim3(block), lds,
                        stream.stream(), xp, M, C, S, slab.data_ptr<float>());
     HIP_CHECK_LAST();
-    hipLaunchKernelGGL(k_bn_combine, dim3(ceil_div_i(C, 256)), dim3(256), 0,
+    hipLaunchKernelGGL(k_bn_combine, dim3(ceil_div_i(C, 4)), dim3(256), 0,
                        stream.stream(), slab.data_ptr<float>(), S, C, M,
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
                        running_mean.data_ptr<float>(),
@@ -273,20 +300,12 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                        save_invstd.data_ptr<float>(),
                        scale.data_ptr<float>(), shift.data_ptr<float>());
     HIP_CHECK_LAST();
-    long total = M * C;
+    long tv = M * C / 8;
     bf16* yp = reinterpret_cast<bf16*>(y.data_ptr());
-    if (C % 8 == 0) {
-        long tv = total / 8;
-        int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
-        hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
-                           stream.stream(), xp, yp, scale.data_ptr<float>(),
-                           shift.data_ptr<float>(), tv, C / 8, fuse_relu);
-    } else {
-        int blocks = std::min<long>(4096, ceil_div_i(total, 256));
-        hipLaunchKernelGGL(k_bn_apply_scalar, dim3(blocks), dim3(256), 0,
-                           stream.stream(), xp, yp, scale.data_ptr<float>(),
-                           shift.data_ptr<float>(), total, C, fuse_relu);
-    }
+    int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
+    hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
+                       stream.stream(), xp, yp, scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), tv, C / 8, fuse_relu);
     HIP_CHECK_LAST();
     return {y, save_mean, save_invstd};
 }
@@ -296,6 +315,7 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        double eps, bool fuse_relu) {
     long M; int C;
     shape_mc(x, M, C);
+    TORCH_CHECK(C % 8 == 0);
     auto stream = at::hip::getCurrentHIPStream();
     auto fopts = gamma.options().dtype(at::kFloat);
     auto scale = at::empty({C}, fopts);
@@ -309,21 +329,14 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        running_var.data_ptr<float>(), (float)eps, C,
                        scale.data_ptr<float>(), shift.data_ptr<float>());
     HIP_CHECK_LAST();
-    const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
-    bf16* yp = reinterpret_cast<bf16*>(y.data_ptr());
-    long total = M * C;
-    if (C % 8 == 0) {
-        long tv = total / 8;
-        int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
-        hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
-                           stream.stream(), xp, yp, scale.data_ptr<float>(),
-                           shift.data_ptr<float>(), tv, C / 8, fuse_relu);
-    } else {
-        int blocks = std::min<long>(4096, ceil_div_i(total, 256));
-        hipLaunchKernelGGL(k_bn_apply_scalar, dim3(blocks), dim3(256), 0,
-                           stream.stream(), xp, yp, scale.data_ptr<float>(),
-                           shift.data_ptr<float>(), total, C, fuse_relu);
-    }
+    long tv = M * C / 8;
+    int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
+    hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(), tv,
+                       C / 8, fuse_relu);
     HIP_CHECK_LAST();
     return y;
 }
@@ -333,10 +346,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor y, bool fuse_relu) {
     long M; int C;
     shape_mc(x, M, C);
+    TORCH_CHECK(C % 8 == 0 && C / 8 <= 256);
     auto stream = at::hip::getCurrentHIPStream();
     auto fopts = gamma.options().dtype(at::kFloat);
-    int S = bn_splits(M, C);
-    auto slab = at::empty({S, C, 2}, fopts);
+    int block = pick_block(C);
+    int nw = block / (C / 8);
+    int S = bn_splits(M, nw);
+    auto slab = at::empty({2, C, S}, fopts);
     auto dgamma = at::empty({C}, fopts);
     auto dbeta = at::empty({C}, fopts);
     auto ca = at::empty({C}, fopts);
@@ -348,28 +364,28 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
     const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
     const bf16* dyp = reinterpret_cast<const bf16*>(dy.data_ptr());
     const bf16* yp = reinterpret_cast<const bf16*>(y.data_ptr());
+    int lds = 2 * nw * C * 4;
 
-    dim3 pgrid(ceil_div_i(C, BN_CB), S);
-    hipLaunchKernelGGL(k_bn_bwd_partial, pgrid, dim3(BN_CB * BN_RW), 0,
+    hipLaunchKernelGGL(k_bn_bwd_partial, dim3(S), dim3(block), lds,
                        stream.stream(), xp, dyp, yp,
                        save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>(), M, C, S, fuse_relu,
                        slab.data_ptr<float>());
     HIP_CHECK_LAST();
-    hipLaunchKernelGGL(k_bn_bwd_combine, dim3(ceil_div_i(C, 256)), dim3(256), 0,
+    hipLaunchKernelGGL(k_bn_bwd_combine, dim3(ceil_div_i(C, 4)), dim3(256), 0,
                        stream.stream(), slab.data_ptr<float>(), S, C, M,
                        gamma.data_ptr<float>(), save_invstd.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                        ca.data_ptr<float>(), cb.data_ptr<float>(),
                        cc.data_ptr<float>());
     HIP_CHECK_LAST();
-    long total = M * C;
-    int blocks = std::min<long>(4096, ceil_div_i(total, 256));
+    long tv = M * C / 8;
+    int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
     hipLaunchKernelGGL(k_bn_bwd_dx, dim3(blocks), dim3(256), 0, stream.stream(),
                        xp, dyp, yp, save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>(), ca.data_ptr<float>(),
                        cb.data_ptr<float>(), cc.data_ptr<float>(),
-                       reinterpret_cast<bf16*>(dx.data_ptr()), total, C,
+                       reinterpret_cast<bf16*>(dx.data_ptr()), tv, C / 8,
                        fuse_relu);
     HIP_CHECK_LAST();
     return {dx, dgamma, dbeta};

@@ -349,6 +349,45 @@ __global__ void k_wgrad_combine(const float* __restrict__ slab, int S,
     }
 }
 
+// conv1-style wgrad fast path (RSC <= 16, Ko <= 64): ONE pass over dy per
+// split — all RSC taps accumulate in registers; dy loads are lane-coalesced
+// along Ko, x loads are wave-uniform broadcasts.
+__global__ void k_wgrad_small_rsc(const bf16* __restrict__ dy,
+                                  const bf16* __restrict__ x,
+                                  float* __restrict__ slab, ConvShape cs,
+                                  long M, int S, int RSC) {
+    int ko = threadIdx.x & 63;
+    int walker = threadIdx.x >> 6;    // 4 walkers
+    int split = blockIdx.x;
+    float acc[16] = {};
+    for (long gm = (long)split * 4 + walker; gm < M; gm += (long)S * 4) {
+        int q = gm % cs.Q; long rem = gm / cs.Q;
+        int p = rem % cs.P; int n = rem / cs.P;
+        float g = (ko < cs.Ko) ? bf2f(dy[gm * cs.Ko + ko]) : 0.f;
+        #pragma unroll 1
+        for (int k = 0; k < RSC; ++k) {
+            int c = k % cs.C; int rs = k / cs.C;
+            int r = rs / cs.S, s = rs % cs.S;
+            int h = p * cs.stride + r - cs.pad;
+            int wcol = q * cs.stride + s - cs.pad;
+            float xv = (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                ? bf2f(x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c])
+                : 0.f;
+            acc[k] = fmaf(g, xv, acc[k]);
+        }
+    }
+    __shared__ float red[4][64];
+    for (int k = 0; k < RSC; ++k) {
+        __syncthreads();
+        red[walker][ko] = acc[k];
+        __syncthreads();
+        if (walker == 0 && ko < cs.Ko) {
+            float t = red[0][ko] + red[1][ko] + red[2][ko] + red[3][ko];
+            slab[((long)split * RSC + k) * cs.Ko + ko] = t;
+        }
+    }
+}
+
 // conv1-style wgrad (Cin < 8): per-(rs, split) block, lanes along Ko
 __global__ void k_wgrad_small_cin(const bf16* __restrict__ dy,
                                   const bf16* __restrict__ x,
@@ -483,6 +522,22 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
 
     if (cs.C < 8) {
         int rsc_total = Kgemm;
+        if (rsc_total <= 16 && cs.Ko <= 64) {
+            int S_ = 512;
+            auto slab = at::empty({S_, rsc_total, cs.Ko},
+                                  x.options().dtype(at::kFloat));
+            hipLaunchKernelGGL(k_wgrad_small_rsc, dim3(S_), dim3(256), 0,
+                               stream.stream(), dyp, xp,
+                               slab.data_ptr<float>(), cs, M, S_, rsc_total);
+            HIP_CHECK_LAST();
+            int total = rsc_total * cs.Ko;
+            hipLaunchKernelGGL(k_wgrad_small_combine,
+                               dim3(ceil_div_i(total, 256)), dim3(256), 0,
+                               stream.stream(), slab.data_ptr<float>(), S_,
+                               cs, rsc_total, dw.data_ptr<float>());
+            HIP_CHECK_LAST();
+            return dw;
+        }
         int S_ = 64;
         auto slab = at::empty({S_, rsc_total, cs.Ko},
                               x.options().dtype(at::kFloat));

@@ -7,8 +7,10 @@
 // whole parameter list in one launch, bf16<->fp32 casts.
 //
 // Multi-tensor scheme: host packs up to MT_MAX tensor pointers + a prefix
-// of chunk offsets into the kernarg struct; blocks binary-map to
-// (tensor, chunk) so one launch covers the whole list.
+// of chunk offsets into the kernarg struct; blocks map to (tensor, chunk)
+// so one launch covers the whole list.  16 Ki-element chunks give ≳350
+// blocks on the 23 MB flagship parameter set (256-CU chip wants ≫256
+// workgroups); bodies are f32x4-vectorized.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -17,7 +19,7 @@
 #include <vector>
 
 constexpr int MT_MAX = 32;
-constexpr int MT_CHUNK = 1 << 16;  // elements per block-chunk
+constexpr int MT_CHUNK = 1 << 14;  // elements per block-chunk
 constexpr int MT_BLOCK = 256;
 
 struct MTMeta {
@@ -30,24 +32,38 @@ struct MTMeta {
 };
 
 DEV_INLINE int find_tensor(const MTMeta& m, int chunk, int& local_chunk) {
-    // linear scan (ntensors <= 32): cheap relative to the memory work
     int t = 0;
     while (t + 1 < m.ntensors && m.chunk_start[t + 1] <= chunk) ++t;
     local_chunk = chunk - m.chunk_start[t];
     return t;
 }
 
+DEV_INLINE void chunk_range(const MTMeta& m, int& t, long& base, long& end) {
+    int local;
+    t = find_tensor(m, blockIdx.x, local);
+    long n = m.sizes[t];
+    base = (long)local * MT_CHUNK;
+    end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
+}
+
 // ---------------------------------------------------------------- unscale ---
 
 __global__ void k_mt_unscale(MTMeta m, float inv_scale, float* found_inf) {
-    int local;
-    int t = find_tensor(m, blockIdx.x, local);
+    int t; long base, end;
+    chunk_range(m, t, base, end);
     float* g = m.b[t];
-    long n = m.sizes[t];
-    long base = (long)local * MT_CHUNK;
-    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
+    long vb = base >> 2, ve = end >> 2;
     bool bad = false;
-    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+    for (long i = vb + threadIdx.x; i < ve; i += MT_BLOCK) {
+        f32x4 v = reinterpret_cast<f32x4*>(g)[i];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            v[j] *= inv_scale;
+            if (!isfinite(v[j])) bad = true;
+        }
+        reinterpret_cast<f32x4*>(g)[i] = v;
+    }
+    for (long i = ve * 4 + threadIdx.x; i < end; i += MT_BLOCK) {
         float v = g[i] * inv_scale;
         if (!isfinite(v)) bad = true;
         g[i] = v;
@@ -85,20 +101,37 @@ void multi_tensor_unscale(std::vector<at::Tensor> grads, at::Tensor found_inf,
 // -------------------------------------------------------------- fused SGD ---
 
 // buf = mu*buf + g ; d = nesterov ? g + mu*buf : buf ; p -= lr*d
-// If found_inf != nullptr and *found_inf != 0, the step is a device-side
-// no-op (amp overflow skip without a host sync).
+// With found_inf set, the step is a device-side no-op (amp overflow skip
+// without a host sync).
 __global__ void k_mt_sgd(MTMeta m, float lr, float mu, float wd, float nesterov,
                          const float* __restrict__ found_inf) {
     if (found_inf && *found_inf != 0.0f) return;
-    int local;
-    int t = find_tensor(m, blockIdx.x, local);
+    int t; long base, end;
+    chunk_range(m, t, base, end);
     const float* __restrict__ g = m.a[t];
     float* __restrict__ p = m.b[t];
     float* __restrict__ buf = m.c[t];
-    long n = m.sizes[t];
-    long base = (long)local * MT_CHUNK;
-    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
-    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+    long vb = base >> 2, ve = end >> 2;
+    for (long i = vb + threadIdx.x; i < ve; i += MT_BLOCK) {
+        f32x4 gv = reinterpret_cast<const f32x4*>(g)[i];
+        f32x4 pv = reinterpret_cast<f32x4*>(p)[i];
+        f32x4 bv = mu != 0.f ? reinterpret_cast<f32x4*>(buf)[i] : f32x4{};
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            float gi = gv[j];
+            if (wd != 0.0f) gi = fmaf(wd, pv[j], gi);
+            float d = gi;
+            if (mu != 0.0f) {
+                float b = fmaf(mu, bv[j], gi);
+                bv[j] = b;
+                d = (nesterov != 0.0f) ? fmaf(mu, b, gi) : b;
+            }
+            pv[j] = fmaf(-lr, d, pv[j]);
+        }
+        reinterpret_cast<f32x4*>(p)[i] = pv;
+        if (mu != 0.f) reinterpret_cast<f32x4*>(buf)[i] = bv;
+    }
+    for (long i = ve * 4 + threadIdx.x; i < end; i += MT_BLOCK) {
         float gi = g[i];
         if (wd != 0.0f) gi = fmaf(wd, p[i], gi);
         float d = gi;
@@ -149,14 +182,24 @@ void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 __global__ void k_mt_lookahead(MTMeta m, float alpha,
                                const float* __restrict__ found_inf) {
     if (found_inf && *found_inf != 0.0f) return;
-    int local;
-    int t = find_tensor(m, blockIdx.x, local);
+    int t; long base, end;
+    chunk_range(m, t, base, end);
     float* __restrict__ fast = m.b[t];
     float* __restrict__ slow = m.c[t];
-    long n = m.sizes[t];
-    long base = (long)local * MT_CHUNK;
-    long end = base + MT_CHUNK < n ? base + MT_CHUNK : n;
-    for (long i = base + threadIdx.x; i < end; i += MT_BLOCK) {
+    long vb = base >> 2, ve = end >> 2;
+    for (long i = vb + threadIdx.x; i < ve; i += MT_BLOCK) {
+        f32x4 fv = reinterpret_cast<f32x4*>(fast)[i];
+        f32x4 sv = reinterpret_cast<f32x4*>(slow)[i];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            float s = fmaf(alpha, fv[j] - sv[j], sv[j]);
+            sv[j] = s;
+            fv[j] = s;
+        }
+        reinterpret_cast<f32x4*>(fast)[i] = fv;
+        reinterpret_cast<f32x4*>(slow)[i] = sv;
+    }
+    for (long i = ve * 4 + threadIdx.x; i < end; i += MT_BLOCK) {
         float s = fmaf(alpha, fast[i] - slow[i], slow[i]);
         slow[i] = s;
         fast[i] = s;
@@ -195,16 +238,14 @@ __global__ void k_f32_to_bf16(const float* __restrict__ in,
                               unsigned short* __restrict__ out, long n) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    for (; i < n / 4; i += stride) {
-        f32x4 v = reinterpret_cast<const f32x4*>(in)[i];
+    long nv = n >> 2;
+    for (long v = i; v < nv; v += stride) {
+        f32x4 f = reinterpret_cast<const f32x4*>(in)[v];
         s16x4 o;
-        o[0] = f2us(v[0]); o[1] = f2us(v[1]); o[2] = f2us(v[2]); o[3] = f2us(v[3]);
-        reinterpret_cast<s16x4*>(out)[i] = o;
+        o[0] = f2us(f[0]); o[1] = f2us(f[1]); o[2] = f2us(f[2]); o[3] = f2us(f[3]);
+        reinterpret_cast<s16x4*>(out)[v] = o;
     }
-    // tail
-    long tail = n & ~3L;
-    for (long j = tail + (blockIdx.x * blockDim.x + threadIdx.x);
-         j < n; j += stride)
+    for (long j = nv * 4 + i; j < n; j += stride)
         out[j] = f2us(in[j]);
 }
 
@@ -212,7 +253,7 @@ at::Tensor cast_to_bf16(at::Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat && x.is_contiguous());
     auto y = at::empty_like(x, x.options().dtype(at::kBFloat16));
     long n = x.numel();
-    int blocks = std::min<long>(2048, (n / 4 + 255) / 256 + 1);
+    int blocks = (int)std::min<long>(2048, (n / 4 + 255) / 256 + 1);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(k_f32_to_bf16, dim3(blocks), dim3(256), 0, stream.stream(),
                        x.data_ptr<float>(),

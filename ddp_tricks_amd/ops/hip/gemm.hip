@@ -248,35 +248,66 @@ __global__ void k_transpose_bf16(const bf16* __restrict__ in,
 
 // -------------------------------------------------------------- column sum ---
 
-// colsum[c] = sum_m in[m][c] (bf16 in, fp32 out), deterministic slab scheme
+// colsum[c] = sum_m in[m][c] (bf16 in, fp32 out), deterministic slab
+// scheme, bf16x8-vectorized; slab layout [C][S] so combine waves read each
+// channel's splits coalesced.
 __global__ void k_colsum_partial(const bf16* __restrict__ x, long M, int C,
                                  int S, float* __restrict__ slab) {
-    int cb = blockIdx.x * 64 + (threadIdx.x % 64);
-    int s = blockIdx.y;
-    int walker = threadIdx.x / 64;
-    __shared__ float red[4][64];
-    float sum = 0.f;
-    if (cb < C) {
-        for (long r = s * 4 + walker; r < M; r += (long)S * 4)
-            sum += bf2f(x[r * C + cb]);
-    }
-    red[walker][threadIdx.x % 64] = sum;
-    __syncthreads();
-    if (walker == 0 && cb < C) {
-        float t = 0.f;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* red = reinterpret_cast<float*>(smem);   // [nw][C]
+    const int cpg = C >> 3;
+    const int c8 = threadIdx.x % cpg;
+    const int walker = threadIdx.x / cpg;
+    const int nw = blockDim.x / cpg;
+    const int s = blockIdx.x;
+    float acc[8] = {};
+    for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
+        s16x8 v = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         #pragma unroll
-        for (int w = 0; w < 4; ++w) t += red[w][threadIdx.x % 64];
-        slab[(long)s * C + cb] = t;
+        for (int j = 0; j < 8; ++j) acc[j] += us2f((unsigned short)v[j]);
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) red[walker * C + c8 * 8 + j] = acc[j];
+    __syncthreads();
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float t = 0.f;
+        for (int w = 0; w < nw; ++w) t += red[w * C + c];
+        slab[(long)c * S + s] = t;
+    }
+}
+
+// scalar fallback for C % 8 != 0 (tiny tensors, e.g. the 10-class head)
+__global__ void k_colsum_partial_scalar(const bf16* __restrict__ x, long M,
+                                        int C, int S, float* __restrict__ slab) {
+    int c = threadIdx.x % C;
+    int walker = threadIdx.x / C;
+    int nw = blockDim.x / C;
+    int s = blockIdx.x;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* red = reinterpret_cast<float*>(smem);   // [nw][C]
+    float sum = 0.f;
+    if (walker < nw) {
+        for (long r = (long)s * nw + walker; r < M; r += (long)S * nw)
+            sum += bf2f(x[r * C + c]);
+        red[walker * C + c] = sum;
+    }
+    __syncthreads();
+    for (int cc = threadIdx.x; cc < C; cc += blockDim.x) {
+        float t = 0.f;
+        for (int w = 0; w < nw; ++w) t += red[w * C + cc];
+        slab[(long)cc * S + s] = t;
     }
 }
 
 __global__ void k_colsum_combine(const float* __restrict__ slab, int S, int C,
                                  float* __restrict__ out) {
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
     if (c >= C) return;
     float v = 0.f;
-    for (int s = 0; s < S; ++s) v += slab[(long)s * C + c];
-    out[c] = v;
+    for (int s = lane; s < S; s += 64) v += slab[(long)c * S + s];
+    v = wave_reduce_sum(v);
+    if (lane == 0) out[c] = v;
 }
 
 // ------------------------------------------------------------------ hosts ---
@@ -363,17 +394,37 @@ at::Tensor col_sum(at::Tensor x) {
                 x.is_contiguous());
     long M = x.size(0);
     int C = x.size(1);
-    int S = (int)std::max<long>(1, std::min<long>(64, M / 4));
-    auto slab = at::empty({S, C}, x.options().dtype(at::kFloat));
-    auto out = at::empty({C}, x.options().dtype(at::kFloat));
     auto stream = at::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(k_colsum_partial, dim3(ceil_div_i(C, 64), S), dim3(256),
-                       0, stream.stream(),
+    auto out = at::empty({C}, x.options().dtype(at::kFloat));
+    if (C % 8 == 0 && C / 8 <= 256) {
+        int cpg = C / 8;
+        int block = (256 / cpg) * cpg;
+        int nw = block / cpg;
+        int S = (int)std::max<long>(1, std::min<long>(1024, M / (nw * 4)));
+        auto slab = at::empty({C, S}, x.options().dtype(at::kFloat));
+        hipLaunchKernelGGL(k_colsum_partial, dim3(S), dim3(block),
+                           nw * C * 4, stream.stream(),
+                           reinterpret_cast<const bf16*>(x.data_ptr()), M, C,
+                           S, slab.data_ptr<float>());
+        HIP_CHECK_LAST();
+        hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(C, 4)), dim3(256),
+                           0, stream.stream(), slab.data_ptr<float>(), S, C,
+                           out.data_ptr<float>());
+        HIP_CHECK_LAST();
+        return out;
+    }
+    TORCH_CHECK(C <= 256, "col_sum scalar path needs C<=256");
+    int nw = 256 / C;
+    int block = nw * C;
+    int S = (int)std::max<long>(1, std::min<long>(512, M / (nw * 4)));
+    auto slab = at::empty({C, S}, x.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(k_colsum_partial_scalar, dim3(S), dim3(block),
+                       nw * C * 4, stream.stream(),
                        reinterpret_cast<const bf16*>(x.data_ptr()), M, C, S,
                        slab.data_ptr<float>());
     HIP_CHECK_LAST();
-    hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(C, 256)), dim3(256),
-                       0, stream.stream(), slab.data_ptr<float>(), S, C,
+    hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(C, 4)), dim3(256), 0,
+                       stream.stream(), slab.data_ptr<float>(), S, C,
                        out.data_ptr<float>());
     HIP_CHECK_LAST();
     return out;
