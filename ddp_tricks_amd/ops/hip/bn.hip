@@ -214,7 +214,9 @@ __global__ void k_bn_bwd_combine(const float* __restrict__ slab, int S, int C,
     coef_c[c] = sum_dyx / M;
 }
 
-// dx = a[c] * (dy_eff - b[c] - xhat * c[c]) — bf16x8
+// dx = a[c] * (dy_eff - b[c] - xhat * c[c]) — bf16x8.  Threads own a fixed
+// 8-channel group (like the partial pass) so the five per-channel
+// coefficients live in registers, not per-element loads.
 __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const bf16* __restrict__ dy,
                             const bf16* __restrict__ y,
@@ -224,24 +226,36 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const float* __restrict__ coef_b,
                             const float* __restrict__ coef_c,
                             bf16* __restrict__ dx,
-                            long total_v, int Cv, bool relu) {
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total_v; i += stride) {
-        int cv = (i % Cv) * 8;
-        s16x8 vx = reinterpret_cast<const s16x8*>(x)[i];
-        s16x8 vg = reinterpret_cast<const s16x8*>(dy)[i];
-        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y)[i] : vg;
+                            long M, int C, bool relu) {
+    const int cpg = C >> 3;
+    const int c8 = threadIdx.x % cpg;
+    const int walker = threadIdx.x / cpg;
+    const int nw = blockDim.x / cpg;
+    float mean[8], invstd[8], ca[8], cb[8], cc[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        int c = c8 * 8 + j;
+        mean[j] = save_mean[c];
+        invstd[j] = save_invstd[c];
+        ca[j] = coef_a[c];
+        cb[j] = coef_b[c];
+        cc[j] = coef_c[c];
+    }
+    long r0 = (long)blockIdx.x * nw + walker;
+    long rstride = (long)gridDim.x * nw;
+    for (long r = r0; r < M; r += rstride) {
+        s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y + r * C)[c8] : vg;
         s16x8 o;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            int c = cv + j;
             float g = us2f((unsigned short)vg[j]);
             if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
-            float xh = (us2f((unsigned short)vx[j]) - save_mean[c]) * save_invstd[c];
-            o[j] = (short)f2us(coef_a[c] * (g - coef_b[c] - xh * coef_c[c]));
+            float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
+            o[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
         }
-        reinterpret_cast<s16x8*>(dx)[i] = o;
+        reinterpret_cast<s16x8*>(dx + r * C)[c8] = o;
     }
 }
 
@@ -379,13 +393,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                        ca.data_ptr<float>(), cb.data_ptr<float>(),
                        cc.data_ptr<float>());
     HIP_CHECK_LAST();
-    long tv = M * C / 8;
-    int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
-    hipLaunchKernelGGL(k_bn_bwd_dx, dim3(blocks), dim3(256), 0, stream.stream(),
+    int dxblocks = (int)std::max<long>(1, std::min<long>(2048, M / (nw * 2)));
+    hipLaunchKernelGGL(k_bn_bwd_dx, dim3(dxblocks), dim3(block), 0,
+                       stream.stream(),
                        xp, dyp, yp, save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>(), ca.data_ptr<float>(),
                        cb.data_ptr<float>(), cc.data_ptr<float>(),
-                       reinterpret_cast<bf16*>(dx.data_ptr()), tv, C / 8,
+                       reinterpret_cast<bf16*>(dx.data_ptr()), M, C,
                        fuse_relu);
     HIP_CHECK_LAST();
     return {dx, dgamma, dbeta};

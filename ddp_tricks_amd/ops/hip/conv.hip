@@ -29,21 +29,31 @@ struct ConvShape {
 
 // ---------------------------------------------------------------- forward ---
 
-// A-gather: element (gm, gk): gm -> (n,p,q); gk -> (rs, c)
-// addr = x[((n*H + p*st + r - pad)*W + q*st + s - pad)*C + c]
+// Unified implicit-GEMM kernel.  MODE 0 = forward (A gathered from x with
+// k = (r,s,c)); MODE 1 = backward-data (A gathered from dy with
+// k = (r,s,ko), B = WT2).  TBN selects the output-channel tile width
+// (128, or 64 for narrow outputs like dgrad into C=64); the 256-thread
+// block is WAVES_M x WAVES_N waves each owning a (128/WAVES_M) x
+// (TBN/WAVES_N) sub-tile.
+template <int MODE, int TBN, int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(256)
-void k_conv_fwd(const bf16* __restrict__ x, const bf16* __restrict__ w,
-                const float* __restrict__ bias, bf16* __restrict__ y,
-                ConvShape cs, int M, int Kgemm) {
+void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
+                 const float* __restrict__ bias, bf16* __restrict__ out,
+                 ConvShape cs, int M, int Kgemm, int Nout) {
+    constexpr int WM = CBM / WAVES_M;
+    constexpr int WN = TBN / WAVES_N;
+    constexpr int MI = WM / 16;
+    constexpr int NI = WN / 16;
     __shared__ bf16 lds_a[CBM][CLDK];
-    __shared__ bf16 lds_b[CBN][CLDK];
+    __shared__ bf16 lds_b[TBN][CLDK];
     const int m0 = blockIdx.x * CBM;
-    const int n0 = blockIdx.y * CBN;
+    const int n0 = blockIdx.y * TBN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
-    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+    const int wid = tid >> 6;
+    const int wr = wid / WAVES_N, wc = wid % WAVES_N;
 
-    f32x4 acc[4][4] = {};
+    f32x4 acc[MI][NI] = {};
     const int ld_row = tid >> 3;
     const int ld_col = (tid & 7) * 8;
 
@@ -54,43 +64,56 @@ void k_conv_fwd(const bf16* __restrict__ x, const bf16* __restrict__ w,
             int gm = m0 + row, gk = kt + ld_col;
             bf16x8_t va = {};
             if (gm < M && gk < Kgemm) {
-                int q = gm % cs.Q, rem = gm / cs.Q;
-                int p = rem % cs.P, n = rem / cs.P;
-                int c = gk % cs.C, rs = gk / cs.C;
-                int r = rs / cs.S, s = rs % cs.S;
-                int h = p * cs.stride + r - cs.pad;
-                int wcol = q * cs.stride + s - cs.pad;
-                if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W) {
-                    va = *reinterpret_cast<const bf16x8_t*>(
-                        &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                if (MODE == 0) {
+                    int q = gm % cs.Q, rem = gm / cs.Q;
+                    int p = rem % cs.P, n = rem / cs.P;
+                    int c = gk % cs.C, rs = gk / cs.C;
+                    int r = rs / cs.S, s = rs % cs.S;
+                    int h = p * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + s - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        va = *reinterpret_cast<const bf16x8_t*>(
+                            &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                } else {
+                    int wcol = gm % cs.W; long rem = gm / cs.W;
+                    int h = rem % cs.H; int n = rem / cs.H;
+                    int ko = gk % cs.Ko, rs = gk / cs.Ko;
+                    int r = rs / cs.S, s = rs % cs.S;
+                    int p = h + cs.pad - r;
+                    int q = wcol + cs.pad - s;
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        va = *reinterpret_cast<const bf16x8_t*>(
+                            &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko]);
                 }
             }
             *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
 
-            int gn = n0 + row;
-            bf16x8_t vb = {};
-            if (gn < cs.Ko && gk < Kgemm)
-                vb = *reinterpret_cast<const bf16x8_t*>(
-                    &w[(long)gn * Kgemm + gk]);
-            *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
+            if (row < TBN) {
+                int gn = n0 + row;
+                bf16x8_t vb = {};
+                if (gn < Nout && gk < Kgemm)
+                    vb = *reinterpret_cast<const bf16x8_t*>(
+                        &Bsrc[(long)gn * Kgemm + gk]);
+                *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
+            }
         }
         __syncthreads();
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
-            bf16x8_t af[4], bfr[4];
+            bf16x8_t af[MI], bfr[NI];
             const int kcol = ks + (lane >> 4) * 8;
             #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
                 af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+                    &lds_a[wr * WM + mi * 16 + (lane & 15)][kcol]);
             #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < NI; ++ni)
                 bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+                    &lds_b[wc * WN + ni * 16 + (lane & 15)][kcol]);
             #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
                 #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
+                for (int ni = 0; ni < NI; ++ni)
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
         }
@@ -98,17 +121,17 @@ void k_conv_fwd(const bf16* __restrict__ x, const bf16* __restrict__ w,
     }
 
     #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+    for (int mi = 0; mi < MI; ++mi)
         #pragma unroll
-        for (int ni = 0; ni < 4; ++ni) {
-            int col = n0 + wc * 64 + ni * 16 + (lane & 15);
-            if (col >= cs.Ko) continue;
+        for (int ni = 0; ni < NI; ++ni) {
+            int col = n0 + wc * WN + ni * 16 + (lane & 15);
+            if (col >= Nout) continue;
             float badd = bias ? bias[col] : 0.f;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                int row = m0 + wr * WM + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= M) continue;
-                y[(long)row * cs.Ko + col] = f2bf(acc[mi][ni][r] + badd);
+                out[(long)row * Nout + col] = f2bf(acc[mi][ni][r] + badd);
             }
         }
 }
@@ -155,90 +178,6 @@ __global__ void k_conv_small_cin(const bf16* __restrict__ x,
             o[j] = (short)f2us(acc[j] + (bias ? bias[ko8 + j] : 0.f));
         *reinterpret_cast<bf16x8_t*>(&y[gm * cs.Ko + ko8]) = o;
     }
-}
-
-// ---------------------------------------------------------- backward data ---
-
-// dx[M=N·H·W, C] = Σ_{r,s,ko} dy[n, (h+pad-r)/st, (w+pad-s)/st, ko] ·
-//                  WT2[c, (r·S+s)·Ko+ko]        (stride 1 only for now)
-__global__ __launch_bounds__(256)
-void k_conv_dgrad(const bf16* __restrict__ dy, const bf16* __restrict__ wt2,
-                  bf16* __restrict__ dx, ConvShape cs, int M, int Kgemm) {
-    __shared__ bf16 lds_a[CBM][CLDK];
-    __shared__ bf16 lds_b[CBN][CLDK];
-    const int m0 = blockIdx.x * CBM;
-    const int n0 = blockIdx.y * CBN;
-    const int tid = threadIdx.x;
-    const int lane = tid & 63;
-    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
-
-    f32x4 acc[4][4] = {};
-    const int ld_row = tid >> 3;
-    const int ld_col = (tid & 7) * 8;
-
-    for (int kt = 0; kt < Kgemm; kt += CBK) {
-        #pragma unroll
-        for (int p4 = 0; p4 < 4; ++p4) {
-            int row = p4 * 32 + ld_row;
-            int gm = m0 + row, gk = kt + ld_col;
-            bf16x8_t va = {};
-            if (gm < M && gk < Kgemm) {
-                int wcol = gm % cs.W; long rem = gm / cs.W;
-                int h = rem % cs.H; int n = rem / cs.H;
-                int ko = gk % cs.Ko, rs = gk / cs.Ko;
-                int r = rs / cs.S, s = rs % cs.S;
-                int p = h + cs.pad - r;
-                int q = wcol + cs.pad - s;
-                if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q) {
-                    va = *reinterpret_cast<const bf16x8_t*>(
-                        &dy[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko]);
-                }
-            }
-            *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
-
-            int gn = n0 + row;
-            bf16x8_t vb = {};
-            if (gn < cs.C && gk < Kgemm)
-                vb = *reinterpret_cast<const bf16x8_t*>(
-                    &wt2[(long)gn * Kgemm + gk]);
-            *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
-        }
-        __syncthreads();
-        #pragma unroll
-        for (int ks = 0; ks < CBK; ks += 32) {
-            bf16x8_t af[4], bfr[4];
-            const int kcol = ks + (lane >> 4) * 8;
-            #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-                af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
-            #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
-                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
-            #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-                #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
-                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
-        }
-        __syncthreads();
-    }
-
-    #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni) {
-            int col = n0 + wc * 64 + ni * 16 + (lane & 15);
-            if (col >= cs.C) continue;
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
-                if (row >= M) continue;
-                dx[(long)row * cs.C + col] = f2bf(acc[mi][ni][r]);
-            }
-        }
 }
 
 // -------------------------------------------------------- backward weight ---
@@ -383,9 +322,27 @@ __global__ void k_wgrad_small_rsc(const bf16* __restrict__ dy,
         __syncthreads();
         if (walker == 0 && ko < cs.Ko) {
             float t = red[0][ko] + red[1][ko] + red[2][ko] + red[3][ko];
-            slab[((long)split * RSC + k) * cs.Ko + ko] = t;
+            // slab layout [RSC*Ko][S]: combine waves sweep the splits coalesced
+            slab[((long)k * cs.Ko + ko) * gridDim.x + split] = t;
         }
     }
+}
+
+// combine for the fast small-RSC path: wave per (k, ko), lanes sweep S
+__global__ void k_wgrad_small_rsc_combine(const float* __restrict__ slab,
+                                          int S, ConvShape cs, int RSC,
+                                          float* __restrict__ dw) {
+    int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (i >= RSC * cs.Ko) return;
+    float v = 0.f;
+    for (int s = lane; s < S; s += 64) v += slab[(long)i * S + s];
+    v = wave_reduce_sum(v);
+    if (lane) return;
+    int ko = i % cs.Ko;
+    int k = i / cs.Ko;
+    int c = k % cs.C, rs = k / cs.C;
+    dw[((long)ko * cs.C + c) * (cs.R * cs.S) + rs] = v;
 }
 
 // conv1-style wgrad (Cin < 8): per-(rs, split) block, lanes along Ko
@@ -475,9 +432,17 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
                            stream.stream(), xp, wp, bp, yp, cs, M);
     } else {
         TORCH_CHECK(cs.C % 8 == 0, "conv fwd needs C % 8 == 0");
-        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, CBN));
-        hipLaunchKernelGGL(k_conv_fwd, grid, dim3(256), 0, stream.stream(),
-                           xp, wp, bp, yp, cs, (int)M, Kgemm);
+        if (cs.Ko >= 128) {
+            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 128));
+            hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid, dim3(256), 0,
+                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                               Kgemm, cs.Ko);
+        } else {
+            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
+            hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
+                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                               Kgemm, cs.Ko);
+        }
     }
     HIP_CHECK_LAST();
     return y;
@@ -497,12 +462,20 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     auto dx = at::empty({(long)cs.N, (long)cs.C, (long)cs.H, (long)cs.W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
     auto stream = at::hip::getCurrentHIPStream();
-    dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, CBN));
-    hipLaunchKernelGGL(k_conv_dgrad, grid, dim3(256), 0, stream.stream(),
-                       reinterpret_cast<const bf16*>(dy.data_ptr()),
-                       reinterpret_cast<const bf16*>(wt2.data_ptr()),
-                       reinterpret_cast<bf16*>(dx.data_ptr()), cs, (int)M,
-                       Kgemm);
+    const bf16* dyp_ = reinterpret_cast<const bf16*>(dy.data_ptr());
+    const bf16* wt2p = reinterpret_cast<const bf16*>(wt2.data_ptr());
+    bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
+    if (cs.C >= 128) {
+        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
+        hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2>), grid, dim3(256), 0,
+                           stream.stream(), dyp_, wt2p, nullptr, dxp, cs,
+                           (int)M, Kgemm, cs.C);
+    } else {
+        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
+        hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1>), grid, dim3(256), 0,
+                           stream.stream(), dyp_, wt2p, nullptr, dxp, cs,
+                           (int)M, Kgemm, cs.C);
+    }
     HIP_CHECK_LAST();
     return dx;
 }
@@ -523,16 +496,16 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     if (cs.C < 8) {
         int rsc_total = Kgemm;
         if (rsc_total <= 16 && cs.Ko <= 64) {
-            int S_ = 512;
-            auto slab = at::empty({S_, rsc_total, cs.Ko},
+            int S_ = (int)std::max<long>(1, std::min<long>(2048, M / 8));
+            auto slab = at::empty({rsc_total * cs.Ko, S_},
                                   x.options().dtype(at::kFloat));
             hipLaunchKernelGGL(k_wgrad_small_rsc, dim3(S_), dim3(256), 0,
                                stream.stream(), dyp, xp,
                                slab.data_ptr<float>(), cs, M, S_, rsc_total);
             HIP_CHECK_LAST();
             int total = rsc_total * cs.Ko;
-            hipLaunchKernelGGL(k_wgrad_small_combine,
-                               dim3(ceil_div_i(total, 256)), dim3(256), 0,
+            hipLaunchKernelGGL(k_wgrad_small_rsc_combine,
+                               dim3(ceil_div_i(total, 4)), dim3(256), 0,
                                stream.stream(), slab.data_ptr<float>(), S_,
                                cs, rsc_total, dw.data_ptr<float>());
             HIP_CHECK_LAST();
