@@ -22,6 +22,7 @@ that exact API shape with MI355X-native mechanics:
 from __future__ import annotations
 
 import contextlib
+import os
 from typing import Optional
 
 import torch
@@ -56,9 +57,41 @@ class _AmpState:
         self.optimizer = None
         self.ddp_model = None
         self.last_overflow = False
+        # async (device-side) overflow tracking — GPU HIP path only
+        self.found_inf = None        # per-step flag, read by the fused
+        #                              optimizer kernels (device-side skip)
+        self.overflow_count = None   # running count, host-synced per epoch
+        self.async_mode = False
 
 
 _state = _AmpState()
+
+
+def _device_buffers(device):
+    if _state.found_inf is None or _state.found_inf.device != device:
+        _state.found_inf = torch.zeros(1, dtype=torch.float32, device=device)
+        _state.overflow_count = torch.zeros(1, dtype=torch.float32,
+                                            device=device)
+    return _state.found_inf
+
+
+def pending_found_inf():
+    """Device-side overflow flag for the fused optimizer kernels (or None)."""
+    return _state.found_inf if _state.async_mode else None
+
+
+def maybe_sync_scaler() -> None:
+    """Epoch-level host sync of the deferred overflow count (async mode):
+    folds any overflows seen this epoch into the dynamic scaler.  With bf16
+    compute overflows are not expected; this keeps the apex policy live
+    without a per-step device->host sync."""
+    if not _state.async_mode or _state.overflow_count is None:
+        return
+    n = int(_state.overflow_count.item())
+    if n > 0:
+        for _ in range(n):
+            _state.scaler.update(found_inf=True)
+        _state.overflow_count.zero_()
 
 
 def is_enabled() -> bool:
@@ -155,6 +188,22 @@ def scale_loss(loss, optimizer, delay_unscale: bool = False):
     if tensors is None:
         tensors = _collect_grads(optimizer)
     inv = 1.0 / (scaler.scale * world)
+
+    # Async path (GPU + HIP ext): unscale+check leaves the flag on device;
+    # the fused SGD/Lookahead kernels skip themselves when it is set and
+    # the scaler syncs once per epoch (maybe_sync_scaler).
+    if tensors and tensors[0].is_cuda and os.environ.get(
+            "DDPX_SYNC_AMP", "0") != "1":
+        from ..ops import load_extension
+        ext = load_extension(required=False)
+        if ext is not None:
+            _state.async_mode = True
+            fi = _device_buffers(tensors[0].device)
+            fi.zero_()  # stream-ordered: prior step's kernels already read it
+            ext.multi_tensor_unscale(tensors, fi, inv)
+            _state.overflow_count += fi
+            return
+
     found_inf = _unscale_and_check(tensors, inv)
     scaler.update(found_inf)
     _state.last_overflow = found_inf

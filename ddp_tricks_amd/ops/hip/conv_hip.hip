@@ -36,6 +36,17 @@ struct ConvShape {
 // (128, or 64 for narrow outputs like dgrad into C=64); the 256-thread
 // block is WAVES_M x WAVES_N waves each owning a (128/WAVES_M) x
 // (TBN/WAVES_N) sub-tile.
+//
+// Staging is __builtin_amdgcn_global_load_lds (direct HBM->LDS, 16 B/lane,
+// no VGPR round trip — cdna_hip_programming.md §5 step 3: +69% on the GEMM
+// ladder).  glds writes lane-linear, so the bank-conflict swizzle lives on
+// the per-lane SOURCE address and is repeated on the fragment reads
+// (rule 21): physical 16B segment = logical k-segment ^ (row & 7), which
+// caps ds_read_b128 conflicts at 2-way on the 128 B rows.  Out-of-range
+// lanes are pointed at a zeroed device buffer (glds has no execution mask
+// that leaves LDS deterministic).
+__device__ __align__(16) unsigned char g_zero16[16];
+
 template <int MODE, int TBN, int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
@@ -45,8 +56,10 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     constexpr int WN = TBN / WAVES_N;
     constexpr int MI = WM / 16;
     constexpr int NI = WN / 16;
-    __shared__ bf16 lds_a[CBM][CLDK];
-    __shared__ bf16 lds_b[TBN][CLDK];
+    constexpr int A_CHUNKS = CBM / 8;      // 1 KiB glds chunks (8 rows)
+    constexpr int B_CHUNKS = TBN / 8;
+    __shared__ bf16 lds_a[CBM][CBK];       // unpadded: glds dest is linear
+    __shared__ bf16 lds_b[TBN][CBK];
     const int m0 = blockIdx.x * CBM;
     const int n0 = blockIdx.y * TBN;
     const int tid = threadIdx.x;
@@ -55,15 +68,17 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     const int wr = wid / WAVES_N, wc = wid % WAVES_N;
 
     f32x4 acc[MI][NI] = {};
-    const int ld_row = tid >> 3;
-    const int ld_col = (tid & 7) * 8;
+    // per-lane piece within a wave chunk: 8 rows x 8 segments of 16 B
+    const int pl_row = lane >> 3;
+    const int pl_segp = lane & 7;          // physical segment (LDS-linear)
 
     for (int kt = 0; kt < Kgemm; kt += CBK) {
-        #pragma unroll
-        for (int p4 = 0; p4 < 4; ++p4) {
-            int row = p4 * 32 + ld_row;
-            int gm = m0 + row, gk = kt + ld_col;
-            bf16x8_t va = {};
+        // ---- A tile: 16 chunks round-robined over the 4 waves ----
+        for (int ch = wid; ch < A_CHUNKS; ch += 4) {
+            int row = ch * 8 + pl_row;
+            int seg = pl_segp ^ (row & 7);     // logical k-segment
+            int gm = m0 + row, gk = kt + seg * 8;
+            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
             if (gm < M && gk < Kgemm) {
                 if (MODE == 0) {
                     int q = gm % cs.Q, rem = gm / cs.Q;
@@ -73,8 +88,7 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     int h = p * cs.stride + r - cs.pad;
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        va = *reinterpret_cast<const bf16x8_t*>(
-                            &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                        src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
                 } else {
                     int wcol = gm % cs.W; long rem = gm / cs.W;
                     int h = rem % cs.H; int n = rem / cs.H;
@@ -83,34 +97,44 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     int p = h + cs.pad - r;
                     int q = wcol + cs.pad - s;
                     if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        va = *reinterpret_cast<const bf16x8_t*>(
-                            &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko]);
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 }
             }
-            *reinterpret_cast<bf16x8_t*>(&lds_a[row][ld_col]) = va;
-
-            if (row < TBN) {
-                int gn = n0 + row;
-                bf16x8_t vb = {};
-                if (gn < Nout && gk < Kgemm)
-                    vb = *reinterpret_cast<const bf16x8_t*>(
-                        &Bsrc[(long)gn * Kgemm + gk]);
-                *reinterpret_cast<bf16x8_t*>(&lds_b[row][ld_col]) = vb;
-            }
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned short*)src,
+                (__attribute__((address_space(3))) unsigned short*)&lds_a[ch * 8][0],
+                16, 0, 0);
         }
-        __syncthreads();
+        // ---- B tile (row-contiguous source) ----
+        for (int ch = wid; ch < B_CHUNKS; ch += 4) {
+            int row = ch * 8 + pl_row;
+            int seg = pl_segp ^ (row & 7);
+            int gn = n0 + row, gk = kt + seg * 8;
+            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
+            if (gn < Nout && gk < Kgemm)
+                src = &Bsrc[(long)gn * Kgemm + gk];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned short*)src,
+                (__attribute__((address_space(3))) unsigned short*)&lds_b[ch * 8][0],
+                16, 0, 0);
+        }
+        __syncthreads();   // hipcc inserts the vmcnt(0) for in-flight glds
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
             bf16x8_t af[MI], bfr[NI];
-            const int kcol = ks + (lane >> 4) * 8;
+            const int kgrp = (ks >> 3) + (lane >> 4);
             #pragma unroll
-            for (int mi = 0; mi < MI; ++mi)
+            for (int mi = 0; mi < MI; ++mi) {
+                int row = wr * WM + mi * 16 + (lane & 15);
                 af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[wr * WM + mi * 16 + (lane & 15)][kcol]);
+                    &lds_a[row][(kgrp ^ (row & 7)) * 8]);
+            }
             #pragma unroll
-            for (int ni = 0; ni < NI; ++ni)
+            for (int ni = 0; ni < NI; ++ni) {
+                int row = wc * WN + ni * 16 + (lane & 15);
                 bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[wc * WN + ni * 16 + (lane & 15)][kcol]);
+                    &lds_b[row][(kgrp ^ (row & 7)) * 8]);
+            }
             #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
                 #pragma unroll

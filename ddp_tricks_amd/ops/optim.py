@@ -62,8 +62,10 @@ class FusedSGD(Optimizer):
             from . import load_extension
             ext = load_extension(required=False)
             if ext is not None:
+                from .. import amp as amp_mod
                 ext.fused_sgd(params, grads, bufs, lr, momentum, wd,
-                              1.0 if nesterov else 0.0)
+                              1.0 if nesterov else 0.0,
+                              amp_mod.pending_found_inf())
                 # raw-pointer writes don't bump Tensor._version — drop the
                 # per-version bf16 weight cast cache so the next forward
                 # re-casts the updated masters

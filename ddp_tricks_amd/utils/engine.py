@@ -63,6 +63,8 @@ def iterate_loader(
         correct_sum += F_ops.argmax_correct(outputs.detach(), target)
         num += image.shape[0]
 
+    if training:
+        amp.maybe_sync_scaler()  # async-amp: fold deferred overflow count
     loss = (loss_sum / num).item()
     acc = (correct_sum.double() / num).item()
 

@@ -63,7 +63,9 @@ class Lookahead(Optimizer):
             from ..ops import load_extension
             ext = load_extension(required=False)
             if ext is not None:
-                ext.fused_lookahead(fasts, slows, self.alpha, None)
+                from .. import amp as amp_mod
+                ext.fused_lookahead(fasts, slows, self.alpha,
+                                    amp_mod.pending_found_inf())
                 from ..ops.functional import clear_weight_cache
                 clear_weight_cache()  # raw writes don't bump _version
                 return
