@@ -47,8 +47,12 @@ def _worker_grad_equiv(rank, port, results):
     loss.backward()
     ddp.finalize_backward(average=True)
     if rank == 0:
-        grads = {k: p.grad.detach().clone() for k, p in model.named_parameters()}
+        # serialize by value (numpy) — tensor fd-sharing over the queue races
+        # with worker exit and the parent's detach can hit a dead listener
+        grads = {k: p.grad.detach().clone().numpy()
+                 for k, p in model.named_parameters()}
         results.put(grads)
+    dist.barrier()
     dist.destroy_process_group()
 
 
@@ -84,7 +88,7 @@ def test_ddp_grads_match_fullbatch():
     # accumulate grads from both replicas
     for (k, p), (k2, p2) in zip(model.named_parameters(), model2.named_parameters()):
         g = (p.grad if p.grad is not None else 0) + (p2.grad if p2.grad is not None else 0)
-        assert torch.allclose(grads[k], g, atol=1e-5), k
+        assert torch.allclose(torch.from_numpy(grads[k]), g, atol=1e-5), k
 
 
 def _worker_broadcast(rank, port, results):
@@ -94,7 +98,9 @@ def _worker_broadcast(rank, port, results):
     model = TinyNet()
     DDP(model)
     if rank == 1:
-        results.put({k: v.detach().clone() for k, v in model.state_dict().items()})
+        results.put({k: v.detach().clone().numpy()
+                     for k, v in model.state_dict().items()})
+    dist.barrier()
     dist.destroy_process_group()
 
 
@@ -112,7 +118,7 @@ def test_param_broadcast_from_rank0():
     torch.manual_seed(100)  # rank 0 init
     ref = TinyNet().state_dict()
     for k in ref:
-        assert torch.allclose(sd1[k], ref[k]), k
+        assert torch.allclose(torch.from_numpy(sd1[k]).float(), ref[k].float()), k
 
 
 def test_singleproc_ddp_grad_views():
