@@ -1,15 +1,19 @@
-"""In-tree build of the _hip_ops extension for gfx950.
+"""In-tree build of the _hip_ops extension for gfx950 — direct hipcc.
 
-Uses torch.utils.cpp_extension (which drives hipcc for the .hip sources
-under PYTORCH_ROCM_ARCH=gfx950) with the build directory inside the package
-so the resulting _hip_ops.so travels with the repo snapshot to GPU boxes.
-hipcc cross-compiles without a GPU — this runs on CPU-only machines too.
+No torch-cpp_extension JIT machinery and no hipify pass: every source under
+``ops/hip/`` is hand-written HIP/CDNA4 and compiled verbatim with hipcc
+(``--offload-arch=gfx950``).  Objects live under ``ops/build/`` (gitignored);
+the linked ``_hip_ops.so`` is copied next to ``ops/__init__.py`` so it travels
+with repo snapshots to GPU boxes.  hipcc cross-compiles without a GPU — this
+runs on CPU-only machines too.  Include/library paths come from the installed
+torch so the extension links against the same libtorch the process loads.
 """
 from __future__ import annotations
 
 import os
-import shutil
+import subprocess
 import sys
+from concurrent.futures import ThreadPoolExecutor
 
 OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(OPS_DIR, "hip")
@@ -25,27 +29,80 @@ SOURCES = [
     "conv.hip",
 ]
 
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+
+
+def _flags():
+    import sysconfig
+
+    import torch
+    from torch.utils import cpp_extension as ce
+
+    inc = list(ce.include_paths(device_type="cuda"))
+    inc.append(sysconfig.get_paths()["include"])
+    lib = list(ce.library_paths(device_type="cuda"))
+    abi = int(getattr(torch._C, "_GLIBCXX_USE_CXX11_ABI", True))
+    cflags = (
+        ["-O3", "-std=c++17", "-fPIC",
+         "-DTORCH_EXTENSION_NAME=_hip_ops",
+         "-DTORCH_API_INCLUDE_EXTENSION_H",
+         "-DUSE_ROCM=1", "-D__HIP_PLATFORM_AMD__=1",
+         f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+         "-DHIP_ENABLE_WARP_SYNC_BUILTINS=1"]
+        + [f"-I{d}" for d in inc]
+    )
+    dev = ["--offload-arch=gfx950", "-fno-gpu-rdc",
+           "-Wno-unused-result", "-ffast-math"]
+    ldflags = ([f"-L{d}" for d in lib]
+               + ["-lc10", "-lc10_hip", "-ltorch", "-ltorch_cpu",
+                  "-ltorch_hip", "-ltorch_python", "-lamdhip64"]
+               + [f"-Wl,-rpath,{d}" for d in lib])
+    return cflags, dev, ldflags
+
+
+def _stale(obj: str, src: str) -> bool:
+    if not os.path.exists(obj):
+        return True
+    omt = os.path.getmtime(obj)
+    hdrs = [os.path.join(HIP_DIR, h) for h in os.listdir(HIP_DIR)
+            if h.endswith(".h")]
+    return any(os.path.getmtime(f) > omt for f in [src] + hdrs)
+
 
 def build(verbose: bool = False):
-    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
-    os.environ.setdefault("MAX_JOBS", "8")
-    from torch.utils.cpp_extension import load
-
     os.makedirs(BUILD_DIR, exist_ok=True)
-    module = load(
-        name="_hip_ops",
-        sources=[os.path.join(HIP_DIR, s) for s in SOURCES],
-        extra_cflags=["-O3"],
-        extra_cuda_cflags=["-O3", "--offload-arch=gfx950", "-std=c++17"],
-        build_directory=BUILD_DIR,
-        verbose=verbose,
-        is_python_module=True,
-    )
-    # copy the .so next to ops/__init__.py so `import _hip_ops` finds it
-    so_path = os.path.join(BUILD_DIR, "_hip_ops.so")
-    if os.path.exists(so_path):
-        shutil.copy2(so_path, os.path.join(OPS_DIR, "_hip_ops.so"))
-    return module
+    cflags, dev, ldflags = _flags()
+
+    def compile_one(src_name: str) -> str:
+        src = os.path.join(HIP_DIR, src_name)
+        obj = os.path.join(BUILD_DIR, os.path.splitext(src_name)[0] + ".o")
+        if not _stale(obj, src):
+            return obj
+        if src_name.endswith(".hip"):
+            cmd = [HIPCC, "-c", src, "-o", obj] + cflags + dev
+        else:  # host-only bindings: plain C++ compile
+            cmd = ["g++", "-c", src, "-o", obj] + cflags
+        if verbose:
+            print(" ".join(cmd))
+        r = subprocess.run(cmd, capture_output=True, text=True)
+        if r.returncode != 0:
+            raise RuntimeError(
+                f"hipcc failed on {src_name}:\n{r.stdout}\n{r.stderr}")
+        return obj
+
+    with ThreadPoolExecutor(max_workers=min(8, len(SOURCES))) as ex:
+        objs = list(ex.map(compile_one, SOURCES))
+
+    so_path = os.path.join(OPS_DIR, "_hip_ops.so")
+    if _stale(so_path, max(objs, key=os.path.getmtime)) or any(
+            os.path.getmtime(o) > os.path.getmtime(so_path) for o in objs):
+        cmd = [HIPCC, "-shared", "-o", so_path] + objs + ldflags
+        if verbose:
+            print(" ".join(cmd))
+        r = subprocess.run(cmd, capture_output=True, text=True)
+        if r.returncode != 0:
+            raise RuntimeError(f"link failed:\n{r.stdout}\n{r.stderr}")
+    return so_path
 
 
 if __name__ == "__main__":
