@@ -117,7 +117,8 @@ def initialize(model, optimizers, opt_level: str = "O1", loss_scale="dynamic"):
         _state.scaler = DynamicLossScaler(init_scale=float(loss_scale),
                                           growth_interval=10 ** 12)
     _state.optimizer = optimizers
-    _patch_step(optimizers)
+    if optimizers is not None:
+        _patch_step(optimizers)
     return model, optimizers
 
 

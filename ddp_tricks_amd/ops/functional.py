@@ -176,20 +176,22 @@ class _HIPBatchNorm(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                training, momentum, eps, fuse_relu):
+                training, momentum, eps, fuse_relu, residual):
         ext = require_ext_for(x)
         xb = _chlast(_to_bf16(x))
+        rb = None if residual is None else _chlast(_to_bf16(residual))
         if training:
             y, save_mean, save_invstd = ext.bn_fwd_train(
                 xb, weight.detach(), bias.detach(), running_mean, running_var,
-                momentum, eps, fuse_relu)
+                momentum, eps, fuse_relu, rb)
             ctx.save_for_backward(xb, weight, save_mean, save_invstd, y)
         else:
             y = ext.bn_fwd_eval(xb, weight.detach(), bias.detach(),
-                                running_mean, running_var, eps, fuse_relu)
+                                running_mean, running_var, eps, fuse_relu, rb)
         ctx.fuse_relu = fuse_relu
         ctx.training = training
         ctx.x_dtype = x.dtype
+        ctx.has_residual = residual is not None
         return y
 
     @staticmethod
@@ -198,21 +200,28 @@ class _HIPBatchNorm(torch.autograd.Function):
         xb, weight, save_mean, save_invstd, y = ctx.saved_tensors
         ext = require_ext_for(dy)
         dyb = _chlast(_to_bf16(dy))
-        dx, dweight, dbias = ext.bn_bwd(xb, dyb, weight.detach(), save_mean,
-                                        save_invstd, y, ctx.fuse_relu)
+        out = ext.bn_bwd(xb, dyb, weight.detach(), save_mean,
+                         save_invstd, y, ctx.fuse_relu, ctx.has_residual)
+        dx, dweight, dbias = out[0], out[1], out[2]
+        dresid = out[3] if ctx.has_residual else None
         if ctx.x_dtype == torch.float32:
             dx = dx.float()
-        return dx, dweight, dbias, None, None, None, None, None, None
+        return (dx, dweight, dbias, None, None, None, None, None, None,
+                dresid)
 
 
 def batch_norm(x, running_mean, running_var, weight, bias,
-               training, momentum, eps, fuse_relu=False):
+               training, momentum, eps, fuse_relu=False, residual=None):
+    """BN with optionally fused ReLU and residual add: relu(bn(x) + residual)
+    — the ResNet block epilogue in one kernel (skip grad returned in bwd)."""
     if x.is_cuda and require_ext_for(x) is not None:
         return _HIPBatchNorm.apply(x, weight, bias, running_mean, running_var,
-                                   training, momentum, eps, fuse_relu)
+                                   training, momentum, eps, fuse_relu, residual)
     xf = x.float() if x.dtype != torch.float32 else x
     y = F.batch_norm(xf, running_mean, running_var, weight, bias,
                      training, momentum, eps)
+    if residual is not None:
+        y = y + residual.float()
     if fuse_relu:
         y = F.relu(y)
     return y.to(x.dtype) if x.is_cuda and amp_mod.is_enabled() else y
@@ -239,13 +248,60 @@ class _HIPMaxPool2x2(torch.autograd.Function):
         return dx
 
 
-def max_pool2d(x, kernel_size=2, stride=None):
+class _HIPMaxPool(torch.autograd.Function):
+    """General max pool (kernel ks, stride st, padding pad) — ResNet stem."""
+
+    @staticmethod
+    def forward(ctx, x, ks, st, pad):
+        ext = require_ext_for(x)
+        xb = _chlast(_to_bf16(x))
+        y, idx = ext.maxpool_fwd(xb, ks, st, pad)
+        ctx.save_for_backward(idx)
+        ctx.meta = (xb.shape[2], xb.shape[3], ks, st, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        H, W, ks, st, pad = ctx.meta
+        ext = require_ext_for(dy)
+        dx = ext.maxpool_bwd(_chlast(_to_bf16(dy)), idx, H, W, ks, st, pad)
+        return dx, None, None, None
+
+
+def max_pool2d(x, kernel_size=2, stride=None, padding=0):
     ks = kernel_size[0] if isinstance(kernel_size, (tuple, list)) else kernel_size
     st = stride or ks
     st = st[0] if isinstance(st, (tuple, list)) else st
-    if x.is_cuda and require_ext_for(x) is not None and ks == 2 and st == 2:
-        return _HIPMaxPool2x2.apply(x)
-    return F.max_pool2d(x, kernel_size, stride)
+    pad = padding[0] if isinstance(padding, (tuple, list)) else padding
+    if x.is_cuda and require_ext_for(x) is not None:
+        if ks == 2 and st == 2 and pad == 0:
+            return _HIPMaxPool2x2.apply(x)
+        return _HIPMaxPool.apply(x, ks, st, pad)
+    return F.max_pool2d(x, kernel_size, stride, padding)
+
+
+class _HIPGlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_ext_for(x)
+        xb = _chlast(_to_bf16(x))
+        ctx.hw = (xb.shape[2], xb.shape[3])
+        return ext.global_avgpool_fwd(xb)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext_for(dy)
+        return ext.global_avgpool_bwd(_to_bf16(dy.contiguous()), *ctx.hw)
+
+
+def global_avg_pool2d(x):
+    """[N,C,H,W] -> [N,C] mean over H,W (ResNet head: adaptive avgpool 1x1
+    + flatten in one op)."""
+    if x.is_cuda and require_ext_for(x) is not None:
+        return _HIPGlobalAvgPool.apply(x)
+    return x.float().mean(dim=(2, 3)).to(x.dtype) \
+        if x.is_cuda and amp_mod.is_enabled() else x.mean(dim=(2, 3))
 
 
 def relu(x, inplace=False):

@@ -20,18 +20,25 @@ at::Tensor argmax_correct(at::Tensor logits, at::Tensor target);
 // pool.hip
 std::vector<at::Tensor> maxpool2x2_fwd(at::Tensor x);
 at::Tensor maxpool2x2_bwd(at::Tensor dy, at::Tensor idx, long H, long W);
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long ks, long st, long pad);
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
+                       long ks, long st, long pad);
+at::Tensor global_avgpool_fwd(at::Tensor x);
+at::Tensor global_avgpool_bwd(at::Tensor dy, long H, long W);
 
 // bn.hip
 std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
-                                     double eps, bool fuse_relu);
+                                     double eps, bool fuse_relu,
+                                     c10::optional<at::Tensor> residual);
 at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        at::Tensor running_mean, at::Tensor running_var,
-                       double eps, bool fuse_relu);
+                       double eps, bool fuse_relu,
+                       c10::optional<at::Tensor> residual);
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
-                               at::Tensor y, bool fuse_relu);
+                               at::Tensor y, bool fuse_relu, bool want_dresid);
 
 // gemm.hip
 at::Tensor gemm_tn(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> bias,
@@ -66,9 +73,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("argmax_correct", &argmax_correct);
     m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
     m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
-    m.def("bn_fwd_train", &bn_fwd_train);
-    m.def("bn_fwd_eval", &bn_fwd_eval);
-    m.def("bn_bwd", &bn_bwd);
+    m.def("maxpool_fwd", &maxpool_fwd);
+    m.def("maxpool_bwd", &maxpool_bwd);
+    m.def("global_avgpool_fwd", &global_avgpool_fwd);
+    m.def("global_avgpool_bwd", &global_avgpool_bwd);
+    m.def("bn_fwd_train", &bn_fwd_train,
+          py::arg("x"), py::arg("gamma"), py::arg("beta"),
+          py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+          py::arg("eps"), py::arg("fuse_relu") = false,
+          py::arg("residual") = c10::nullopt);
+    m.def("bn_fwd_eval", &bn_fwd_eval,
+          py::arg("x"), py::arg("gamma"), py::arg("beta"),
+          py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
+          py::arg("fuse_relu") = false, py::arg("residual") = c10::nullopt);
+    m.def("bn_bwd", &bn_bwd,
+          py::arg("x"), py::arg("dy"), py::arg("gamma"), py::arg("save_mean"),
+          py::arg("save_invstd"), py::arg("y"), py::arg("fuse_relu") = false,
+          py::arg("want_dresid") = false);
     m.def("gemm_tn", &gemm_tn, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
     m.def("transpose_bf16", &transpose_bf16);

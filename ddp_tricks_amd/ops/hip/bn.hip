@@ -113,20 +113,25 @@ __global__ void k_bn_eval_coeffs(const float* __restrict__ gamma,
     shift[c] = b - running_mean[c] * sc;
 }
 
-// apply: y = relu?(x*scale[c] + shift[c]) — bf16x8
+// apply: y = relu?(x*scale[c] + shift[c] (+ residual)) — bf16x8.  The
+// optional residual add feeds ResNet skip connections without a separate
+// elementwise pass.
 __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
                               const float* __restrict__ scale,
                               const float* __restrict__ shift,
-                              long total_v, int Cv, bool relu) {
+                              long total_v, int Cv, bool relu,
+                              const bf16* __restrict__ resid) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
     for (; i < total_v; i += stride) {
         int cv = (i % Cv) * 8;
         s16x8 v = reinterpret_cast<const s16x8*>(x)[i];
+        s16x8 rv = resid ? reinterpret_cast<const s16x8*>(resid)[i] : s16x8{};
         s16x8 o;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float f = fmaf(us2f((unsigned short)v[j]), scale[cv + j], shift[cv + j]);
+            if (resid) f += us2f((unsigned short)rv[j]);
             if (relu) f = fmaxf(f, 0.f);
             o[j] = (short)f2us(f);
         }
@@ -226,7 +231,8 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const float* __restrict__ coef_b,
                             const float* __restrict__ coef_c,
                             bf16* __restrict__ dx,
-                            long M, int C, bool relu) {
+                            long M, int C, bool relu,
+                            bf16* __restrict__ dresid) {
     const int cpg = C >> 3;
     const int c8 = threadIdx.x % cpg;
     const int walker = threadIdx.x / cpg;
@@ -247,15 +253,18 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
         s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y + r * C)[c8] : vg;
-        s16x8 o;
+        s16x8 o, og;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float g = us2f((unsigned short)vg[j]);
             if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
             float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
             o[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
+            og[j] = (short)f2us(g);
         }
         reinterpret_cast<s16x8*>(dx + r * C)[c8] = o;
+        if (dresid)   // skip-connection grad = relu-masked dy
+            reinterpret_cast<s16x8*>(dresid + r * C)[c8] = og;
     }
 }
 
@@ -281,7 +290,8 @@ static int pick_block(int C) {
 std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
-                                     double eps, bool fuse_relu) {
+                                     double eps, bool fuse_relu,
+                                     c10::optional<at::Tensor> residual) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0 && C / 8 <= 256, "bn HIP path needs C%8==0");
@@ -317,16 +327,19 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
     long tv = M * C / 8;
     bf16* yp = reinterpret_cast<bf16*>(y.data_ptr());
     int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
+    const bf16* rp = residual.has_value()
+        ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;
     hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
                        stream.stream(), xp, yp, scale.data_ptr<float>(),
-                       shift.data_ptr<float>(), tv, C / 8, fuse_relu);
+                       shift.data_ptr<float>(), tv, C / 8, fuse_relu, rp);
     HIP_CHECK_LAST();
     return {y, save_mean, save_invstd};
 }
 
 at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        at::Tensor running_mean, at::Tensor running_var,
-                       double eps, bool fuse_relu) {
+                       double eps, bool fuse_relu,
+                       c10::optional<at::Tensor> residual) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0);
@@ -345,19 +358,21 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
     HIP_CHECK_LAST();
     long tv = M * C / 8;
     int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
+    const bf16* rp = residual.has_value()
+        ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;
     hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
                        scale.data_ptr<float>(), shift.data_ptr<float>(), tv,
-                       C / 8, fuse_relu);
+                       C / 8, fuse_relu, rp);
     HIP_CHECK_LAST();
     return y;
 }
 
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
-                               at::Tensor y, bool fuse_relu) {
+                               at::Tensor y, bool fuse_relu, bool want_dresid) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0 && C / 8 <= 256);
@@ -393,6 +408,14 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                        ca.data_ptr<float>(), cb.data_ptr<float>(),
                        cc.data_ptr<float>());
     HIP_CHECK_LAST();
+    at::Tensor dresid;
+    bf16* drp = nullptr;
+    if (want_dresid) {
+        dresid = x.dim() == 4
+            ? at::empty_like(x, x.options().memory_format(at::MemoryFormat::ChannelsLast))
+            : at::empty_like(x);
+        drp = reinterpret_cast<bf16*>(dresid.data_ptr());
+    }
     int dxblocks = (int)std::max<long>(1, std::min<long>(2048, M / (nw * 2)));
     hipLaunchKernelGGL(k_bn_bwd_dx, dim3(dxblocks), dim3(block), 0,
                        stream.stream(),
@@ -400,7 +423,8 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                        save_invstd.data_ptr<float>(), ca.data_ptr<float>(),
                        cb.data_ptr<float>(), cc.data_ptr<float>(),
                        reinterpret_cast<bf16*>(dx.data_ptr()), M, C,
-                       fuse_relu);
+                       fuse_relu, drp);
     HIP_CHECK_LAST();
+    if (want_dresid) return {dx, dgamma, dbeta, dresid};
     return {dx, dgamma, dbeta};
 }

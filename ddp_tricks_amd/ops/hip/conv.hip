@@ -93,9 +93,12 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     int h = rem % cs.H; int n = rem / cs.H;
                     int ko = gk % cs.Ko, rs = gk / cs.Ko;
                     int r = rs / cs.S, s = rs % cs.S;
-                    int p = h + cs.pad - r;
-                    int q = wcol + cs.pad - s;
-                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                    int pn = h + cs.pad - r;
+                    int qn = wcol + cs.pad - s;
+                    // strided conv: only taps where stride divides contribute
+                    int p = pn / cs.stride, q = qn / cs.stride;
+                    if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
+                        qn == q * cs.stride && p < cs.P && q < cs.Q)
                         src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 }
             }
@@ -475,7 +478,6 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
                         long H, long W, long R, long S, long stride, long pad) {
     // dy: NCHW logical / channels_last bf16 [N,Ko,P,Q]; wt2: [C, R*S*Ko] bf16
-    TORCH_CHECK(stride == 1, "dgrad stride>1 not implemented yet");
     ConvShape cs;
     cs.N = N; cs.C = C; cs.H = H; cs.W = W;
     cs.Ko = dy.size(1); cs.P = dy.size(2); cs.Q = dy.size(3);

@@ -26,7 +26,7 @@ class Linear(nn.Linear):
 
 
 class _BatchNormBase:
-    def _bn_forward(self, x):
+    def _bn_forward(self, x, residual=None):
         self._check_input_dim(x)
         if self.training:
             if self.num_batches_tracked is not None:
@@ -36,7 +36,7 @@ class _BatchNormBase:
         return F_ops.batch_norm(
             x, self.running_mean, self.running_var, self.weight, self.bias,
             self.training or not self.track_running_stats, momentum, self.eps,
-            fuse_relu=fuse_relu)
+            fuse_relu=fuse_relu, residual=residual)
 
 
 class BatchNorm2d(nn.BatchNorm2d, _BatchNormBase):
@@ -44,8 +44,10 @@ class BatchNorm2d(nn.BatchNorm2d, _BatchNormBase):
         super().__init__(*args, **kwargs)
         self.fuse_relu = fuse_relu
 
-    def forward(self, x):
-        return self._bn_forward(x)
+    def forward(self, x, residual=None):
+        # residual: optional skip tensor added before the (fused) ReLU —
+        # relu(bn(x) + residual) in one kernel on GPU (ResNet blocks)
+        return self._bn_forward(x, residual)
 
 
 class BatchNorm1d(nn.BatchNorm1d, _BatchNormBase):
@@ -64,7 +66,15 @@ class ReLU(nn.ReLU):
 
 class MaxPool2d(nn.MaxPool2d):
     def forward(self, x):
-        return F_ops.max_pool2d(x, self.kernel_size, self.stride)
+        return F_ops.max_pool2d(x, self.kernel_size, self.stride, self.padding)
+
+
+class AdaptiveAvgPool2d(nn.AdaptiveAvgPool2d):
+    """Only the (1,1) global case is supported (the ResNet head)."""
+
+    def forward(self, x):
+        assert self.output_size in (1, (1, 1))
+        return F_ops.global_avg_pool2d(x).reshape(x.shape[0], x.shape[1], 1, 1)
 
 
 class Flatten(nn.Flatten):

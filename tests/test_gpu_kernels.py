@@ -313,3 +313,126 @@ def test_multi_tensor_unscale():
     g[0][17] = float("inf")
     ext.multi_tensor_unscale(g, fi, 1.0)
     assert fi.item() == 1.0
+
+
+# ------------------------------------------------------- ResNet kernel set ---
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 32, 32, 128, 3, 2, 1),    # stride-2 3x3 (ResNet transition)
+    (8, 64, 32, 32, 128, 1, 2, 0),    # stride-2 1x1 (downsample)
+    (8, 64, 32, 32, 64, 1, 1, 0),     # 1x1
+    (4, 3, 64, 64, 64, 7, 2, 3),      # ImageNet stem (small-Cin direct path)
+])
+def test_conv_fwd_resnet_shapes(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(5)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    y = ext.conv2d_fwd(x, w, None, stride, pad)
+    _close(y, _conv_ref(x, w, None, stride, pad), name=f"conv fwd {shape}")
+
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 32, 32, 128, 3, 2, 1),
+    (8, 64, 32, 32, 128, 1, 2, 0),
+    (8, 128, 16, 16, 128, 3, 1, 1),
+])
+def test_conv_dgrad_strided(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(6)
+    P = (H + 2 * pad - R) // stride + 1
+    dy = torch.randn(N, K, P, P, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16)
+    wt2 = w.permute(1, 2, 3, 0).reshape(C, R * R * K).contiguous()
+    dx = ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R, stride, pad)
+    ref = torch.nn.grad.conv2d_input((N, C, H, W), w.float(), dy.float(),
+                                     stride=stride, padding=pad)
+    _close(dx, ref, name=f"conv dgrad {shape}")
+
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 32, 32, 128, 3, 2, 1),
+    (8, 64, 32, 32, 128, 1, 2, 0),
+    (4, 3, 64, 64, 64, 7, 2, 3),      # stem wgrad (small-Cin path)
+])
+def test_conv_wgrad_strided(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(7)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    P = (H + 2 * pad - R) // stride + 1
+    dy = torch.randn(N, K, P, P, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    dw = ext.conv2d_wgrad(dy, x, R, R, stride, pad)
+    ref = torch.nn.grad.conv2d_weight(x.float(), (K, C, R, R), dy.float(),
+                                      stride=stride, padding=pad)
+    _close(dw, ref, rel=2e-2, atol=1.0, name=f"conv wgrad {shape}")
+
+
+def test_maxpool_3x3s2p1():
+    N, C, H, W = 8, 64, 56, 56
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    y, idx = ext.maxpool_fwd(x, 3, 2, 1)
+    ref = torch.nn.functional.max_pool2d(x.float(), 3, 2, 1)
+    _close(y, ref, rel=0, atol=1e-6, name="maxpool3 fwd")
+    dy = torch.randn_like(y).contiguous(memory_format=CL)
+    dx = ext.maxpool_bwd(dy, idx, H, W, 3, 2, 1)
+    xf = x.float().requires_grad_(True)
+    torch.nn.functional.max_pool2d(xf, 3, 2, 1).backward(dy.float())
+    # ties in a window can pick a different argmax than torch after bf16
+    # round-trip of dy; inputs are continuous so ties have measure ~0
+    _close(dx, xf.grad, rel=1e-2, atol=1e-2, name="maxpool3 bwd")
+
+
+def test_global_avgpool():
+    N, C, H, W = 16, 512, 7, 7
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    y = ext.global_avgpool_fwd(x)
+    _close(y, x.float().mean(dim=(2, 3)), rel=1e-2, atol=1e-3, name="gap fwd")
+    dy = torch.randn(N, C, device=DEV).to(torch.bfloat16)
+    dx = ext.global_avgpool_bwd(dy, H, W)
+    ref = (dy.float() / (H * W)).unsqueeze(-1).unsqueeze(-1).expand(N, C, H, W)
+    _close(dx, ref, rel=1e-2, atol=1e-4, name="gap bwd")
+
+
+def test_bn_residual_fused():
+    N, C, H, W = 16, 64, 14, 14
+    torch.manual_seed(8)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    res = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    g = torch.randn(C, device=DEV).abs() + 0.5
+    b = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True, res)
+
+    xf = x.float().detach().requires_grad_(True)
+    rf = res.float().detach().requires_grad_(True)
+    g2 = g.detach().requires_grad_(True)
+    b2 = b.detach().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        xf, torch.zeros(C, device=DEV), torch.ones(C, device=DEV),
+        g2, b2, True, 0.1, 1e-5)
+    ref = (ref + rf).relu()
+    _close(y, ref, name="bn+res fwd")
+
+    dy = torch.randn_like(ref).to(torch.bfloat16)
+    ref.backward(dy.float())
+    out = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si, y, True, True)
+    dx, dg, db, dres = out
+    _close(dg, g2.grad, rel=2e-2, atol=0.1, name="bn+res dgamma")
+    _close(db, b2.grad, rel=2e-2, atol=0.1, name="bn+res dbeta")
+    _close(dx, xf.grad, rel=5e-2, atol=2e-2, name="bn+res dx")
+    _close(dres, rf.grad, rel=2e-2, atol=1e-2, name="bn+res dresid")
+
+
+def test_bn_eval_residual():
+    N, C, H, W = 8, 64, 12, 12
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    res = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).contiguous(memory_format=CL)
+    g = torch.randn(C, device=DEV)
+    b = torch.randn(C, device=DEV)
+    rm = torch.randn(C, device=DEV)
+    rv = torch.rand(C, device=DEV) + 0.5
+    y = ext.bn_fwd_eval(x, g, b, rm, rv, 1e-5, True, res)
+    ref = (torch.nn.functional.batch_norm(x.float(), rm, rv, g, b, False,
+                                          0.1, 1e-5) + res.float()).relu()
+    _close(y, ref, name="bn eval resid")

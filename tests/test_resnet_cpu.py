@@ -1,0 +1,74 @@
+"""ResNet model family on CPU: shapes, torchvision-compatible state_dict
+layout, registry, residual BN fallback semantics (BASELINE configs 4-5)."""
+import torch
+
+from ddp_tricks_amd.models import build_model, resnet18, resnet50
+
+
+def test_resnet18_cifar_shapes():
+    m = resnet18(num_classes=10, cifar_stem=True)
+    y = m(torch.randn(2, 3, 32, 32))
+    assert y.shape == (2, 10)
+    y.sum().backward()
+    assert all(p.grad is not None for p in m.parameters())
+
+
+def test_resnet50_imagenet_shapes():
+    m = resnet50()
+    y = m(torch.randn(1, 3, 224, 224))
+    assert y.shape == (1, 1000)
+
+
+def test_param_counts_match_torchvision():
+    # canonical torchvision counts for these architectures
+    assert sum(p.numel() for p in resnet18().parameters()) == 11_689_512
+    assert sum(p.numel() for p in resnet50().parameters()) == 25_557_032
+
+
+def test_state_dict_keys_torchvision_layout():
+    sd = resnet18().state_dict()
+    for k in ("conv1.weight", "bn1.weight", "bn1.running_mean",
+              "layer1.0.conv1.weight", "layer2.0.downsample.0.weight",
+              "layer2.0.downsample.1.running_var", "layer4.1.bn2.bias",
+              "fc.weight", "fc.bias"):
+        assert k in sd, k
+    sd50 = resnet50().state_dict()
+    for k in ("layer1.0.conv3.weight", "layer1.0.bn3.weight",
+              "layer1.0.downsample.0.weight", "layer3.5.conv2.weight"):
+        assert k in sd50, k
+
+
+def test_registry():
+    m = build_model("resnet18", num_classes=10, cifar_stem=True)
+    assert m(torch.randn(1, 3, 32, 32)).shape == (1, 10)
+
+
+def test_residual_bn_cpu_semantics():
+    """CPU fallback of batch_norm(residual=...) == bn -> +res -> relu."""
+    from ddp_tricks_amd.ops.functional import batch_norm
+    torch.manual_seed(0)
+    x = torch.randn(4, 8, 5, 5)
+    res = torch.randn(4, 8, 5, 5)
+    w = torch.rand(8) + 0.5
+    b = torch.randn(8)
+    rm = torch.zeros(8)
+    rv = torch.ones(8)
+    y = batch_norm(x, rm, rv, w, b, True, 0.1, 1e-5, fuse_relu=True,
+                   residual=res)
+    ref = (torch.nn.functional.batch_norm(
+        x, torch.zeros(8), torch.ones(8), w, b, True, 0.1, 1e-5) + res).relu()
+    assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_maxpool_padding_cpu():
+    from ddp_tricks_amd.ops.functional import max_pool2d
+    x = torch.randn(2, 4, 9, 9)
+    y = max_pool2d(x, 3, 2, 1)
+    ref = torch.nn.functional.max_pool2d(x, 3, 2, 1)
+    assert torch.equal(y, ref)
+
+
+def test_global_avgpool_cpu():
+    from ddp_tricks_amd.ops.functional import global_avg_pool2d
+    x = torch.randn(2, 4, 7, 7)
+    assert torch.allclose(global_avg_pool2d(x), x.mean(dim=(2, 3)))
