@@ -51,8 +51,9 @@ def _flags():
          "-DHIP_ENABLE_WARP_SYNC_BUILTINS=1"]
         + [f"-I{d}" for d in inc]
     )
-    dev = ["--offload-arch=gfx950", "-fno-gpu-rdc",
-           "-Wno-unused-result", "-ffast-math"]
+    # NOTE: no -ffast-math — finite-math-only would fold away the isfinite()
+    # overflow checks in the amp unscale kernel
+    dev = ["--offload-arch=gfx950", "-fno-gpu-rdc", "-Wno-unused-result"]
     ldflags = ([f"-L{d}" for d in lib]
                + ["-lc10", "-lc10_hip", "-ltorch", "-ltorch_cpu",
                   "-ltorch_hip", "-ltorch_python", "-lamdhip64"]

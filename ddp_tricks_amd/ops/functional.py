@@ -29,9 +29,12 @@ _WCACHE = {}  # param -> {"version": v, variant: tensor}
 def weight_variant(w: torch.Tensor, variant: str) -> torch.Tensor:
     """Per-step cached transforms of fp32 master weights.
 
-    variants: "flat"  — bf16, native contiguous (linear [N,K])
-              "nhwc"  — bf16 channels_last (conv KRSC memory)
-              "wt2"   — bf16 [C, R*S*K] (conv dgrad operand)
+    variants: "flat"     — bf16, native contiguous (linear [N,K])
+              "nhwc"     — bf16 channels_last (conv KRSC memory)
+              "wt2"      — bf16 [C, R*S*K] (conv dgrad operand)
+              "nhwc_p8"  — bf16 channels_last with Cin zero-padded to 8
+                           (MFMA path for C<8 stems, e.g. ResNet RGB)
+              "wt2_p8"   — dgrad operand of the padded weights
     """
     ent = _WCACHE.get(w)
     if ent is None or ent["version"] != w._version:
@@ -46,6 +49,14 @@ def weight_variant(w: torch.Tensor, variant: str) -> torch.Tensor:
         elif variant == "wt2":
             K, C, R, S = w.shape
             ent[variant] = wb.permute(1, 2, 3, 0).reshape(C, R * S * K).contiguous()
+        elif variant in ("nhwc_p8", "wt2_p8"):
+            K, C, R, S = w.shape
+            wp = torch.zeros(K, 8, R, S, dtype=torch.bfloat16, device=w.device)
+            wp[:, :C] = wb
+            if variant == "nhwc_p8":
+                ent[variant] = wp.contiguous(memory_format=CL)
+            else:
+                ent[variant] = wp.permute(1, 2, 3, 0).reshape(8, R * S * K).contiguous()
         else:
             raise KeyError(variant)
     return ent[variant]
@@ -78,35 +89,60 @@ def _nhwc_2d(x: torch.Tensor) -> torch.Tensor:
 
 # ----------------------------------------------------------------- conv2d ---
 
+def _pad8(xb: torch.Tensor) -> torch.Tensor:
+    """Zero-pad channels to 8 (channels_last) so C<8 inputs can take the
+    MFMA implicit-GEMM path — worth it at ResNet-stem scale where the
+    direct VALU kernel is compute-bound."""
+    N, C, H, W = xb.shape
+    xp = torch.zeros(N, 8, H, W, dtype=xb.dtype, device=xb.device)
+    xp = xp.contiguous(memory_format=CL)
+    xp[:, :C] = xb
+    return xp
+
+
+def _use_pad8(C: int, H: int, W: int) -> bool:
+    return C < 8 and H * W >= 1024
+
+
 class _HIPConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
         ext = require_ext_for(x)
         xb = _chlast(_to_bf16(x))
-        wb = weight_variant(weight, "nhwc")
+        pad8 = _use_pad8(weight.shape[1], xb.shape[2], xb.shape[3])
+        if pad8:
+            xb = _pad8(xb)
+            wb = weight_variant(weight, "nhwc_p8")
+        else:
+            wb = weight_variant(weight, "nhwc")
         bias_f = bias if bias is None else bias.detach().float()
         y = ext.conv2d_fwd(xb, wb, bias_f, stride, padding)
         ctx.save_for_backward(xb, weight)
-        ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None)
+        ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
+                    pad8)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         xb, weight = ctx.saved_tensors
-        stride, padding, x_dtype, w_shape, has_bias = ctx.meta
+        stride, padding, x_dtype, w_shape, has_bias, pad8 = ctx.meta
         ext = require_ext_for(dy)
         dyb = _chlast(_to_bf16(dy))
         K, C, R, S = w_shape
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            wt2 = weight_variant(weight, "wt2")
-            dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], C,
+            wt2 = weight_variant(weight, "wt2_p8" if pad8 else "wt2")
+            dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], xb.shape[1],
                                   xb.shape[2], xb.shape[3], R, S,
                                   stride, padding)
+            if pad8:
+                dx = dx[:, :C]
             if x_dtype == torch.float32:
                 dx = dx.float()
         if ctx.needs_input_grad[1]:
             dw = ext.conv2d_wgrad(dyb, xb, R, S, stride, padding)
+            if pad8:
+                dw = dw[:, :C].contiguous()
         if has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(_nhwc_2d(dyb))
         return dx, dw, db, None, None

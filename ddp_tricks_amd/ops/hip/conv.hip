@@ -211,16 +211,20 @@ __global__ void k_conv_small_cin(const bf16* __restrict__ x,
 
 // slab[s][ko, rsc] = Σ_{m in split s} dy[m][ko] · im2col(x)[m][rsc]
 // 64×64 output tile per block (4 waves, 32×32 each), contraction staged
-// 32-deep with LDS-transposed writes (sources are m-major).
-constexpr int WBM = 64, WBN = 64, WBK = 32;
-constexpr int WLDK = WBK + 8;
+// 64-deep through DOUBLE-BUFFERED LDS with register staging (the writes
+// transpose m-major sources to [ko|rsc][m] images, which glds cannot do —
+// cdna_hip_programming.md T10/T14 note).  One barrier per 64-m step: the
+// t+1 tile is gathered into registers while t's MFMAs run, stored to the
+// other buffer, then a single barrier publishes it.
+constexpr int WBM = 64, WBN = 64, WBK = 64;
+constexpr int WLDM = WBK + 8;
 
 __global__ __launch_bounds__(256)
 void k_conv_wgrad(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                   float* __restrict__ slab, ConvShape cs, long M, int Kgemm,
                   int S) {
-    __shared__ bf16 lds_a[WBM][WLDK];   // [ko][m]
-    __shared__ bf16 lds_b[WBN][WLDK];   // [rsc][m]
+    __shared__ bf16 lds_a[2][WBM][WLDM];   // [buf][ko][m]
+    __shared__ bf16 lds_b[2][WBN][WLDM];   // [buf][rsc][m]
     const int ko0 = blockIdx.x * WBM;
     const int rc0 = blockIdx.y * WBN;
     const int split = blockIdx.z;
@@ -229,56 +233,89 @@ void k_conv_wgrad(const bf16* __restrict__ dy, const bf16* __restrict__ x,
     const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
 
     f32x4 acc[2][2] = {};
-    // staging assignment: thread -> (mloc = tid%32, j8 = (tid/32)*8)
-    const int mloc = tid & 31;
-    const int j8 = (tid >> 5) * 8;
+    // staging assignment: thread -> (mloc = tid%64, half j-rows j8 and j8+32)
+    const int mloc = tid & 63;
+    const int jbase = (tid >> 6) * 8;
 
+    const long mstep = (long)S * WBK;
     const long m_begin = (long)split * WBK;
-    for (long mt = m_begin; mt < M; mt += (long)S * WBK) {
-        long gm = mt + mloc;
-        // A: dy[m][ko0+j8..+7] -> lds_a[j][m]
-        bf16x8_t va = {};
-        if (gm < M && ko0 + j8 < cs.Ko)
-            va = *reinterpret_cast<const bf16x8_t*>(
-                &dy[gm * cs.Ko + ko0 + j8]);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) lds_a[j8 + j][mloc] = ((bf16*)&va)[j];
-        // B: im2col(x)[m][rc0+j8..+7] -> lds_b[j][m]
-        bf16x8_t vb = {};
-        int gk = rc0 + j8;
-        if (gm < M && gk < Kgemm) {
-            int q = gm % cs.Q; long rem = gm / cs.Q;
-            int p = rem % cs.P; int n = rem / cs.P;
-            int c = gk % cs.C, rs = gk / cs.C;
-            int r = rs / cs.S, s = rs % cs.S;
-            int h = p * cs.stride + r - cs.pad;
-            int wcol = q * cs.stride + s - cs.pad;
-            if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                vb = *reinterpret_cast<const bf16x8_t*>(
-                    &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
-        }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) lds_b[j8 + j][mloc] = ((bf16*)&vb)[j];
-        __syncthreads();
 
-        // MFMA: contraction over the 32 staged m's
-        bf16x8_t af[2], bfr[2];
-        const int kcol = (lane >> 4) * 8;
+    bf16x8_t va[2], vb[2];
+    auto gather = [&](long mt) {
+        long gm = mt + mloc;
         #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
-            af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
+        for (int half = 0; half < 2; ++half) {
+            int j8 = jbase + half * 32;
+            va[half] = bf16x8_t{};
+            vb[half] = bf16x8_t{};
+            if (gm < M) {
+                if (ko0 + j8 < cs.Ko)
+                    va[half] = *reinterpret_cast<const bf16x8_t*>(
+                        &dy[gm * cs.Ko + ko0 + j8]);
+                int gk = rc0 + j8;
+                if (gk < Kgemm) {
+                    int q = gm % cs.Q; long rem = gm / cs.Q;
+                    int p = rem % cs.P; int n = rem / cs.P;
+                    int c = gk % cs.C, rs = gk / cs.C;
+                    int r = rs / cs.S, s = rs % cs.S;
+                    int h = p * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + s - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        vb[half] = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+            }
+        }
+    };
+    auto store = [&](int buf) {
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
-            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+        for (int half = 0; half < 2; ++half) {
+            int j8 = jbase + half * 32;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                lds_a[buf][j8 + j][mloc] = ((bf16*)&va[half])[j];
+                lds_b[buf][j8 + j][mloc] = ((bf16*)&vb[half])[j];
+            }
+        }
+    };
+    auto compute = [&](int buf) {
         #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+        for (int ks = 0; ks < WBK; ks += 32) {
+            bf16x8_t af[2], bfr[2];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[buf][wr * 32 + mi * 16 + (lane & 15)][kcol]);
             #pragma unroll
             for (int ni = 0; ni < 2; ++ni)
-                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[buf][wc * 32 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 2; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
+    };
+
+    if (m_begin < M) {
+        gather(m_begin);
+        store(0);
         __syncthreads();
+        int buf = 0;
+        for (long mt = m_begin;;) {
+            long mt_next = mt + mstep;
+            bool more = mt_next < M;
+            if (more) gather(mt_next);
+            compute(buf);
+            if (!more) break;
+            store(buf ^ 1);
+            __syncthreads();
+            buf ^= 1;
+            mt = mt_next;
+        }
     }
 
     #pragma unroll
@@ -312,6 +349,76 @@ __global__ void k_wgrad_combine(const float* __restrict__ slab, int S,
         int rc = i % Kgemm;
         int c = rc % cs.C, rs = rc / cs.C;
         dw[((long)ko * cs.C + c) * RS + rs] = v;
+    }
+}
+
+// conv1 wgrad, fully specialized C==1 / 3x3 (the MNIST stem): each walker
+// wave sweeps a CONTIGUOUS output range with incremental (n,p,q) tracking —
+// no divisions in the inner loop — and a 3x3 sliding x-window in registers
+// (q+1 reuses 6 of 9 taps).  dy loads are lane-coalesced along Ko; x loads
+// are wave-uniform scalar broadcasts.  slab layout matches
+// k_wgrad_small_rsc ([k][Ko][split]) so the combine kernel is shared.
+__global__ void k_wgrad_c1_r3(const bf16* __restrict__ dy,
+                              const bf16* __restrict__ x,
+                              float* __restrict__ slab, ConvShape cs,
+                              long M, int S) {
+    const int ko = threadIdx.x & 63;
+    const int walker = threadIdx.x >> 6;          // 4 walker waves / block
+    const long wslot = (long)blockIdx.x * 4 + walker;
+    const long nslots = (long)S * 4;
+    const long chunk = (M + nslots - 1) / nslots;
+    long gm = wslot * chunk;
+    const long gm_end = min(M, gm + chunk);
+    float acc[9] = {};
+    if (gm < gm_end && ko < cs.Ko) {
+        int q = (int)(gm % cs.Q);
+        long rem = gm / cs.Q;
+        int p = (int)(rem % cs.P);
+        int n = (int)(rem / cs.P);
+        float xw[9];           // window: xw[r*3+s] = x[p*st+r-pad][q*st+s-pad]
+        auto ld = [&](int h, int wc) -> float {
+            return (h >= 0 && h < cs.H && wc >= 0 && wc < cs.W)
+                ? bf2f(x[((long)n * cs.H + h) * cs.W + wc]) : 0.f;
+        };
+        auto fill = [&]() {
+            #pragma unroll
+            for (int r = 0; r < 3; ++r)
+                #pragma unroll
+                for (int s = 0; s < 3; ++s)
+                    xw[r * 3 + s] = ld(p * cs.stride + r - cs.pad,
+                                       q * cs.stride + s - cs.pad);
+        };
+        fill();
+        for (;;) {
+            float g = bf2f(dy[gm * cs.Ko + ko]);
+            #pragma unroll
+            for (int k = 0; k < 9; ++k) acc[k] = fmaf(g, xw[k], acc[k]);
+            if (++gm >= gm_end) break;
+            if (++q == cs.Q) {
+                q = 0;
+                if (++p == cs.P) { p = 0; ++n; }
+                fill();
+            } else if (cs.stride == 1) {
+                #pragma unroll
+                for (int r = 0; r < 3; ++r) {
+                    xw[r * 3] = xw[r * 3 + 1];
+                    xw[r * 3 + 1] = xw[r * 3 + 2];
+                    xw[r * 3 + 2] = ld(p + r - cs.pad, q + 2 - cs.pad);
+                }
+            } else {
+                fill();
+            }
+        }
+    }
+    __shared__ float red[4][64];
+    for (int k = 0; k < 9; ++k) {
+        __syncthreads();
+        red[walker][ko] = acc[k];
+        __syncthreads();
+        if (walker == 0 && ko < cs.Ko) {
+            float t = red[0][ko] + red[1][ko] + red[2][ko] + red[3][ko];
+            slab[((long)k * cs.Ko + ko) * gridDim.x + blockIdx.x] = t;
+        }
     }
 }
 
@@ -525,6 +632,11 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
             int S_ = (int)std::max<long>(1, std::min<long>(2048, M / 8));
             auto slab = at::empty({rsc_total * cs.Ko, S_},
                                   x.options().dtype(at::kFloat));
+            if (cs.C == 1 && cs.R == 3 && cs.S == 3)
+                hipLaunchKernelGGL(k_wgrad_c1_r3, dim3(S_), dim3(256), 0,
+                                   stream.stream(), dyp, xp,
+                                   slab.data_ptr<float>(), cs, M, S_);
+            else
             hipLaunchKernelGGL(k_wgrad_small_rsc, dim3(S_), dim3(256), 0,
                                stream.stream(), dyp, xp,
                                slab.data_ptr<float>(), cs, M, S_, rsc_total);
