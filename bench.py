@@ -26,7 +26,7 @@ import torch.distributed as dist
 os.environ.setdefault("DDPX_NO_TQDM", "1")
 
 from ddp_tricks_amd import amp, same_seeds  # noqa: E402
-from ddp_tricks_amd.models.toy_net import Toy_Net  # noqa: E402
+from ddp_tricks_amd.models import build_model  # noqa: E402
 from ddp_tricks_amd.ops.functional import cross_entropy_loss  # noqa: E402
 from ddp_tricks_amd.ops.optim import FusedSGD  # noqa: E402
 from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP  # noqa: E402
@@ -34,32 +34,45 @@ from ddp_tricks_amd.utils.lookahead import Lookahead  # noqa: E402
 
 BASELINE_IMG_PER_SEC = 12000.0  # BASELINE.md implied train throughput
 
+# model -> (ctor kwargs, input CHW, n classes, per-GPU batch, dataset label)
+MODEL_CONFIGS = {
+    "toy_net": ({}, (1, 28, 28), 10, 1024, "MNIST(synthetic)"),
+    "resnet18": ({"num_classes": 10, "cifar_stem": True}, (3, 32, 32), 10,
+                 1024, "CIFAR-10(synthetic)"),
+    "resnet50": ({"num_classes": 1000}, (3, 224, 224), 1000, 256,
+                 "ImageNet(synthetic)"),
+}
+
 
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=1024,
-                   help="per-GPU batch size (reference default)")
+    p.add_argument("--batch-size", type=int, default=None,
+                   help="per-GPU batch size (default: model-specific)")
+    p.add_argument("--model", default="toy_net",
+                   choices=list(MODEL_CONFIGS),
+                   help="benchmark config (BASELINE.json); headline = toy_net")
     p.add_argument("--no-h2d", action="store_true",
                    help="skip the per-step H2D copy (ablation only)")
     return p.parse_args()
 
 
-class SyntheticMNIST:
-    """Pinned-host synthetic MNIST batches with one-ahead H2D prefetch on a
+class SyntheticData:
+    """Pinned-host synthetic image batches with one-ahead H2D prefetch on a
     copy stream (mirrors the training pipeline's prefetch)."""
 
-    def __init__(self, batch: int, device, n_buffers: int = 4, h2d: bool = True):
+    def __init__(self, batch: int, device, shape=(1, 28, 28), classes=10,
+                 n_buffers: int = 4, h2d: bool = True):
         g = torch.Generator().manual_seed(1234)
         self.h2d = h2d and device.type == "cuda"
         self.device = device
         self.host_images = []
         self.host_targets = []
         for _ in range(n_buffers):
-            img = torch.rand(batch, 1, 28, 28, generator=g)
-            tgt = torch.randint(0, 10, (batch,), generator=g)
+            img = torch.rand(batch, *shape, generator=g)
+            tgt = torch.randint(0, classes, (batch,), generator=g)
             if self.h2d:
                 img = img.pin_memory()
                 tgt = tgt.pin_memory()
@@ -119,15 +132,19 @@ def main():
         backend = "gloo"
     dist.init_process_group(backend=backend, init_method="env://")
 
+    kw, chw, classes, default_batch, dataset = MODEL_CONFIGS[args.model]
+    if args.batch_size is None:
+        args.batch_size = default_batch
     same_seeds(42)
-    model = Toy_Net().to(device)
+    model = build_model(args.model, **kw).to(device)
     optimizer = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, nesterov=True)
     lookahead = Lookahead(optimizer, k=10, alpha=0.5)
     model, apex_optimizer = amp.initialize(model, lookahead, opt_level="O1")
     ddp = DDP(model)
     ddp.train()
 
-    data = SyntheticMNIST(args.batch_size, device, h2d=not args.no_h2d)
+    data = SyntheticData(args.batch_size, device, shape=chw, classes=classes,
+                         h2d=not args.no_h2d)
 
     def step():
         apex_optimizer.zero_grad()
@@ -177,12 +194,13 @@ def main():
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": value / BASELINE_IMG_PER_SEC,
+            "vs_baseline": (value / BASELINE_IMG_PER_SEC
+                            if args.model == "toy_net" else None),
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "Toy_Net",
-                "dataset": "MNIST(synthetic)",
+                "model": "Toy_Net" if args.model == "toy_net" else args.model,
+                "dataset": dataset,
                 "global_batch": args.batch_size * world,
                 "per_gpu_batch": args.batch_size,
                 "seq_len": None,
