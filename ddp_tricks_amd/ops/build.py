@@ -21,6 +21,7 @@ BUILD_DIR = os.path.join(OPS_DIR, "build")
 
 SOURCES = [
     "bindings.cpp",
+    "comm.cpp",
     "elementwise.hip",
     "ce.hip",
     "pool.hip",
@@ -56,7 +57,7 @@ def _flags():
     dev = ["--offload-arch=gfx950", "-fno-gpu-rdc", "-Wno-unused-result"]
     ldflags = ([f"-L{d}" for d in lib]
                + ["-lc10", "-lc10_hip", "-ltorch", "-ltorch_cpu",
-                  "-ltorch_hip", "-ltorch_python", "-lamdhip64"]
+                  "-ltorch_hip", "-ltorch_python", "-lamdhip64", "-lrccl"]
                + [f"-Wl,-rpath,{d}" for d in lib])
     return cflags, dev, ldflags
 
@@ -73,6 +74,21 @@ def _stale(obj: str, src: str) -> bool:
 def build(verbose: bool = False):
     os.makedirs(BUILD_DIR, exist_ok=True)
     cflags, dev, ldflags = _flags()
+
+    # flag changes must invalidate every object, not just newer-than-source
+    import hashlib
+    sig = hashlib.sha1(" ".join(cflags + dev + ldflags).encode()).hexdigest()
+    sig_file = os.path.join(BUILD_DIR, "build_sig")
+    old_sig = None
+    if os.path.exists(sig_file):
+        with open(sig_file) as f:
+            old_sig = f.read().strip()
+    if old_sig != sig:
+        for f in os.listdir(BUILD_DIR):
+            if f.endswith(".o"):
+                os.remove(os.path.join(BUILD_DIR, f))
+        with open(sig_file, "w") as f:
+            f.write(sig)
 
     def compile_one(src_name: str) -> str:
         src = os.path.join(HIP_DIR, src_name)

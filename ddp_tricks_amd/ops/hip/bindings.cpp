@@ -57,7 +57,11 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C, long H,
 at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
                         long stride, long pad);
 
+// comm.cpp
+void register_comm(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    register_comm(m);
     m.def("multi_tensor_unscale", &multi_tensor_unscale,
           "fused unscale + inf check over grad tensors");
     m.def("fused_sgd", &fused_sgd, "fused nesterov-momentum SGD step",

@@ -74,6 +74,22 @@ class DistributedDataParallel(nn.Module):
         self._param_bucket = {}
         self._launched: List[_Bucket] = []
 
+        # Native RCCL communicator: rendezvous the ncclUniqueId
+        # over the torch TCP store, then talk to RCCL directly — bucket
+        # all-reduces and buffer broadcasts bypass ProcessGroupNCCL.
+        self._rccl = None
+        self._ext = None
+        if self._is_cuda and self.world_size > 1 \
+                and os.environ.get("DDPX_NATIVE_RCCL", "1") == "1":
+            from ..ops import load_extension
+            ext = load_extension()
+            if ext is not None and hasattr(ext, "rccl_unique_id"):
+                obj = [ext.rccl_unique_id() if self.rank == 0 else None]
+                dist.broadcast_object_list(obj, src=0, group=self.process_group)
+                self._rccl = ext.rccl_comm_init(self.world_size, self.rank,
+                                                obj[0])
+                self._ext = ext
+
         if self.world_size > 1:
             self._broadcast_params()
         self._coalesce_buffers()
@@ -86,6 +102,15 @@ class DistributedDataParallel(nn.Module):
 
     def _broadcast_params(self) -> None:
         with torch.no_grad():
+            if self._rccl is not None:
+                self._ext.rccl_group_start()
+                for p in self.module.parameters():
+                    self._ext.rccl_broadcast(p.data, 0, self._rccl)
+                for b in self.module.buffers():
+                    self._ext.rccl_broadcast(b.data, 0, self._rccl)
+                self._ext.rccl_group_end()
+                torch.cuda.synchronize(self._device)
+                return
             for p in self.module.parameters():
                 dist.broadcast(p.data, src=0, group=self.process_group)
             for b in self.module.buffers():
@@ -160,8 +185,13 @@ class DistributedDataParallel(nn.Module):
             ev.record(torch.cuda.current_stream(self._device))
             self._comm_stream.wait_event(ev)
             with torch.cuda.stream(self._comm_stream):
-                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                         group=self.process_group, async_op=True)
+                if self._rccl is not None:
+                    self._ext.rccl_all_reduce(b.flat, self._rccl, False)
+                    b.work = None
+                else:
+                    b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                             group=self.process_group,
+                                             async_op=True)
         else:
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                      group=self.process_group, async_op=True)
@@ -190,8 +220,14 @@ class DistributedDataParallel(nn.Module):
     def forward(self, *args, **kwargs):
         if (self.broadcast_buffers and self.world_size > 1
                 and self.module.training and self._buf_flats):
-            for flat in self._buf_flats:
-                dist.broadcast(flat, src=0, group=self.process_group)
+            if self._rccl is not None:
+                self._ext.rccl_group_start()
+                for flat in self._buf_flats:
+                    self._ext.rccl_broadcast(flat, 0, self._rccl)
+                self._ext.rccl_group_end()
+            else:
+                for flat in self._buf_flats:
+                    dist.broadcast(flat, src=0, group=self.process_group)
         return self.module(*args, **kwargs)
 
     # ------------------------------------------------------- passthrough ---

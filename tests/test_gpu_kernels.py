@@ -436,3 +436,25 @@ def test_bn_eval_residual():
     ref = (torch.nn.functional.batch_norm(x.float(), rm, rv, g, b, False,
                                           0.1, 1e-5) + res.float()).relu()
     _close(y, ref, name="bn eval resid")
+
+
+def test_conv_pad8_stem_path_isolated():
+    """The pad-to-8 MFMA stem path (functional.conv2d with C=3, 32x32) vs
+    torch fp32 — isolates the path the ResNet CIFAR stem uses so oracle
+    drift upstream can't hide a kernel bug."""
+    from ddp_tricks_amd.ops import functional as F_ops
+    torch.manual_seed(11)
+    x = torch.randn(32, 3, 32, 32, device=DEV).to(torch.bfloat16).float()
+    x.requires_grad_(True)
+    w = torch.randn(64, 3, 3, 3, device=DEV).to(torch.bfloat16).float()
+    w = torch.nn.Parameter(w)
+    y = F_ops.conv2d(x, w, None, stride=1, padding=1)
+    xr = x.detach().clone().requires_grad_(True)
+    wr = x.new_tensor(w.detach().cpu().numpy()).requires_grad_(True)
+    yr = torch.nn.functional.conv2d(xr, wr, None, 1, 1)
+    _close(y, yr, name="pad8 fwd")
+    dy = torch.randn_like(yr).to(torch.bfloat16).float()
+    y.backward(dy)
+    yr.backward(dy)
+    _close(w.grad, wr.grad, rel=2e-2, atol=0.5, name="pad8 wgrad")
+    _close(x.grad, xr.grad, rel=3e-2, atol=3e-2, name="pad8 dgrad")
