@@ -34,11 +34,19 @@ def iterate_loader(
     local_rank,
     apex_optimizer: Optional[torch.optim.Optimizer],
     training: bool = False,
+    timers=None,
 ):
     device = next(model.parameters()).device
     loss_sum = torch.zeros((), dtype=torch.float64, device=device)
     correct_sum = torch.zeros((), dtype=torch.long, device=device)
     num = 0
+
+    if timers is None and os.environ.get("DDPX_PHASE_TIMERS", "0") == "1":
+        from .timers import PhaseTimers
+        timers = iterate_loader._timers = getattr(
+            iterate_loader, "_timers", None) or PhaseTimers(device)
+    from contextlib import nullcontext
+    ph = timers.phase if timers is not None else (lambda name: nullcontext())
 
     iterator = loader
     if os.environ.get("DDPX_NO_TQDM", "0") != "1":
@@ -48,15 +56,19 @@ def iterate_loader(
     for image, target in iterator:
         if training:
             apex_optimizer.zero_grad()
-        image = image.to(device, non_blocking=True)
-        target = target.to(device, dtype=torch.long, non_blocking=True)
-        outputs = model(image)
-        batch_loss = loss_function(outputs, target)
-        batch_loss = batch_loss / outputs.shape[0]  # reference's double normalization
+        with ph("h2d"):
+            image = image.to(device, non_blocking=True)
+            target = target.to(device, dtype=torch.long, non_blocking=True)
+        with ph("fwd"):
+            outputs = model(image)
+            batch_loss = loss_function(outputs, target)
+            batch_loss = batch_loss / outputs.shape[0]  # reference's double normalization
         if training:
-            with amp.scale_loss(batch_loss, apex_optimizer) as scaled_loss:
-                scaled_loss.backward()
-            apex_optimizer.step()
+            with ph("bwd"):
+                with amp.scale_loss(batch_loss, apex_optimizer) as scaled_loss:
+                    scaled_loss.backward()
+            with ph("opt"):
+                apex_optimizer.step()
 
         # device-side metric accumulation (one host sync per epoch)
         loss_sum += batch_loss.detach().double() * image.shape[0]

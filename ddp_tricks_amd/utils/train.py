@@ -127,11 +127,17 @@ def train(args):
             tb.add_scalars("Loss", {"train": train_loss, "valid": valid_loss}, epoch)
             tb.add_scalars("Acc", {"train": train_acc, "valid": valid_acc}, epoch)
 
+        phase_info = ""
+        timers = getattr(iterate_loader, "_timers", None)
+        if timers is not None:  # DDPX_PHASE_TIMERS=1
+            phase_info = f", phases[{timers.format()}]"
+            timers.reset()
         print(f"epoch: {epoch:03d}/{args.epochs}, "
               f"time: {time.time() - epoch_start_time:.2f}s, "
               f"learning_rate: {curr_lr}, "
               f"train_loss: {train_loss:.4f}, train_acc: {train_acc:.4f}, "
-              f"valid_loss: {valid_loss:.4f}, valid_acc: {valid_acc:.4f}")
+              f"valid_loss: {valid_loss:.4f}, valid_acc: {valid_acc:.4f}"
+              f"{phase_info}")
 
         # scheduler gating exactly as the reference (utils/train.py:104-106)
         if epoch <= args.warmup_epochs:
