@@ -94,6 +94,20 @@ def maybe_sync_scaler() -> None:
         _state.overflow_count.zero_()
 
 
+def state_dict() -> dict:
+    """Serializable scaler state (checkpoint/resume — SURVEY §5.4 ext)."""
+    if _state.scaler is None:
+        return {}
+    sc = _state.scaler
+    return {"scale": sc.scale, "good_steps": sc._good_steps}
+
+
+def load_state_dict(sd: dict) -> None:
+    if _state.scaler is not None and sd:
+        _state.scaler.scale = sd["scale"]
+        _state.scaler._good_steps = sd["good_steps"]
+
+
 def is_enabled() -> bool:
     return _state.enabled
 

@@ -46,7 +46,7 @@ struct ConvShape {
 // that leaves LDS deterministic).
 __device__ __align__(16) unsigned char g_zero16[16];
 
-template <int MODE, int TBN, int WAVES_M, int WAVES_N>
+template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
@@ -88,6 +88,15 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
                         src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
+                } else if (STRIDE1) {
+                    int wcol = gm % cs.W; long rem = gm / cs.W;
+                    int h = rem % cs.H; int n = rem / cs.H;
+                    int ko = gk % cs.Ko, rs = gk / cs.Ko;
+                    int r = rs / cs.S, s = rs % cs.S;
+                    int p = h + cs.pad - r;
+                    int q = wcol + cs.pad - s;
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 } else {
                     int wcol = gm % cs.W; long rem = gm / cs.W;
                     int h = rem % cs.H; int n = rem / cs.H;
@@ -352,6 +361,89 @@ __global__ void k_wgrad_combine(const float* __restrict__ slab, int S,
     }
 }
 
+// Single-buffer 32-deep wgrad variant (the pre-pipelining shape; kept for
+// A/B selection via DDPX_WGRAD_V=sb — the profiler decides, not intuition).
+constexpr int SBK = 32;
+constexpr int SLDK = SBK + 8;
+
+__global__ __launch_bounds__(256)
+void k_conv_wgrad_sb(const bf16* __restrict__ dy, const bf16* __restrict__ x,
+                     float* __restrict__ slab, ConvShape cs, long M, int Kgemm,
+                     int S) {
+    __shared__ bf16 lds_a[WBM][SLDK];   // [ko][m]
+    __shared__ bf16 lds_b[WBN][SLDK];   // [rsc][m]
+    const int ko0 = blockIdx.x * WBM;
+    const int rc0 = blockIdx.y * WBN;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[2][2] = {};
+    const int mloc = tid & 31;
+    const int j8 = (tid >> 5) * 8;
+
+    const long m_begin = (long)split * SBK;
+    for (long mt = m_begin; mt < M; mt += (long)S * SBK) {
+        long gm = mt + mloc;
+        bf16x8_t va = {};
+        if (gm < M && ko0 + j8 < cs.Ko)
+            va = *reinterpret_cast<const bf16x8_t*>(
+                &dy[gm * cs.Ko + ko0 + j8]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) lds_a[j8 + j][mloc] = ((bf16*)&va)[j];
+        bf16x8_t vb = {};
+        int gk = rc0 + j8;
+        if (gm < M && gk < Kgemm) {
+            int q = gm % cs.Q; long rem = gm / cs.Q;
+            int p = rem % cs.P; int n = rem / cs.P;
+            int c = gk % cs.C, rs = gk / cs.C;
+            int r = rs / cs.S, s = rs % cs.S;
+            int h = p * cs.stride + r - cs.pad;
+            int wcol = q * cs.stride + s - cs.pad;
+            if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                vb = *reinterpret_cast<const bf16x8_t*>(
+                    &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) lds_b[j8 + j][mloc] = ((bf16*)&vb)[j];
+        __syncthreads();
+
+        bf16x8_t af[2], bfr[2];
+        const int kcol = (lane >> 4) * 8;
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+            int col = rc0 + wc * 32 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
 // conv1 wgrad, fully specialized C==1 / 3x3 (the MNIST stem): each walker
 // wave sweeps a CONTIGUOUS output range with incremental (n,p,q) tracking —
 // no divisions in the inner loop — and a 3x3 sliding x-window in registers
@@ -600,14 +692,24 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
     if (cs.C >= 128) {
         dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
-        hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2>), grid, dim3(256), 0,
-                           stream.stream(), dyp_, wt2p, nullptr, dxp, cs,
-                           (int)M, Kgemm, cs.C);
+        if (cs.stride == 1)
+            hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+        else
+            hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, false>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
     } else {
         dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
-        hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1>), grid, dim3(256), 0,
-                           stream.stream(), dyp_, wt2p, nullptr, dxp, cs,
-                           (int)M, Kgemm, cs.C);
+        if (cs.stride == 1)
+            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+        else
+            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, false>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
     }
     HIP_CHECK_LAST();
     return dx;
@@ -666,14 +768,22 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     }
 
     TORCH_CHECK(cs.C % 8 == 0 && cs.Ko % 8 == 0);
+    static const char* wv = getenv("DDPX_WGRAD_V");
+    const bool use_sb = wv && wv[0] == 's';
+    const int depth = use_sb ? SBK : WBK;
     int gk = ceil_div_i(cs.Ko, WBM), gr = ceil_div_i(Kgemm, WBN);
     int S_ = 1;
-    while (gk * gr * S_ < 1024 && S_ < 64 && (M / (S_ * 2)) >= WBK) S_ *= 2;
+    while (gk * gr * S_ < 1024 && S_ < 64 && (M / (S_ * 2)) >= depth) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
-    hipLaunchKernelGGL(k_conv_wgrad, dim3(gk, gr, S_), dim3(256), 0,
-                       stream.stream(), dyp, xp, slab.data_ptr<float>(), cs,
-                       M, Kgemm, S_);
+    if (use_sb)
+        hipLaunchKernelGGL(k_conv_wgrad_sb, dim3(gk, gr, S_), dim3(256), 0,
+                           stream.stream(), dyp, xp, slab.data_ptr<float>(),
+                           cs, M, Kgemm, S_);
+    else
+        hipLaunchKernelGGL(k_conv_wgrad, dim3(gk, gr, S_), dim3(256), 0,
+                           stream.stream(), dyp, xp, slab.data_ptr<float>(),
+                           cs, M, Kgemm, S_);
     HIP_CHECK_LAST();
     long total = (long)cs.Ko * Kgemm;
     int blocks = std::min<long>(4096, ceil_div_i(total, 256));

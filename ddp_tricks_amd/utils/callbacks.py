@@ -64,3 +64,12 @@ class EarlyStopping:
         else:
             self.best_score = score
             self.counter = 0
+
+    def state_dict(self):
+        return {"counter": self.counter, "best_score": self.best_score,
+                "early_stop": self.early_stop}
+
+    def load_state_dict(self, state):
+        self.counter = state["counter"]
+        self.best_score = state["best_score"]
+        self.early_stop = state["early_stop"]

@@ -88,6 +88,78 @@ class MNIST(Dataset):
         return self.images[idx], int(self.labels[idx])
 
 
+class CIFAR10(Dataset):
+    """CIFAR-10 (synthetic stand-in when the binary batches are absent —
+    no network in this environment).  Returns (float32 [3,32,32] in [0,1],
+    int64 label); BASELINE.json config 4."""
+
+    def __init__(self, root: str = "./datasets/", train: bool = True,
+                 download: bool = False, synthetic: Optional[bool] = None,
+                 num_samples: Optional[int] = None):
+        batch_dir = os.path.join(root, "cifar-10-batches-py")
+        self.synthetic = synthetic if synthetic is not None \
+            else not os.path.isdir(batch_dir)
+        n_default = 50000 if train else 10000
+        if not self.synthetic:
+            import pickle
+            names = [f"data_batch_{i}" for i in range(1, 6)] if train \
+                else ["test_batch"]
+            imgs, lbls = [], []
+            for nm in names:
+                with open(os.path.join(batch_dir, nm), "rb") as f:
+                    d = pickle.load(f, encoding="bytes")
+                imgs.append(np.asarray(d[b"data"], dtype=np.float32) / 255.0)
+                lbls.extend(d[b"labels"])
+            self.images = torch.from_numpy(
+                np.concatenate(imgs).reshape(-1, 3, 32, 32))
+            self.labels = torch.tensor(lbls, dtype=torch.int64)
+        else:
+            env_cap = os.environ.get("DDPX_SYNTH_SAMPLES")
+            if num_samples is None and env_cap:
+                num_samples = min(int(env_cap), n_default)
+            n = num_samples or n_default
+            g = torch.Generator().manual_seed(777 if train else 778)
+            labels = torch.randint(0, 10, (n,), generator=g)
+            templates = torch.rand(10, 3, 32, 32, generator=g)
+            noise = torch.rand(n, 3, 32, 32, generator=g)
+            self.images = (0.6 * templates[labels] + 0.4 * noise).clamp_(0, 1)
+            self.labels = labels
+
+    def __len__(self) -> int:
+        return self.labels.shape[0]
+
+    def __getitem__(self, idx: int):
+        return self.images[idx], int(self.labels[idx])
+
+
+class SyntheticImageNet(Dataset):
+    """Synthetic ImageNet-shaped data ([3,224,224], 1000 classes) for the
+    ResNet-50 config (BASELINE.json config 5) — generated per-index so the
+    full 1.28M-image epoch never materializes in host memory."""
+
+    def __init__(self, root: str = "./datasets/", train: bool = True,
+                 download: bool = False, num_samples: Optional[int] = None):
+        n_default = 10240 if train else 2048
+        env_cap = os.environ.get("DDPX_SYNTH_SAMPLES")
+        if num_samples is None and env_cap:
+            num_samples = int(env_cap)
+        self.n = num_samples or n_default
+        self.train = train
+
+    def __len__(self) -> int:
+        return self.n
+
+    def __getitem__(self, idx: int):
+        g = torch.Generator().manual_seed((idx << 1) | (1 if self.train else 0))
+        label = int(torch.randint(0, 1000, (1,), generator=g))
+        img = torch.rand(3, 224, 224, generator=g)
+        return img, label
+
+
+DATASETS = {"mnist": MNIST, "cifar10": CIFAR10,
+            "imagenet_synthetic": SyntheticImageNet}
+
+
 class DistributedSampler(torch.utils.data.Sampler):
     """torch.utils.data.distributed.DistributedSampler semantics
     (seeded shuffle, pad to divisible, set_epoch) re-implemented."""

@@ -24,11 +24,11 @@ import torch.distributed as torch_dist
 from torch.utils.data import DataLoader
 
 from .. import amp
-from ..models.toy_net import Toy_Net
+from ..models import build_model
 from ..ops.optim import FusedSGD
 from ..parallel.ddp import DistributedDataParallel as DDP
 from .callbacks import EarlyStopping, same_seeds
-from .data import MNIST, DistributedSampler, CudaPrefetcher
+from .data import DATASETS, DistributedSampler, CudaPrefetcher
 from .engine import iterate_loader
 from .lookahead import Lookahead
 from .schedulers import ReduceLROnPlateau, WarmupLambdaLR
@@ -47,15 +47,26 @@ def train(args):
     local_rank = getattr(args, "local_rank", 0) or 0
     device = _resolve_device(local_rank)
 
+    # Extensions over the reference's 10-flag contract:
+    # --model / --dataset select from the registries; --resume restarts
+    # from the sidecar checkpoint.  Defaults reproduce the reference.
+    model_name = getattr(args, "model", None) or "toy_net"
+    dataset_name = getattr(args, "dataset", None) or "mnist"
+    model_kwargs = dict(getattr(args, "model_kwargs", None) or {})
+    if model_name.startswith("resnet") and dataset_name == "cifar10":
+        model_kwargs.setdefault("num_classes", 10)
+        model_kwargs.setdefault("cifar_stem", True)
+    Dataset = DATASETS[dataset_name]
+
     # Data pipeline (reference utils/train.py:24-30: sharded train loader,
     # UNSHARDED valid loader evaluated in full on every rank)
-    train_set = MNIST(root=args.data_path, train=True, download=True)
+    train_set = Dataset(root=args.data_path, train=True, download=True)
     train_sampler = DistributedSampler(train_set)
     same_seeds(args.seed_num)
     train_loader = DataLoader(train_set, batch_size=args.batch_size,
                               shuffle=False, pin_memory=device.type == "cuda",
                               sampler=train_sampler)
-    valid_set = MNIST(root=args.data_path, train=False, download=True)
+    valid_set = Dataset(root=args.data_path, train=False, download=True)
     valid_loader = DataLoader(valid_set, batch_size=args.batch_size,
                               shuffle=False, pin_memory=device.type == "cuda")
     if device.type == "cuda":
@@ -67,7 +78,7 @@ def train(args):
     # Model (seeded identically on all ranks before the DDP broadcast —
     # reference utils/train.py:34-36, README rationale)
     same_seeds(args.seed_num)
-    model = Toy_Net()
+    model = build_model(model_name, **model_kwargs)
     model = model.to(device)
 
     os.makedirs(os.path.join(args.model_path, "logs"), exist_ok=True)
@@ -102,10 +113,30 @@ def train(args):
                                            opt_level="O1")
     parallel_model = DDP(model)
 
+    # Resume from the sidecar checkpoint (extension; the reference saves
+    # best weights only and has no resume path — SURVEY §5.4).  Every rank
+    # loads the same file, so state stays rank-consistent.
+    start_epoch = 0
+    resume_path = f"{latest_model_path}.resume.pt"
+    resume_on = bool(getattr(args, "resume", False))
+    if resume_on and os.path.exists(resume_path):
+        ck = torch.load(resume_path, map_location=device, weights_only=False)
+        parallel_model.module.load_state_dict(ck["model"])
+        apex_optimizer.load_state_dict(ck["optimizer"])
+        scheduler_wu.load_state_dict(ck["scheduler_wu"])
+        scheduler_re.load_state_dict(ck["scheduler_re"])
+        early_stopping.load_state_dict(ck["early_stopping"])
+        amp.load_state_dict(ck.get("amp", {}))
+        best_valid_acc = ck["best_valid_acc"]
+        start_epoch = ck["epoch"] + 1
+        from ..ops.functional import clear_weight_cache
+        clear_weight_cache()
+        print(f"resumed {args.exp_name} at epoch {start_epoch}")
+
     if local_rank == 0:
         tb = SummaryWriter(os.path.join(args.model_path, "logs", args.exp_name))
 
-    for epoch in range(args.epochs):
+    for epoch in range(start_epoch, args.epochs):
         epoch_start_time = time.time()
         train_sampler.set_epoch(epoch)
 
@@ -151,6 +182,18 @@ def train(args):
             best_valid_acc = valid_acc
             torch.save(parallel_model.module.state_dict(),
                        f"{latest_model_path}.pt")
+
+        if local_rank == 0 and resume_on:
+            torch.save({
+                "epoch": epoch,
+                "model": parallel_model.module.state_dict(),
+                "optimizer": apex_optimizer.state_dict(),
+                "scheduler_wu": scheduler_wu.state_dict(),
+                "scheduler_re": scheduler_re.state_dict(),
+                "early_stopping": early_stopping.state_dict(),
+                "amp": amp.state_dict(),
+                "best_valid_acc": best_valid_acc,
+            }, resume_path)
 
     if local_rank == 0:
         tb.close()

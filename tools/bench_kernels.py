@@ -1,0 +1,82 @@
+"""Micro-benchmark of the HIP conv kernels at the workload's shapes.
+
+Usage (on a GPU box):  python tools/bench_kernels.py [--resnet]
+
+Times conv fwd/dgrad/wgrad per shape with HIP events (200 reps after 20
+warmup) and prints achieved TFLOP/s next to microseconds, so kernel A/B
+decisions are measurements, not guesses.  DDPX_WGRAD_V=sb selects the
+single-buffer wgrad variant for comparison.
+"""
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
+
+from ddp_tricks_amd.ops import load_extension  # noqa: E402
+
+CL = torch.channels_last
+
+TOYNET = [
+    # (name, N, C, H, W, K, R, stride, pad)
+    ("conv2", 1024, 64, 26, 26, 128, 3, 1, 0),
+    ("conv3", 1024, 128, 12, 12, 256, 3, 1, 0),
+    ("conv4", 1024, 256, 10, 10, 512, 3, 1, 0),
+    ("conv1", 1024, 1, 28, 28, 64, 3, 1, 0),
+]
+RESNET = [
+    ("r18s1", 256, 64, 32, 32, 64, 3, 1, 1),
+    ("r18s2", 256, 64, 32, 32, 128, 3, 2, 1),
+    ("r18d2", 256, 64, 32, 32, 128, 1, 2, 0),
+    ("r50c1", 256, 64, 56, 56, 64, 1, 1, 0),
+    ("r50c3", 256, 64, 56, 56, 256, 1, 1, 0),
+    ("r50l3", 256, 256, 14, 14, 1024, 1, 1, 0),
+    ("stem50", 64, 8, 224, 224, 64, 7, 2, 3),
+]
+
+
+def time_op(fn, reps=200, warmup=20):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    s = torch.cuda.Event(enable_timing=True)
+    e = torch.cuda.Event(enable_timing=True)
+    s.record()
+    for _ in range(reps):
+        fn()
+    e.record()
+    torch.cuda.synchronize()
+    return s.elapsed_time(e) / reps * 1000.0  # us
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--resnet", action="store_true")
+    ap.add_argument("--reps", type=int, default=200)
+    args = ap.parse_args()
+    ext = load_extension(required=True)
+    dev = torch.device("cuda:0")
+    shapes = TOYNET + (RESNET if args.resnet else [])
+    print(f"{'shape':8s} {'op':6s} {'us':>9s} {'TFLOP/s':>9s}")
+    for name, N, C, H, W, K, R, stride, pad in shapes:
+        P = (H + 2 * pad - R) // stride + 1
+        x = torch.randn(N, C, H, W, device=dev).to(torch.bfloat16).contiguous(memory_format=CL)
+        w = torch.randn(K, C, R, R, device=dev).to(torch.bfloat16).contiguous(memory_format=CL)
+        dy = torch.randn(N, K, P, P, device=dev).to(torch.bfloat16).contiguous(memory_format=CL)
+        wt2 = w.to(torch.bfloat16).permute(1, 2, 3, 0).reshape(C, R * R * K).contiguous()
+        flops = 2.0 * N * P * P * K * C * R * R
+        us = time_op(lambda: ext.conv2d_fwd(x, w, None, stride, pad), args.reps)
+        print(f"{name:8s} {'fwd':6s} {us:9.1f} {flops/us/1e6:9.1f}")
+        if C >= 8:
+            us = time_op(lambda: ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R,
+                                                  stride, pad), args.reps)
+            print(f"{name:8s} {'dgrad':6s} {us:9.1f} {flops/us/1e6:9.1f}")
+        us = time_op(lambda: ext.conv2d_wgrad(dy, x, R, R, stride, pad),
+                     args.reps)
+        print(f"{name:8s} {'wgrad':6s} {us:9.1f} {flops/us/1e6:9.1f}")
+
+
+if __name__ == "__main__":
+    main()
