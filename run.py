@@ -38,6 +38,15 @@ def parse_args():
                         type=str, help="path of model")
     parser.add_argument("--local_rank", type=int, default=None,
                         help="local rank for DistributedDataParallel")
+    # extensions beyond the reference's 10-flag contract
+    parser.add_argument("--model", default="toy_net", type=str,
+                        help="model registry name (toy_net, resnet18, "
+                             "resnet34, resnet50)")
+    parser.add_argument("--dataset", default="mnist", type=str,
+                        help="dataset registry name (mnist, cifar10, "
+                             "imagenet_synthetic)")
+    parser.add_argument("--resume", action="store_true",
+                        help="save/load the sidecar resume checkpoint")
     args = parser.parse_args()
     if args.local_rank is None:
         args.local_rank = int(os.environ.get("LOCAL_RANK", 0))

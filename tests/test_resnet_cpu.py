@@ -72,3 +72,32 @@ def test_global_avgpool_cpu():
     from ddp_tricks_amd.ops.functional import global_avg_pool2d
     x = torch.randn(2, 4, 7, 7)
     assert torch.allclose(global_avg_pool2d(x), x.mean(dim=(2, 3)))
+
+
+def test_train_resnet18_cifar10_cpu(tmp_path, monkeypatch):
+    """train(args) with --model resnet18 --dataset cifar10 (config 4 path)."""
+    import os
+    import types
+
+    import torch.distributed as dist
+
+    from ddp_tricks_amd import amp
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "128")
+    monkeypatch.setenv("DDPX_NO_TQDM", "1")
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29682")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    amp._state.__init__()
+    from ddp_tricks_amd.utils.train import train
+    args = types.SimpleNamespace(
+        exp_name="R18", learning_rate=0.05, batch_size=64, epochs=1,
+        warmup_epochs=1, warmup_type="linear", seed_num=42,
+        data_path="/nonexistent", model_path=str(tmp_path), local_rank=0,
+        model="resnet18", dataset="cifar10")
+    train(args)
+    import torch
+    sd = torch.load(os.path.join(tmp_path, "R18.pt"), weights_only=True)
+    assert "layer4.1.bn2.running_var" in sd and sd["fc.weight"].shape == (10, 512)
