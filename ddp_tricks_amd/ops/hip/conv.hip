@@ -768,8 +768,11 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     }
 
     TORCH_CHECK(cs.C % 8 == 0 && cs.Ko % 8 == 0);
+    // measured on MI355X (profiles/, tools/bench_kernels.py): the single-
+    // buffer 32-deep variant beats the double-buffered 64-deep one at every
+    // Toy_Net shape (e.g. conv4 678 vs 856 us) — default sb, env overrides
     static const char* wv = getenv("DDPX_WGRAD_V");
-    const bool use_sb = wv && wv[0] == 's';
+    const bool use_sb = !(wv && wv[0] == 'p');
     const int depth = use_sb ? SBK : WBK;
     int gk = ceil_div_i(cs.Ko, WBM), gr = ceil_div_i(Kgemm, WBN);
     int S_ = 1;
