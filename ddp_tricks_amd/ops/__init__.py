@@ -29,6 +29,28 @@ def _package_dir() -> str:
     return os.path.dirname(os.path.abspath(__file__))
 
 
+class _SyncDebugProxy:
+    """DDPX_SYNC_DEBUG=1: every extension call is followed by a
+    device synchronize + error check, so an async kernel fault surfaces at
+    the faulting op's Python call site instead of a later sync point."""
+
+    def __init__(self, ext):
+        self._ext = ext
+
+    def __getattr__(self, name):
+        fn = getattr(self._ext, name)
+        if not callable(fn):
+            return fn
+
+        def wrapped(*args, **kwargs):
+            import torch
+            out = fn(*args, **kwargs)
+            torch.cuda.synchronize()
+            return out
+
+        return wrapped
+
+
 def load_extension(required: bool = False):
     """Load the in-tree _hip_ops extension; cache the module object."""
     global _EXT, _EXT_TRIED
@@ -43,6 +65,8 @@ def load_extension(required: bool = False):
     try:
         import torch  # noqa: F401  (extension links against torch libs)
         _EXT = importlib.import_module("_hip_ops")
+        if os.environ.get("DDPX_SYNC_DEBUG", "0") == "1":
+            _EXT = _SyncDebugProxy(_EXT)
     except ImportError as e:
         _EXT = None
         if required:

@@ -444,6 +444,109 @@ void k_conv_wgrad_sb(const bf16* __restrict__ dy, const bf16* __restrict__ x,
         }
 }
 
+// Wide wgrad variant: 128(ko) x 128(rsc) tile, 32-deep contraction — four
+// times the MFMA work per barrier pair of the 64x64 tile, same per-thread
+// staging cost per element (2 transposed bf16x8 stores per operand).
+// Requires Ko >= 128 and Kgemm >= 128 (host falls back to sb otherwise).
+__global__ __launch_bounds__(256)
+void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
+                       float* __restrict__ slab, ConvShape cs, long M,
+                       int Kgemm, int S) {
+    __shared__ bf16 lds_a[128][SLDK];   // [ko][m]
+    __shared__ bf16 lds_b[128][SLDK];   // [rsc][m]
+    const int ko0 = blockIdx.x * 128;
+    const int rc0 = blockIdx.y * 128;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[4][4] = {};
+    const int mloc = tid & 31;
+    const int j8 = (tid >> 5) * 8;
+
+    const long m_begin = (long)split * SBK;
+    for (long mt = m_begin; mt < M; mt += (long)S * SBK) {
+        long gm = mt + mloc;
+        const bool valid = gm < M;
+        #pragma unroll
+        for (int half = 0; half < 2; ++half) {
+            const int j = j8 + half * 64;
+            bf16x8_t va = {};
+            if (valid && ko0 + j < cs.Ko)
+                va = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[gm * cs.Ko + ko0 + j]);
+            #pragma unroll
+            for (int jj = 0; jj < 8; ++jj)
+                lds_a[j + jj][mloc] = ((bf16*)&va)[jj];
+        }
+        if (valid) {
+            int q = gm % cs.Q; long rem = gm / cs.Q;
+            int p = rem % cs.P; int n = rem / cs.P;
+            #pragma unroll
+            for (int half = 0; half < 2; ++half) {
+                const int j = j8 + half * 64;
+                bf16x8_t vb = {};
+                int gk = rc0 + j;
+                if (gk < Kgemm) {
+                    int c = gk % cs.C, rs = gk / cs.C;
+                    int r = rs / cs.S, s = rs % cs.S;
+                    int h = p * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + s - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        vb = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+                #pragma unroll
+                for (int jj = 0; jj < 8; ++jj)
+                    lds_b[j + jj][mloc] = ((bf16*)&vb)[jj];
+            }
+        } else {
+            #pragma unroll
+            for (int half = 0; half < 2; ++half) {
+                const int j = j8 + half * 64;
+                #pragma unroll
+                for (int jj = 0; jj < 8; ++jj)
+                    lds_b[j + jj][mloc] = (bf16)0;
+            }
+        }
+        __syncthreads();
+
+        bf16x8_t af[4], bfr[4];
+        const int kcol = (lane >> 4) * 8;
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = rc0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
 // conv1 wgrad, fully specialized C==1 / 3x3 (the MNIST stem): each walker
 // wave sweeps a CONTIGUOUS output range with incremental (n,p,q) tracking —
 // no divisions in the inner loop — and a 3x3 sliding x-window in registers
@@ -768,18 +871,28 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     }
 
     TORCH_CHECK(cs.C % 8 == 0 && cs.Ko % 8 == 0);
-    // measured on MI355X (profiles/, tools/bench_kernels.py): the single-
-    // buffer 32-deep variant beats the double-buffered 64-deep one at every
-    // Toy_Net shape (e.g. conv4 678 vs 856 us) — default sb, env overrides
+    // Variant choice is MEASURED (profiles/, tools/bench_kernels.py): the
+    // 128x128 wide tile wins when it fits (4x MFMA work per barrier), the
+    // single-buffer 32-deep 64x64 tile beats the double-buffered pipelined
+    // one everywhere else.  DDPX_WGRAD_V = wide|sb|pipe overrides.
     static const char* wv = getenv("DDPX_WGRAD_V");
-    const bool use_sb = !(wv && wv[0] == 'p');
-    const int depth = use_sb ? SBK : WBK;
-    int gk = ceil_div_i(cs.Ko, WBM), gr = ceil_div_i(Kgemm, WBN);
+    const char sel = wv ? wv[0] : 0;
+    const bool can_wide = cs.Ko >= 128 && Kgemm >= 128;
+    const bool use_wide = sel ? sel == 'w' : can_wide;
+    const bool use_sb = sel ? sel == 's' : !can_wide;
+    TORCH_CHECK(!use_wide || can_wide, "wide wgrad needs Ko,Kgemm >= 128");
+    const int depth = (use_wide || use_sb) ? SBK : WBK;
+    const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
+    int gk = ceil_div_i(cs.Ko, tm), gr = ceil_div_i(Kgemm, tn);
     int S_ = 1;
     while (gk * gr * S_ < 1024 && S_ < 64 && (M / (S_ * 2)) >= depth) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
-    if (use_sb)
+    if (use_wide)
+        hipLaunchKernelGGL(k_conv_wgrad_wide, dim3(gk, gr, S_), dim3(256), 0,
+                           stream.stream(), dyp, xp, slab.data_ptr<float>(),
+                           cs, M, Kgemm, S_);
+    else if (use_sb)
         hipLaunchKernelGGL(k_conv_wgrad_sb, dim3(gk, gr, S_), dim3(256), 0,
                            stream.stream(), dyp, xp, slab.data_ptr<float>(),
                            cs, M, Kgemm, S_);

@@ -25,6 +25,7 @@ TOYNET = [
     ("conv3", 1024, 128, 12, 12, 256, 3, 1, 0),
     ("conv4", 1024, 256, 10, 10, 512, 3, 1, 0),
     ("conv1", 1024, 1, 28, 28, 64, 3, 1, 0),
+    ("conv1p8", 1024, 8, 28, 28, 64, 3, 1, 0),
 ]
 RESNET = [
     ("r18s1", 256, 64, 32, 32, 64, 3, 1, 1),
