@@ -109,17 +109,29 @@ class _HIPConv2d(torch.autograd.Function):
     def forward(ctx, x, weight, bias, stride, padding):
         ext = require_ext_for(x)
         xb = _chlast(_to_bf16(x))
-        pad8 = _use_pad8(weight.shape[1], xb.shape[2], xb.shape[3])
-        if pad8:
-            xb = _pad8(xb)
+        C_in = weight.shape[1]
+        if C_in < 8:
+            # fwd always takes the MFMA path on zero-padded channels
+            # (measured faster than the direct VALU kernel even at MNIST
+            # size: 48 vs 64 us).  wgrad for SMALL spatial inputs stays on
+            # the specialized C<8 kernels (measured: MFMA wgrad at Kgemm=72
+            # is 362 us vs 132 us for the sliding-window kernel), so the
+            # UNPADDED x is saved in that case.
+            pad8_bwd = _use_pad8(C_in, xb.shape[2], xb.shape[3])
+            xp = _pad8(xb)
             wb = weight_variant(weight, "nhwc_p8")
-        else:
-            wb = weight_variant(weight, "nhwc")
+            y = ext.conv2d_fwd(xp, wb, bias if bias is None
+                               else bias.detach().float(), stride, padding)
+            ctx.save_for_backward(xp if pad8_bwd else xb, weight)
+            ctx.meta = (stride, padding, x.dtype, weight.shape,
+                        bias is not None, pad8_bwd)
+            return y
+        wb = weight_variant(weight, "nhwc")
         bias_f = bias if bias is None else bias.detach().float()
         y = ext.conv2d_fwd(xb, wb, bias_f, stride, padding)
         ctx.save_for_backward(xb, weight)
         ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
-                    pad8)
+                    False)
         return y
 
     @staticmethod

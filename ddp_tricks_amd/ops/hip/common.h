@@ -76,6 +76,40 @@ DEV_INLINE long long lds_block_reduce_ll(long long v, long long* scratch) {
     return out;  // valid in thread 0
 }
 
+// Fast unsigned division by a runtime constant (libdivide-style
+// multiply-shift, set up once on the host).  The conv gathers decode
+// (m -> n,p,q) and (k -> r,s,c) per staged element; s_div emulation costs
+// tens of VALU cycles each, v_mul_hi_u32 + shift costs ~4.
+// Powers of two (incl. 1) use a plain shift (mul == 0 flags it); otherwise
+// the Granlund–Montgomery round-up form: shift = floor(log2 d),
+// mul = floor(2^(32+shift)/d) + 1, q = (n*mul) >> (32+shift) — exact for
+// n < 2^31 (all index spaces here are far below that).  d is wave-uniform
+// so the mul==0 branch is a scalar branch.
+struct FastDiv {
+    unsigned mul;
+    unsigned shift;
+    unsigned d;
+    __host__ void init(unsigned div) {
+        d = div;
+        if ((div & (div - 1)) == 0) {   // power of two (incl. 1)
+            mul = 0;
+            shift = 0;
+            while ((1u << shift) < div) ++shift;
+            return;
+        }
+        shift = 0;
+        while ((2u << shift) < div) ++shift;   // floor(log2(div))
+        mul = (unsigned)(((1ull << (32 + shift)) / div) + 1);
+    }
+};
+DEV_INLINE unsigned fd_div(unsigned n, FastDiv f) {
+    if (f.mul == 0) return n >> f.shift;
+    return (unsigned)(((unsigned long long)n * f.mul) >> 32) >> f.shift;
+}
+DEV_INLINE unsigned fd_mod(unsigned n, FastDiv f, unsigned q) {
+    return n - q * f.d;
+}
+
 #define HIP_CHECK_LAST()                                                     \
     do {                                                                     \
         hipError_t e = hipGetLastError();                                    \

@@ -25,7 +25,15 @@ struct ConvShape {
     int Ko, R, S;       // weights
     int P, Q;           // output
     int stride, pad;
+    // magic-division decoders for the gathers (set on host)
+    FastDiv fdQ, fdP, fdW, fdH, fdC, fdS, fdKo;
 };
+
+static inline void init_fastdiv(ConvShape& cs) {
+    cs.fdQ.init(cs.Q); cs.fdP.init(cs.P); cs.fdW.init(cs.W);
+    cs.fdH.init(cs.H); cs.fdC.init(cs.C); cs.fdS.init(cs.S);
+    cs.fdKo.init(cs.Ko);
+}
 
 // ---------------------------------------------------------------- forward ---
 
@@ -80,28 +88,40 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
             if (gm < M && gk < Kgemm) {
                 if (MODE == 0) {
-                    int q = gm % cs.Q, rem = gm / cs.Q;
-                    int p = rem % cs.P, n = rem / cs.P;
-                    int c = gk % cs.C, rs = gk / cs.C;
-                    int r = rs / cs.S, s = rs % cs.S;
+                    unsigned rem = fd_div(gm, cs.fdQ);
+                    int q = fd_mod(gm, cs.fdQ, rem);
+                    unsigned n = fd_div(rem, cs.fdP);
+                    int p = fd_mod(rem, cs.fdP, n);
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
                     int h = p * cs.stride + r - cs.pad;
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
                         src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
                 } else if (STRIDE1) {
-                    int wcol = gm % cs.W; long rem = gm / cs.W;
-                    int h = rem % cs.H; int n = rem / cs.H;
-                    int ko = gk % cs.Ko, rs = gk / cs.Ko;
-                    int r = rs / cs.S, s = rs % cs.S;
+                    unsigned rem = fd_div(gm, cs.fdW);
+                    int wcol = fd_mod(gm, cs.fdW, rem);
+                    unsigned n = fd_div(rem, cs.fdH);
+                    int h = fd_mod(rem, cs.fdH, n);
+                    unsigned rs = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
                     int p = h + cs.pad - r;
                     int q = wcol + cs.pad - s;
                     if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
                         src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 } else {
-                    int wcol = gm % cs.W; long rem = gm / cs.W;
-                    int h = rem % cs.H; int n = rem / cs.H;
-                    int ko = gk % cs.Ko, rs = gk / cs.Ko;
-                    int r = rs / cs.S, s = rs % cs.S;
+                    unsigned rem = fd_div(gm, cs.fdW);
+                    int wcol = fd_mod(gm, cs.fdW, rem);
+                    unsigned n = fd_div(rem, cs.fdH);
+                    int h = fd_mod(rem, cs.fdH, n);
+                    unsigned rs = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
                     int pn = h + cs.pad - r;
                     int qn = wcol + cs.pad - s;
                     // strided conv: only taps where stride divides contribute
@@ -263,10 +283,14 @@ void k_conv_wgrad(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                         &dy[gm * cs.Ko + ko0 + j8]);
                 int gk = rc0 + j8;
                 if (gk < Kgemm) {
-                    int q = gm % cs.Q; long rem = gm / cs.Q;
-                    int p = rem % cs.P; int n = rem / cs.P;
-                    int c = gk % cs.C, rs = gk / cs.C;
-                    int r = rs / cs.S, s = rs % cs.S;
+                    unsigned rem = fd_div((unsigned)gm, cs.fdQ);
+                    int q = fd_mod((unsigned)gm, cs.fdQ, rem);
+                    unsigned n = fd_div(rem, cs.fdP);
+                    int p = fd_mod(rem, cs.fdP, n);
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
                     int h = p * cs.stride + r - cs.pad;
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
@@ -395,10 +419,14 @@ void k_conv_wgrad_sb(const bf16* __restrict__ dy, const bf16* __restrict__ x,
         bf16x8_t vb = {};
         int gk = rc0 + j8;
         if (gm < M && gk < Kgemm) {
-            int q = gm % cs.Q; long rem = gm / cs.Q;
-            int p = rem % cs.P; int n = rem / cs.P;
-            int c = gk % cs.C, rs = gk / cs.C;
-            int r = rs / cs.S, s = rs % cs.S;
+            unsigned rem = fd_div((unsigned)gm, cs.fdQ);
+            int q = fd_mod((unsigned)gm, cs.fdQ, rem);
+            unsigned n = fd_div(rem, cs.fdP);
+            int p = fd_mod(rem, cs.fdP, n);
+            unsigned rs = fd_div(gk, cs.fdC);
+            int c = fd_mod(gk, cs.fdC, rs);
+            int r = fd_div(rs, cs.fdS);
+            int s = fd_mod(rs, cs.fdS, r);
             int h = p * cs.stride + r - cs.pad;
             int wcol = q * cs.stride + s - cs.pad;
             if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
@@ -481,16 +509,20 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                 lds_a[j + jj][mloc] = ((bf16*)&va)[jj];
         }
         if (valid) {
-            int q = gm % cs.Q; long rem = gm / cs.Q;
-            int p = rem % cs.P; int n = rem / cs.P;
+            unsigned rem = fd_div((unsigned)gm, cs.fdQ);
+            int q = fd_mod((unsigned)gm, cs.fdQ, rem);
+            unsigned n = fd_div(rem, cs.fdP);
+            int p = fd_mod(rem, cs.fdP, n);
             #pragma unroll
             for (int half = 0; half < 2; ++half) {
                 const int j = j8 + half * 64;
                 bf16x8_t vb = {};
                 int gk = rc0 + j;
                 if (gk < Kgemm) {
-                    int c = gk % cs.C, rs = gk / cs.C;
-                    int r = rs / cs.S, s = rs % cs.S;
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
                     int h = p * cs.stride + r - cs.pad;
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
@@ -732,6 +764,7 @@ static ConvShape make_shape(const at::Tensor& x, int Ko, int R, int S,
     cs.Ko = Ko; cs.R = R; cs.S = S; cs.stride = stride; cs.pad = pad;
     cs.P = (cs.H + 2 * pad - R) / stride + 1;
     cs.Q = (cs.W + 2 * pad - S) / stride + 1;
+    init_fastdiv(cs);
     return cs;
 }
 
@@ -784,6 +817,7 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     cs.N = N; cs.C = C; cs.H = H; cs.W = W;
     cs.Ko = dy.size(1); cs.P = dy.size(2); cs.Q = dy.size(3);
     cs.R = R; cs.S = S; cs.stride = stride; cs.pad = pad;
+    init_fastdiv(cs);
     TORCH_CHECK(cs.Ko % 8 == 0);
     int Kgemm = cs.R * cs.S * cs.Ko;
     long M = (long)cs.N * cs.H * cs.W;
@@ -878,8 +912,12 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     static const char* wv = getenv("DDPX_WGRAD_V");
     const char sel = wv ? wv[0] : 0;
     const bool can_wide = cs.Ko >= 128 && Kgemm >= 128;
-    const bool use_wide = sel ? sel == 'w' : can_wide;
-    const bool use_sb = sel ? sel == 's' : !can_wide;
+    // wide also needs enough 128x128 tiles to fill the machine (measured:
+    // conv2 with 1x5 tiles is slower wide than sb, conv3/conv4 faster)
+    const bool enough = can_wide &&
+        ceil_div_i(cs.Ko, 128) * ceil_div_i(Kgemm, 128) >= 8;
+    const bool use_wide = sel ? sel == 'w' : enough;
+    const bool use_sb = sel ? sel == 's' : !enough;
     TORCH_CHECK(!use_wide || can_wide, "wide wgrad needs Ko,Kgemm >= 128");
     const int depth = (use_wide || use_sb) ? SBK : WBK;
     const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
