@@ -9,50 +9,81 @@
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
 
+// bf16x8-vectorized: each thread owns an 8-channel group of one output
+// pixel — 4×16 B window loads, 16 B result store, 8 B packed argmax.
 __global__ void k_maxpool2x2_fwd(const bf16* __restrict__ x,
                                  bf16* __restrict__ y,
                                  unsigned char* __restrict__ idx,
-                                 long total, int Ho, int Wo, int C, int H, int W) {
+                                 long total_v, int Ho, int Wo, int Cv,
+                                 int H, int W) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total; i += stride) {
-        int c = i % C;
-        long r = i / C;
+    const int C = Cv * 8;
+    for (; i < total_v; i += stride) {
+        int c8 = i % Cv;
+        long r = i / Cv;
         int wo = r % Wo; r /= Wo;
         int ho = r % Ho; long n = r / Ho;
-        const bf16* base = x + ((n * H + 2 * ho) * W + 2 * wo) * C + c;
-        float v0 = bf2f(base[0]);
-        float v1 = bf2f(base[C]);
-        float v2 = bf2f(base[(long)W * C]);
-        float v3 = bf2f(base[(long)W * C + C]);
-        float best = v0; int arg = 0;
-        if (v1 > best) { best = v1; arg = 1; }
-        if (v2 > best) { best = v2; arg = 2; }
-        if (v3 > best) { best = v3; arg = 3; }
-        y[i] = f2bf(best);
-        idx[i] = (unsigned char)arg;
+        const s16x8* base = reinterpret_cast<const s16x8*>(
+            x + ((n * H + 2 * ho) * W + 2 * wo) * C) + c8;
+        const int wv = W * Cv;
+        s16x8 v0 = base[0];
+        s16x8 v1 = base[Cv];
+        s16x8 v2 = base[wv];
+        s16x8 v3 = base[wv + Cv];
+        s16x8 o;
+        unsigned long long packed = 0;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float b = us2f((unsigned short)v0[j]);
+            int arg = 0;
+            float f1 = us2f((unsigned short)v1[j]);
+            if (f1 > b) { b = f1; arg = 1; }
+            float f2 = us2f((unsigned short)v2[j]);
+            if (f2 > b) { b = f2; arg = 2; }
+            float f3 = us2f((unsigned short)v3[j]);
+            if (f3 > b) { b = f3; arg = 3; }
+            o[j] = (short)f2us(b);
+            packed |= (unsigned long long)arg << (8 * j);
+        }
+        reinterpret_cast<s16x8*>(y)[i] = o;
+        reinterpret_cast<unsigned long long*>(idx)[i] = packed;
     }
 }
 
 __global__ void k_maxpool2x2_bwd(const bf16* __restrict__ dy,
                                  const unsigned char* __restrict__ idx,
                                  bf16* __restrict__ dx,
-                                 long total, int Ho, int Wo, int C, int H, int W) {
+                                 long total_v, int Ho, int Wo, int Cv,
+                                 int H, int W) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    bf16 zero = f2bf(0.f);
-    for (; i < total; i += stride) {
-        int c = i % C;
-        long r = i / C;
+    const int C = Cv * 8;
+    for (; i < total_v; i += stride) {
+        int c8 = i % Cv;
+        long r = i / Cv;
         int wo = r % Wo; r /= Wo;
         int ho = r % Ho; long n = r / Ho;
-        bf16* base = dx + ((n * H + 2 * ho) * W + 2 * wo) * C + c;
-        int arg = idx[i];
-        bf16 g = dy[i];
-        base[0] = (arg == 0) ? g : zero;
-        base[C] = (arg == 1) ? g : zero;
-        base[(long)W * C] = (arg == 2) ? g : zero;
-        base[(long)W * C + C] = (arg == 3) ? g : zero;
+        s16x8* base = reinterpret_cast<s16x8*>(
+            dx + ((n * H + 2 * ho) * W + 2 * wo) * C) + c8;
+        const int wv = W * Cv;
+        s16x8 g = reinterpret_cast<const s16x8*>(dy)[i];
+        unsigned long long packed =
+            reinterpret_cast<const unsigned long long*>(idx)[i];
+        s16x8 o0 = {}, o1 = {}, o2 = {}, o3 = {};
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            int arg = (packed >> (8 * j)) & 0xff;
+            short gv = g[j];
+            if (arg == 0) o0[j] = gv;
+            else if (arg == 1) o1[j] = gv;
+            else if (arg == 2) o2[j] = gv;
+            else o3[j] = gv;
+        }
+        base[0] = o0;
+        base[Cv] = o1;
+        base[wv] = o2;
+        base[wv + Cv] = o3;
     }
 }
 
@@ -237,18 +268,20 @@ std::vector<at::Tensor> maxpool2x2_fwd(at::Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
     int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
     TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "maxpool2x2 requires even H,W");
+    TORCH_CHECK(C % 8 == 0, "maxpool2x2 HIP path needs C % 8 == 0");
     int Ho = H / 2, Wo = W / 2;
     auto y = at::empty({N, C, Ho, Wo},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
     auto idx = at::empty({N, Ho, Wo, C}, x.options().dtype(at::kByte));
-    long total = (long)N * Ho * Wo * C;
+    long total_v = (long)N * Ho * Wo * (C / 8);
     auto stream = at::hip::getCurrentHIPStream();
-    int blocks = std::min<long>(4096, ceil_div_i(total, 256));
+    int blocks = std::min<long>(8192, ceil_div_i(total_v, 256));
     hipLaunchKernelGGL(k_maxpool2x2_fwd, dim3(blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
-                       idx.data_ptr<unsigned char>(), total, Ho, Wo, C, H, W);
+                       idx.data_ptr<unsigned char>(), total_v, Ho, Wo, C / 8,
+                       H, W);
     HIP_CHECK_LAST();
     return {y, idx};
 }
@@ -257,15 +290,15 @@ at::Tensor maxpool2x2_bwd(at::Tensor dy, at::Tensor idx, long H, long W) {
     int N = dy.size(0), C = dy.size(1), Ho = dy.size(2), Wo = dy.size(3);
     auto dx = at::empty({N, C, (int)H, (int)W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
-    long total = (long)N * Ho * Wo * C;
+    long total_v = (long)N * Ho * Wo * (C / 8);
     auto stream = at::hip::getCurrentHIPStream();
-    int blocks = std::min<long>(4096, ceil_div_i(total, 256));
+    int blocks = std::min<long>(8192, ceil_div_i(total_v, 256));
     hipLaunchKernelGGL(k_maxpool2x2_bwd, dim3(blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(dy.data_ptr()),
                        idx.data_ptr<unsigned char>(),
                        reinterpret_cast<bf16*>(dx.data_ptr()),
-                       total, Ho, Wo, C, (int)H, (int)W);
+                       total_v, Ho, Wo, C / 8, (int)H, (int)W);
     HIP_CHECK_LAST();
     return dx;
 }
