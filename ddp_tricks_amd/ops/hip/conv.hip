@@ -923,7 +923,10 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
     int gk = ceil_div_i(cs.Ko, tm), gr = ceil_div_i(Kgemm, tn);
     int S_ = 1;
-    while (gk * gr * S_ < 1024 && S_ < 64 && (M / (S_ * 2)) >= depth) S_ *= 2;
+    // split until the machine fills (256 CUs x several blocks); small tile
+    // grids (1x1 convs: one 64x64 tile) need deep splits — measured 17
+    // TFLOP/s at the old S<=64 cap on [Ko=64,C=64] wgrad
+    while (gk * gr * S_ < 1024 && S_ < 512 && (M / (S_ * 2)) >= depth) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
     if (use_wide)

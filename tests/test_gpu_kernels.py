@@ -458,3 +458,26 @@ def test_conv_pad8_stem_path_isolated():
     yr.backward(dy)
     _close(w.grad, wr.grad, rel=2e-2, atol=0.5, name="pad8 wgrad")
     _close(x.grad, xr.grad, rel=3e-2, atol=3e-2, name="pad8 dgrad")
+
+
+def test_conv1x1_gemm_route():
+    """1x1 s1 convs route through the GEMM path — full autograd parity."""
+    from ddp_tricks_amd.ops import functional as F_ops
+    torch.manual_seed(12)
+    x = torch.randn(8, 64, 14, 14, device=DEV).to(torch.bfloat16).float()
+    x.requires_grad_(True)
+    w = torch.nn.Parameter(
+        torch.randn(256, 64, 1, 1, device=DEV).to(torch.bfloat16).float())
+    b = torch.nn.Parameter(torch.randn(256, device=DEV))
+    y = F_ops.conv2d(x, w, b, stride=1, padding=0)
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    yr = torch.nn.functional.conv2d(xr, wr, br, 1, 0)
+    _close(y, yr, name="1x1 fwd")
+    dy = torch.randn_like(yr).to(torch.bfloat16).float()
+    y.backward(dy)
+    yr.backward(dy)
+    _close(x.grad, xr.grad, rel=3e-2, atol=3e-2, name="1x1 dgrad")
+    _close(w.grad, wr.grad, rel=2e-2, atol=0.5, name="1x1 wgrad")
+    _close(b.grad, br.grad, rel=2e-2, atol=0.5, name="1x1 bias")
