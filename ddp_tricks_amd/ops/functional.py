@@ -272,10 +272,10 @@ class _HIPBatchNorm(torch.autograd.Function):
         xb = _chlast(_to_bf16(x))
         rb = None if residual is None else _chlast(_to_bf16(residual))
         if training:
-            y, save_mean, save_invstd = ext.bn_fwd_train(
+            y, save_mean, save_invstd, mask = ext.bn_fwd_train(
                 xb, weight.detach(), bias.detach(), running_mean, running_var,
                 momentum, eps, fuse_relu, rb)
-            ctx.save_for_backward(xb, weight, save_mean, save_invstd, y)
+            ctx.save_for_backward(xb, weight, save_mean, save_invstd, mask)
         else:
             y = ext.bn_fwd_eval(xb, weight.detach(), bias.detach(),
                                 running_mean, running_var, eps, fuse_relu, rb)
@@ -288,11 +288,11 @@ class _HIPBatchNorm(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         assert ctx.training, "backward through eval-mode BN is unsupported"
-        xb, weight, save_mean, save_invstd, y = ctx.saved_tensors
+        xb, weight, save_mean, save_invstd, mask = ctx.saved_tensors
         ext = require_ext_for(dy)
         dyb = _chlast(_to_bf16(dy))
         out = ext.bn_bwd(xb, dyb, weight.detach(), save_mean,
-                         save_invstd, y, ctx.fuse_relu, ctx.has_residual)
+                         save_invstd, mask, ctx.fuse_relu, ctx.has_residual)
         dx, dweight, dbias = out[0], out[1], out[2]
         dresid = out[3] if ctx.has_residual else None
         if ctx.x_dtype == torch.float32:

@@ -38,7 +38,8 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        c10::optional<at::Tensor> residual);
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
-                               at::Tensor y, bool fuse_relu, bool want_dresid);
+                               at::Tensor mask, bool fuse_relu,
+                               bool want_dresid);
 
 // gemm.hip
 at::Tensor gemm_tn(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> bias,
@@ -92,7 +93,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("fuse_relu") = false, py::arg("residual") = c10::nullopt);
     m.def("bn_bwd", &bn_bwd,
           py::arg("x"), py::arg("dy"), py::arg("gamma"), py::arg("save_mean"),
-          py::arg("save_invstd"), py::arg("y"), py::arg("fuse_relu") = false,
+          py::arg("save_invstd"), py::arg("mask"), py::arg("fuse_relu") = false,
           py::arg("want_dresid") = false);
     m.def("gemm_tn", &gemm_tn, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);

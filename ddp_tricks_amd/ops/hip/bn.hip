@@ -120,7 +120,8 @@ __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
                               const float* __restrict__ scale,
                               const float* __restrict__ shift,
                               long total_v, int Cv, bool relu,
-                              const bf16* __restrict__ resid) {
+                              const bf16* __restrict__ resid,
+                              unsigned char* __restrict__ mask) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
     for (; i < total_v; i += stride) {
@@ -128,21 +129,26 @@ __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
         s16x8 v = reinterpret_cast<const s16x8*>(x)[i];
         s16x8 rv = resid ? reinterpret_cast<const s16x8*>(resid)[i] : s16x8{};
         s16x8 o;
+        unsigned m = 0;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float f = fmaf(us2f((unsigned short)v[j]), scale[cv + j], shift[cv + j]);
             if (resid) f += us2f((unsigned short)rv[j]);
-            if (relu) f = fmaxf(f, 0.f);
+            if (relu) {
+                if (f > 0.f) m |= 1u << j;
+                f = fmaxf(f, 0.f);
+            }
             o[j] = (short)f2us(f);
         }
         reinterpret_cast<s16x8*>(y)[i] = o;
+        if (mask) mask[i] = (unsigned char)m;   // byte per 8-channel group
     }
 }
 
 // backward partials: slab[0][c][s] = sum(dy_eff), slab[1][c][s] = sum(dy_eff*xhat)
 __global__ void k_bn_bwd_partial(const bf16* __restrict__ x,
                                  const bf16* __restrict__ dy,
-                                 const bf16* __restrict__ y,  // post-relu mask
+                                 const unsigned char* __restrict__ mask,
                                  const float* __restrict__ save_mean,
                                  const float* __restrict__ save_invstd,
                                  long M, int C, int S, bool relu,
@@ -161,15 +167,14 @@ __global__ void k_bn_bwd_partial(const bf16* __restrict__ x,
         invstd[j] = save_invstd[c8 * 8 + j];
     }
     float sum_dy[8] = {}, sum_dyx[8] = {};
+    const int cpgv = cpg;
     for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
-        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y + r * C)[c8]
-                        : vg;
+        unsigned m = relu ? mask[r * cpgv + c8] : 0xffu;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            float g = us2f((unsigned short)vg[j]);
-            if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
+            float g = (m >> j) & 1u ? us2f((unsigned short)vg[j]) : 0.f;
             float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
             sum_dy[j] += g;
             sum_dyx[j] = fmaf(g, xh, sum_dyx[j]);
@@ -224,7 +229,7 @@ __global__ void k_bn_bwd_combine(const float* __restrict__ slab, int S, int C,
 // coefficients live in registers, not per-element loads.
 __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
                             const bf16* __restrict__ dy,
-                            const bf16* __restrict__ y,
+                            const unsigned char* __restrict__ mask,
                             const float* __restrict__ save_mean,
                             const float* __restrict__ save_invstd,
                             const float* __restrict__ coef_a,
@@ -252,12 +257,11 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
     for (long r = r0; r < M; r += rstride) {
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
-        s16x8 vy = relu ? reinterpret_cast<const s16x8*>(y + r * C)[c8] : vg;
+        unsigned m = relu ? mask[r * cpg + c8] : 0xffu;
         s16x8 o, og;
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            float g = us2f((unsigned short)vg[j]);
-            if (relu && us2f((unsigned short)vy[j]) <= 0.f) g = 0.f;
+            float g = (m >> j) & 1u ? us2f((unsigned short)vg[j]) : 0.f;
             float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
             o[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
             og[j] = (short)f2us(g);
@@ -329,11 +333,19 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
     int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
     const bf16* rp = residual.has_value()
         ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;
+    at::Tensor mask;
+    unsigned char* mp = nullptr;
+    if (fuse_relu) {   // packed relu mask: byte per 8-channel group (bwd
+        mask = at::empty({M, C / 8}, x.options().dtype(at::kByte));
+        mp = mask.data_ptr<unsigned char>();   // reads 1/16 of y's bytes)
+    } else {
+        mask = at::empty({0}, x.options().dtype(at::kByte));
+    }
     hipLaunchKernelGGL(k_bn_apply_v8, dim3(blocks), dim3(256), 0,
                        stream.stream(), xp, yp, scale.data_ptr<float>(),
-                       shift.data_ptr<float>(), tv, C / 8, fuse_relu, rp);
+                       shift.data_ptr<float>(), tv, C / 8, fuse_relu, rp, mp);
     HIP_CHECK_LAST();
-    return {y, save_mean, save_invstd};
+    return {y, save_mean, save_invstd, mask};
 }
 
 at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
@@ -365,14 +377,15 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
                        scale.data_ptr<float>(), shift.data_ptr<float>(), tv,
-                       C / 8, fuse_relu, rp);
+                       C / 8, fuse_relu, rp, nullptr);
     HIP_CHECK_LAST();
     return y;
 }
 
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
-                               at::Tensor y, bool fuse_relu, bool want_dresid) {
+                               at::Tensor mask, bool fuse_relu,
+                               bool want_dresid) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0 && C / 8 <= 256);
@@ -392,7 +405,10 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
         : at::empty_like(x);
     const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
     const bf16* dyp = reinterpret_cast<const bf16*>(dy.data_ptr());
-    const bf16* yp = reinterpret_cast<const bf16*>(y.data_ptr());
+    TORCH_CHECK(!fuse_relu || mask.numel() == M * (C / 8),
+                "bn_bwd needs the packed relu mask from bn_fwd_train");
+    const unsigned char* yp = fuse_relu
+        ? mask.data_ptr<unsigned char>() : nullptr;
     int lds = 2 * nw * C * 4;
 
     hipLaunchKernelGGL(k_bn_bwd_partial, dim3(S), dim3(block), lds,

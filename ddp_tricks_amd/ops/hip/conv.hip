@@ -926,7 +926,10 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     // split until the machine fills (256 CUs x several blocks); small tile
     // grids (1x1 convs: one 64x64 tile) need deep splits — measured 17
     // TFLOP/s at the old S<=64 cap on [Ko=64,C=64] wgrad
-    while (gk * gr * S_ < 1024 && S_ < 512 && (M / (S_ * 2)) >= depth) S_ *= 2;
+    // deepen splits until ~1024 blocks, but keep >=8 contraction steps per
+    // block (measured: S=512 at M=65k quadrupled a 55 us wgrad)
+    while (gk * gr * S_ < 1024 && S_ < 512 &&
+           M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
     if (use_wide)

@@ -169,7 +169,7 @@ def test_bn2d_fwd_bwd(relu):
     b = torch.randn(C, device=DEV)
     rm = torch.zeros(C, device=DEV)
     rv = torch.ones(C, device=DEV)
-    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, relu)
+    y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, relu)
 
     xf = x.float()
     ref_rm = torch.zeros(C)
@@ -192,7 +192,8 @@ def test_bn2d_fwd_bwd(relu):
         ref2 = ref2.relu()
     dy = torch.randn_like(ref2).to(torch.bfloat16)
     ref2.backward(dy.float())
-    dx, dg, db = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si, y, relu)
+    dx, dg, db = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si,
+                            mask, relu)
     _close(dg, g2.grad, rel=2e-2, atol=0.1, name="bn dgamma")
     _close(db, b2.grad, rel=2e-2, atol=0.1, name="bn dbeta")
     _close(dx, xf2.grad, rel=5e-2, atol=2e-2, name="bn dx")
@@ -205,7 +206,7 @@ def test_bn1d_fwd():
     b = torch.randn(C, device=DEV)
     rm = torch.zeros(C, device=DEV)
     rv = torch.ones(C, device=DEV)
-    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True)
+    y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True)
     ref = torch.nn.functional.batch_norm(
         x.float(), torch.zeros(C, device=DEV), torch.ones(C, device=DEV),
         g, b, True, 0.1, 1e-5).relu()
@@ -402,7 +403,7 @@ def test_bn_residual_fused():
     b = torch.randn(C, device=DEV)
     rm = torch.zeros(C, device=DEV)
     rv = torch.ones(C, device=DEV)
-    y, sm, si = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True, res)
+    y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True, res)
 
     xf = x.float().detach().requires_grad_(True)
     rf = res.float().detach().requires_grad_(True)
@@ -416,7 +417,8 @@ def test_bn_residual_fused():
 
     dy = torch.randn_like(ref).to(torch.bfloat16)
     ref.backward(dy.float())
-    out = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si, y, True, True)
+    out = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm, si, mask,
+                     True, True)
     dx, dg, db, dres = out
     _close(dg, g2.grad, rel=2e-2, atol=0.1, name="bn+res dgamma")
     _close(db, b2.grad, rel=2e-2, atol=0.1, name="bn+res dbeta")
