@@ -52,11 +52,44 @@ def time_op(fn, reps=200, warmup=20):
     return s.elapsed_time(e) / reps * 1000.0  # us
 
 
+def bench_functional(shapes, reps):
+    """Times the autograd-level conv path (true dispatch: 1x1 GEMM route,
+    pad8 stems, fused wgrad variants) — fwd and full bwd per shape."""
+    import torch.nn as nn
+
+    from ddp_tricks_amd.ops import functional as F_ops
+    dev = torch.device("cuda:0")
+    print(f"{'shape':8s} {'op':6s} {'us':>9s} {'TFLOP/s':>9s}")
+    for name, N, C, H, W, K, R, stride, pad in shapes:
+        P = (H + 2 * pad - R) // stride + 1
+        x = torch.randn(N, C, H, W, device=dev).to(torch.bfloat16)\
+            .contiguous(memory_format=CL).requires_grad_(True)
+        w = nn.Parameter(torch.randn(K, C, R, R, device=dev))
+        dy = torch.randn(N, K, P, P, device=dev).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        flops = 2.0 * N * P * P * K * C * R * R
+
+        us = time_op(lambda: F_ops.conv2d(x.detach(), w, None, stride, pad),
+                     reps)
+        print(f"{name:8s} {'Ffwd':6s} {us:9.1f} {flops/us/1e6:9.1f}")
+
+        def fb():
+            y = F_ops.conv2d(x, w, None, stride, pad)
+            y.backward(dy)
+        us = time_op(fb, reps // 2)
+        print(f"{name:8s} {'Ffb':6s} {us:9.1f} {3*flops/us/1e6:9.1f}")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--resnet", action="store_true")
+    ap.add_argument("--functional", action="store_true")
     ap.add_argument("--reps", type=int, default=200)
     args = ap.parse_args()
+    if args.functional:
+        load_extension(required=True)
+        bench_functional(TOYNET + (RESNET if args.resnet else []), args.reps)
+        return
     ext = load_extension(required=True)
     dev = torch.device("cuda:0")
     shapes = TOYNET + (RESNET if args.resnet else [])
