@@ -65,8 +65,14 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     constexpr int NI = WN / 16;
     constexpr int A_CHUNKS = CBM / 8;      // 1 KiB glds chunks (8 rows)
     constexpr int B_CHUNKS = TBN / 8;
-    __shared__ bf16 lds_a[CBM][CBK];       // unpadded: glds dest is linear
-    __shared__ bf16 lds_b[TBN][CBK];
+    // DOUBLE-BUFFERED glds ring: tile t+1's HBM->LDS DMA is issued right
+    // after the barrier and lands while tile t's MFMAs run; the next
+    // barrier's implicit vmcnt(0) absorbs whatever latency remains
+    // (cdna_hip_programming.md: 2-buf glds +40% on the GEMM ladder; one
+    // barrier per k-tile — a wave reaching it has finished its reads of
+    // the buffer the following issue overwrites).
+    __shared__ bf16 lds_a[2][CBM][CBK];    // unpadded: glds dest is linear
+    __shared__ bf16 lds_b[2][TBN][CBK];
     const int m0 = blockIdx.x * CBM;
     const int n0 = blockIdx.y * TBN;
     const int tid = threadIdx.x;
@@ -79,8 +85,8 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     const int pl_row = lane >> 3;
     const int pl_segp = lane & 7;          // physical segment (LDS-linear)
 
-    for (int kt = 0; kt < Kgemm; kt += CBK) {
-        // ---- A tile: 16 chunks round-robined over the 4 waves ----
+    auto issue_tile = [&](int kt, int buf) {
+        // ---- A tile: chunks round-robined over the 4 waves ----
         for (int ch = wid; ch < A_CHUNKS; ch += 4) {
             int row = ch * 8 + pl_row;
             int seg = pl_segp ^ (row & 7);     // logical k-segment
@@ -133,7 +139,7 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             }
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) unsigned short*)src,
-                (__attribute__((address_space(3))) unsigned short*)&lds_a[ch * 8][0],
+                (__attribute__((address_space(3))) unsigned short*)&lds_a[buf][ch * 8][0],
                 16, 0, 0);
         }
         // ---- B tile (row-contiguous source) ----
@@ -146,10 +152,17 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                 src = &Bsrc[(long)gn * Kgemm + gk];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) unsigned short*)src,
-                (__attribute__((address_space(3))) unsigned short*)&lds_b[ch * 8][0],
+                (__attribute__((address_space(3))) unsigned short*)&lds_b[buf][ch * 8][0],
                 16, 0, 0);
         }
-        __syncthreads();   // hipcc inserts the vmcnt(0) for in-flight glds
+    };
+
+    issue_tile(0, 0);
+    int buf = 0;
+    for (int kt = 0; kt < Kgemm; kt += CBK) {
+        __syncthreads();   // vmcnt(0) for the in-flight glds + barrier
+        if (kt + CBK < Kgemm)
+            issue_tile(kt + CBK, buf ^ 1);
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
             bf16x8_t af[MI], bfr[NI];
@@ -158,13 +171,13 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             for (int mi = 0; mi < MI; ++mi) {
                 int row = wr * WM + mi * 16 + (lane & 15);
                 af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[row][(kgrp ^ (row & 7)) * 8]);
+                    &lds_a[buf][row][(kgrp ^ (row & 7)) * 8]);
             }
             #pragma unroll
             for (int ni = 0; ni < NI; ++ni) {
                 int row = wc * WN + ni * 16 + (lane & 15);
                 bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[row][(kgrp ^ (row & 7)) * 8]);
+                    &lds_b[buf][row][(kgrp ^ (row & 7)) * 8]);
             }
             #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
@@ -173,7 +186,7 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
         }
-        __syncthreads();
+        buf ^= 1;
     }
 
     #pragma unroll

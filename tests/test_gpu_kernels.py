@@ -462,8 +462,8 @@ def test_conv_pad8_stem_path_isolated():
     _close(x.grad, xr.grad, rel=3e-2, atol=3e-2, name="pad8 dgrad")
 
 
-def test_conv1x1_gemm_route():
-    """1x1 s1 convs route through the GEMM path — full autograd parity."""
+def test_conv1x1_autograd_parity():
+    """1x1 s1 convs on the conv path — full autograd parity."""
     from ddp_tricks_amd.ops import functional as F_ops
     torch.manual_seed(12)
     x = torch.randn(8, 64, 14, 14, device=DEV).to(torch.bfloat16).float()
