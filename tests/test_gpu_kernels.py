@@ -144,6 +144,8 @@ def test_conv_dgrad():
 @pytest.mark.parametrize("shape", [
     (8, 64, 26, 26, 128, 3, 0),
     (8, 1, 28, 28, 64, 3, 0),         # small-Cin path
+    (8, 128, 12, 12, 256, 3, 0),      # wide-tile dispatch (Ko,Kgemm >= 128)
+    (8, 64, 14, 14, 64, 1, 0),        # 1x1 deep-split path
 ])
 def test_conv_wgrad(shape):
     N, C, H, W, K, R, pad = shape
