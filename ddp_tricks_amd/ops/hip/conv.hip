@@ -65,14 +65,13 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     constexpr int NI = WN / 16;
     constexpr int A_CHUNKS = CBM / 8;      // 1 KiB glds chunks (8 rows)
     constexpr int B_CHUNKS = TBN / 8;
-    // DOUBLE-BUFFERED glds ring: tile t+1's HBM->LDS DMA is issued right
-    // after the barrier and lands while tile t's MFMAs run; the next
-    // barrier's implicit vmcnt(0) absorbs whatever latency remains
-    // (cdna_hip_programming.md: 2-buf glds +40% on the GEMM ladder; one
-    // barrier per k-tile — a wave reaching it has finished its reads of
-    // the buffer the following issue overwrites).
-    __shared__ bf16 lds_a[2][CBM][CBK];    // unpadded: glds dest is linear
-    __shared__ bf16 lds_b[2][TBN][CBK];
+    // Single-buffer glds staging.  A 2-buffer prefetch ring was tried and
+    // REVERTED: doubling LDS (64 KB) halved resident blocks/CU and measured
+    // 10-30% SLOWER at every conv shape — this kernel hides HBM latency
+    // with block-level parallelism (grids >> 256 workgroups), unlike the
+    // guide's 256^2-tile GEMM that runs ~1 block/CU.
+    __shared__ bf16 lds_a[1][CBM][CBK];    // unpadded: glds dest is linear
+    __shared__ bf16 lds_b[1][TBN][CBK];
     const int m0 = blockIdx.x * CBM;
     const int n0 = blockIdx.y * TBN;
     const int tid = threadIdx.x;
@@ -157,12 +156,10 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
         }
     };
 
-    issue_tile(0, 0);
-    int buf = 0;
+    constexpr int buf = 0;
     for (int kt = 0; kt < Kgemm; kt += CBK) {
+        issue_tile(kt, 0);
         __syncthreads();   // vmcnt(0) for the in-flight glds + barrier
-        if (kt + CBK < Kgemm)
-            issue_tile(kt + CBK, buf ^ 1);
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
             bf16x8_t af[MI], bfr[NI];
@@ -186,7 +183,7 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
         }
-        buf ^= 1;
+        __syncthreads();
     }
 
     #pragma unroll
