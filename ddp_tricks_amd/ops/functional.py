@@ -41,6 +41,17 @@ def weight_variant(w: torch.Tensor, variant: str) -> torch.Tensor:
         ent = {"version": w._version}
         _WCACHE[w] = ent
     if variant not in ent:
+        ext = require_ext_for(w) if w.is_cuda else None
+        if (ext is not None and w.dtype == torch.float32
+                and variant in ("nhwc", "wt2", "nhwc_p8", "wt2_p8")):
+            # one fused kernel fills BOTH conv operand layouts per step
+            pad8 = variant.endswith("_p8")
+            sfx = "_p8" if pad8 else ""
+            nhwc, wt2 = ext.pack_conv_weight(w.detach().contiguous(), pad8,
+                                             True)
+            ent["nhwc" + sfx] = nhwc
+            ent["wt2" + sfx] = wt2
+            return ent[variant]
         wb = w.detach().to(torch.bfloat16)
         if variant == "flat":
             ent[variant] = wb.contiguous()

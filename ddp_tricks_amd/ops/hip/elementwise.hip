@@ -260,3 +260,59 @@ at::Tensor cast_to_bf16(at::Tensor x) {
     HIP_CHECK_LAST();
     return y;
 }
+
+// ------------------------------------------------- conv weight packing ---
+
+// One pass over the fp32 master weight [K,C,R,S] producing BOTH bf16
+// operand layouts: nhwc [K,R,S,Cp] (forward) and wt2 [Cp, R*S*K] (dgrad),
+// with optional zero-padding of C to Cp=8 (the C<8 stem path).  Replaces
+// the per-step chain of to(bf16) + contiguous(channels_last) + permute +
+// reshape + contiguous ATen launches (SURVEY N4's cast layer).
+__global__ void k_pack_conv_weight(const float* __restrict__ w,
+                                   bf16* __restrict__ nhwc,
+                                   bf16* __restrict__ wt2,
+                                   int K, int C, int RS, int Cp) {
+    long total = (long)K * Cp * RS;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += stride) {
+        int c = i % Cp;
+        long r2 = i / Cp;
+        int rs = r2 % RS;
+        int k = r2 / RS;
+        float v = c < C ? w[((long)k * C + c) * RS + rs] : 0.f;
+        bf16 b = f2bf(v);
+        nhwc[((long)k * RS + rs) * Cp + c] = b;           // [K,R,S,Cp]
+        if (wt2)
+            wt2[((long)c * RS + rs) * K + k] = b;         // [Cp, RS*K]
+    }
+}
+
+std::vector<at::Tensor> pack_conv_weight(at::Tensor w, bool pad8,
+                                         bool want_wt2) {
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kFloat &&
+                w.is_contiguous());
+    int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+    int Cp = pad8 ? 8 : C;
+    int RS = R * S;
+    auto opts = w.options().dtype(at::kBFloat16);
+    // nhwc tensor carries channels_last metadata for [K,Cp,R,S]
+    auto nhwc = at::empty({K, Cp, R, S},
+                          opts.memory_format(at::MemoryFormat::ChannelsLast));
+    at::Tensor wt2;
+    bf16* wt2p = nullptr;
+    if (want_wt2) {
+        wt2 = at::empty({Cp, (long)RS * K}, opts);
+        wt2p = reinterpret_cast<bf16*>(wt2.data_ptr());
+    }
+    long total = (long)K * Cp * RS;
+    auto stream = at::hip::getCurrentHIPStream();
+    int blocks = std::min<long>(2048, ceil_div_i(total, 256));
+    hipLaunchKernelGGL(k_pack_conv_weight, dim3(blocks), dim3(256), 0,
+                       stream.stream(), w.data_ptr<float>(),
+                       reinterpret_cast<bf16*>(nhwc.data_ptr()), wt2p,
+                       K, C, RS, Cp);
+    HIP_CHECK_LAST();
+    if (want_wt2) return {nhwc, wt2};
+    return {nhwc};
+}

@@ -485,3 +485,20 @@ def test_conv1x1_autograd_parity():
     _close(x.grad, xr.grad, rel=3e-2, atol=3e-2, name="1x1 dgrad")
     _close(w.grad, wr.grad, rel=2e-2, atol=0.5, name="1x1 wgrad")
     _close(b.grad, br.grad, rel=2e-2, atol=0.5, name="1x1 bias")
+
+
+def test_pack_conv_weight():
+    """Fused fp32 KCRS -> bf16 KRSC + WT2 packing == the torch-op chain."""
+    torch.manual_seed(13)
+    for K, C, R, pad8 in [(128, 64, 3, False), (64, 3, 7, True),
+                          (256, 64, 1, False)]:
+        w = torch.randn(K, C, R, R, device=DEV)
+        nhwc, wt2 = ext.pack_conv_weight(w, pad8, True)
+        Cp = 8 if pad8 else C
+        wb = torch.zeros(K, Cp, R, R, device=DEV, dtype=torch.bfloat16)
+        wb[:, :C] = w.to(torch.bfloat16)
+        ref_nhwc = wb.contiguous(memory_format=CL)
+        ref_wt2 = wb.permute(1, 2, 3, 0).reshape(Cp, R * R * K).contiguous()
+        assert nhwc.shape == ref_nhwc.shape
+        assert torch.equal(nhwc.float(), ref_nhwc.float())
+        assert torch.equal(wt2.float(), ref_wt2.float())
