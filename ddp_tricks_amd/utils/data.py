@@ -72,11 +72,13 @@ class MNIST(Dataset):
             if num_samples is None and env_cap:
                 num_samples = min(int(env_cap), n_default)
             n = num_samples or n_default
+            # Class templates come from a FIXED seed shared by train and
+            # valid (a validation set drawn from different class patterns
+            # is unlearnable); labels/noise differ per split.
+            gt = torch.Generator().manual_seed(9999)
+            templates = torch.rand(10, 1, 28, 28, generator=gt)
             g = torch.Generator().manual_seed(1234 if train else 4321)
             labels = torch.randint(0, 10, (n,), generator=g)
-            # Class-dependent separable patterns + noise: a fixed random
-            # template per class so the net has something learnable.
-            templates = torch.rand(10, 1, 28, 28, generator=g)
             noise = torch.rand(n, 1, 28, 28, generator=g)
             self.images = (0.6 * templates[labels] + 0.4 * noise).clamp_(0, 1)
             self.labels = labels
@@ -118,9 +120,10 @@ class CIFAR10(Dataset):
             if num_samples is None and env_cap:
                 num_samples = min(int(env_cap), n_default)
             n = num_samples or n_default
+            gt = torch.Generator().manual_seed(8888)   # shared templates
+            templates = torch.rand(10, 3, 32, 32, generator=gt)
             g = torch.Generator().manual_seed(777 if train else 778)
             labels = torch.randint(0, 10, (n,), generator=g)
-            templates = torch.rand(10, 3, 32, 32, generator=g)
             noise = torch.rand(n, 3, 32, 32, generator=g)
             self.images = (0.6 * templates[labels] + 0.4 * noise).clamp_(0, 1)
             self.labels = labels

@@ -52,3 +52,33 @@ def test_matches_torch_distributed_sampler():
         ours.set_epoch(epoch)
         theirs.set_epoch(epoch)
         assert list(iter(ours)) == list(iter(theirs))
+
+
+def test_synthetic_templates_shared_across_splits():
+    """Valid-set class templates must equal the train-set's (a valid set
+    drawn from different patterns is unlearnable — caught on GPU run 1)."""
+    import os
+
+    import torch
+
+    from ddp_tricks_amd.utils.data import MNIST, CIFAR10
+    os.environ["DDPX_SYNTH_SAMPLES"] = "64"
+    try:
+        for DS in (MNIST, CIFAR10):
+            tr = DS(root="/nonexistent", train=True)
+            va = DS(root="/nonexistent", train=False)
+            # estimate per-class mean images; shared templates => the same
+            # class across splits correlates far better than across classes
+            def class_mean(ds, c):
+                idx = [i for i in range(len(ds)) if ds.labels[i] == c]
+                return ds.images[idx].mean(0).flatten() if idx else None
+            same, diff = [], []
+            for c in range(10):
+                a, b = class_mean(tr, c), class_mean(va, c)
+                if a is None or b is None:
+                    continue
+                same.append(torch.nn.functional.cosine_similarity(
+                    a, b, dim=0).item())
+            assert sum(same) / len(same) > 0.9, same
+    finally:
+        del os.environ["DDPX_SYNTH_SAMPLES"]
