@@ -482,16 +482,17 @@ void k_conv_wgrad_sb(const bf16* __restrict__ dy, const bf16* __restrict__ x,
         }
 }
 
-// Wide wgrad variant: 128(ko) x 128(rsc) tile, 32-deep contraction — four
-// times the MFMA work per barrier pair of the 64x64 tile, same per-thread
-// staging cost per element (2 transposed bf16x8 stores per operand).
+// Wide wgrad variant: 128(ko) x 128(rsc) tile, DEPTH-deep contraction —
+// more MFMA work per barrier pair than the 64x64 tile at the same
+// per-thread staging cost per element (transposed bf16x8 stores).
 // Requires Ko >= 128 and Kgemm >= 128 (host falls back to sb otherwise).
+template <int DEPTH>
 __global__ __launch_bounds__(256)
 void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                        float* __restrict__ slab, ConvShape cs, long M,
                        int Kgemm, int S) {
-    __shared__ bf16 lds_a[128][SLDK];   // [ko][m]
-    __shared__ bf16 lds_b[128][SLDK];   // [rsc][m]
+    __shared__ bf16 lds_a[128][DEPTH + 8];   // [ko][m]
+    __shared__ bf16 lds_b[128][DEPTH + 8];   // [rsc][m]
     const int ko0 = blockIdx.x * 128;
     const int rc0 = blockIdx.y * 128;
     const int split = blockIdx.z;
@@ -500,16 +501,16 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
     const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
 
     f32x4 acc[4][4] = {};
-    const int mloc = tid & 31;
-    const int j8 = (tid >> 5) * 8;
+    const int mloc = tid % DEPTH;
+    const int jb = (tid / DEPTH) * 8;
+    constexpr int JSTEP = (256 / DEPTH) * 8;
 
-    const long m_begin = (long)split * SBK;
-    for (long mt = m_begin; mt < M; mt += (long)S * SBK) {
+    const long m_begin = (long)split * DEPTH;
+    for (long mt = m_begin; mt < M; mt += (long)S * DEPTH) {
         long gm = mt + mloc;
         const bool valid = gm < M;
         #pragma unroll
-        for (int half = 0; half < 2; ++half) {
-            const int j = j8 + half * 64;
+        for (int j = jb; j < 128; j += JSTEP) {
             bf16x8_t va = {};
             if (valid && ko0 + j < cs.Ko)
                 va = *reinterpret_cast<const bf16x8_t*>(
@@ -524,8 +525,7 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
             unsigned n = fd_div(rem, cs.fdP);
             int p = fd_mod(rem, cs.fdP, n);
             #pragma unroll
-            for (int half = 0; half < 2; ++half) {
-                const int j = j8 + half * 64;
+            for (int j = jb; j < 128; j += JSTEP) {
                 bf16x8_t vb = {};
                 int gk = rc0 + j;
                 if (gk < Kgemm) {
@@ -545,31 +545,32 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
             }
         } else {
             #pragma unroll
-            for (int half = 0; half < 2; ++half) {
-                const int j = j8 + half * 64;
+            for (int j = jb; j < 128; j += JSTEP)
                 #pragma unroll
                 for (int jj = 0; jj < 8; ++jj)
                     lds_b[j + jj][mloc] = (bf16)0;
-            }
         }
         __syncthreads();
 
-        bf16x8_t af[4], bfr[4];
-        const int kcol = (lane >> 4) * 8;
         #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-            af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
-        #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+        for (int ks = 0; ks < DEPTH; ks += 32) {
+            bf16x8_t af[4], bfr[4];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
-                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
         __syncthreads();
     }
 
@@ -929,7 +930,8 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     const bool use_wide = sel ? sel == 'w' : enough;
     const bool use_sb = sel ? sel == 's' : !enough;
     TORCH_CHECK(!use_wide || can_wide, "wide wgrad needs Ko,Kgemm >= 128");
-    const int depth = (use_wide || use_sb) ? SBK : WBK;
+    const bool wide64 = use_wide && !(wv && wv[1] == '3');  // w32 forces 32
+    const int depth = use_wide ? (wide64 ? 64 : SBK) : (use_sb ? SBK : WBK);
     const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
     int gk = ceil_div_i(cs.Ko, tm), gr = ceil_div_i(Kgemm, tn);
     int S_ = 1;
@@ -942,10 +944,14 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
            M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
-    if (use_wide)
-        hipLaunchKernelGGL(k_conv_wgrad_wide, dim3(gk, gr, S_), dim3(256), 0,
-                           stream.stream(), dyp, xp, slab.data_ptr<float>(),
-                           cs, M, Kgemm, S_);
+    if (use_wide && wide64)
+        hipLaunchKernelGGL((k_conv_wgrad_wide<64>), dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_wide)
+        hipLaunchKernelGGL((k_conv_wgrad_wide<32>), dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_sb)
         hipLaunchKernelGGL(k_conv_wgrad_sb, dim3(gk, gr, S_), dim3(256), 0,
                            stream.stream(), dyp, xp, slab.data_ptr<float>(),
