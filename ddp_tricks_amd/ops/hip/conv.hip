@@ -964,9 +964,9 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     // conv2 with 1x5 tiles is slower wide than sb, conv3/conv4 faster)
     const bool enough = can_wide &&
         ceil_div_i(cs.Ko, 128) * ceil_div_i(Kgemm, 128) >= 8;
-    const bool use_wide = sel ? sel == 'w' : enough;
-    const bool use_sb = sel ? sel == 's' : !enough;
-    TORCH_CHECK(!use_wide || can_wide, "wide wgrad needs Ko,Kgemm >= 128");
+    const bool use_wide = (sel ? sel == 'w' : enough) && can_wide;
+    const bool use_sb = sel ? (sel == 's' || (sel == 'w' && !can_wide))
+                            : !enough;
     const bool wide64 = use_wide && !(wv && wv[1] == '3');  // w32 forces 32
     const int depth = use_wide ? (wide64 ? 64 : SBK) : (use_sb ? SBK : WBK);
     const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
