@@ -246,25 +246,6 @@ __global__ void k_conv_small_cin(const bf16* __restrict__ x,
     }
 }
 
-// Zero-stuff dy for strided dgrad: dy2[p*st][q*st] = dy[p][q], zeros
-// elsewhere — turns a stride-s dgrad into a stride-1 dgrad over dy2
-// (the divisibility-checked gather measured ~3x off the stride-1 path).
-__global__ void k_stuff_dy(const bf16* __restrict__ dy, bf16* __restrict__ dy2,
-                           long total_v, int P, int Q, int Cv,
-                           int P2, int Q2, int st) {
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total_v; i += stride) {
-        int c8 = i % Cv;
-        long r = i / Cv;
-        int q = r % Q; r /= Q;
-        int p = r % P; long n = r / P;
-        s16x8 v = reinterpret_cast<const s16x8*>(dy)[i];
-        reinterpret_cast<s16x8*>(dy2)[
-            ((n * P2 + (long)p * st) * Q2 + (long)q * st) * Cv + c8] = v;
-    }
-}
-
 // -------------------------------------------------------- backward weight ---
 
 // slab[s][ko, rsc] = Σ_{m in split s} dy[m][ko] · im2col(x)[m][rsc]
@@ -854,24 +835,10 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     auto dx = at::empty({(long)cs.N, (long)cs.C, (long)cs.H, (long)cs.W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
     auto stream = at::hip::getCurrentHIPStream();
-    at::Tensor dy2;   // keep alive until launch
-    if (cs.stride > 1) {
-        int P2 = cs.stride * (cs.P - 1) + 1;
-        int Q2 = cs.stride * (cs.Q - 1) + 1;
-        dy2 = at::zeros({(long)cs.N, (long)cs.Ko, (long)P2, (long)Q2},
-                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
-        long tv = (long)cs.N * cs.P * cs.Q * (cs.Ko / 8);
-        int blocks = std::min<long>(4096, ceil_div_i(tv, 256));
-        hipLaunchKernelGGL(k_stuff_dy, dim3(blocks), dim3(256), 0,
-                           stream.stream(),
-                           reinterpret_cast<const bf16*>(dy.data_ptr()),
-                           reinterpret_cast<bf16*>(dy2.data_ptr()),
-                           tv, cs.P, cs.Q, cs.Ko / 8, P2, Q2, cs.stride);
-        HIP_CHECK_LAST();
-        dy = dy2;
-        cs.P = P2; cs.Q = Q2; cs.stride = 1;
-        init_fastdiv(cs);
-    }
+    // NOTE: zero-stuffing dy to reach the stride-1 gather was tried and
+    // REVERTED — measured slower at ResNet's strided shapes (r18 3x3/2:
+    // 113 vs 99 us, 1x1/2: 44 vs 22 us): the 4x stuffed-dy traffic costs
+    // more than the divisibility-checked gather saves.
     const bf16* dyp_ = reinterpret_cast<const bf16*>(dy.data_ptr());
     const bf16* wt2p = reinterpret_cast<const bf16*>(wt2.data_ptr());
     bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
