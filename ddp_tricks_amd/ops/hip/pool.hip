@@ -92,47 +92,66 @@ __global__ void k_maxpool2x2_bwd(const bf16* __restrict__ dy,
 // (< ks*ks <= 256) per output element; backward GATHERS: windows may
 // overlap (st < ks), so each input element sums the dy of every output
 // window that claimed it — no atomics, deterministic.
+// bf16x8-vectorized: thread owns an 8-channel group of one output pixel;
+// argmax bytes packed 8-at-a-time (same layout trick as the 2x2 kernel).
 __global__ void k_maxpool_fwd(const bf16* __restrict__ x,
                               bf16* __restrict__ y,
-                              unsigned char* __restrict__ idx, long total,
-                              int Ho, int Wo, int C, int H, int W,
+                              unsigned char* __restrict__ idx, long total_v,
+                              int Ho, int Wo, int Cv, int H, int W,
                               int ks, int st, int pad) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total; i += stride) {
-        int c = i % C;
-        long r = i / C;
+    const int C = Cv * 8;
+    for (; i < total_v; i += stride) {
+        int c8 = i % Cv;
+        long r = i / Cv;
         int wo = r % Wo; r /= Wo;
         int ho = r % Ho; long n = r / Ho;
-        float best = -3.4e38f; int arg = 0;
+        float best[8];
+        int arg[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) { best[j] = -3.4e38f; arg[j] = 0; }
         for (int kh = 0; kh < ks; ++kh) {
             int h = ho * st + kh - pad;
             if (h < 0 || h >= H) continue;
             for (int kw = 0; kw < ks; ++kw) {
                 int w = wo * st + kw - pad;
                 if (w < 0 || w >= W) continue;
-                float v = bf2f(x[((n * H + h) * W + w) * C + c]);
-                if (v > best) { best = v; arg = kh * ks + kw; }
+                s16x8 v = reinterpret_cast<const s16x8*>(
+                    x + ((n * H + h) * W + w) * C)[c8];
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    float f = us2f((unsigned short)v[j]);
+                    if (f > best[j]) { best[j] = f; arg[j] = kh * ks + kw; }
+                }
             }
         }
-        y[i] = f2bf(best);
-        idx[i] = (unsigned char)arg;
+        s16x8 o;
+        unsigned long long packed = 0;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            o[j] = (short)f2us(best[j]);
+            packed |= (unsigned long long)(unsigned)arg[j] << (8 * j);
+        }
+        reinterpret_cast<s16x8*>(y)[i] = o;
+        reinterpret_cast<unsigned long long*>(idx)[i] = packed;
     }
 }
 
 __global__ void k_maxpool_bwd(const bf16* __restrict__ dy,
                               const unsigned char* __restrict__ idx,
-                              bf16* __restrict__ dx, long total_in,
-                              int Ho, int Wo, int C, int H, int W,
+                              bf16* __restrict__ dx, long total_v,
+                              int Ho, int Wo, int Cv, int H, int W,
                               int ks, int st, int pad) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total_in; i += stride) {
-        int c = i % C;
-        long r = i / C;
+    const int C = Cv * 8;
+    for (; i < total_v; i += stride) {
+        int c8 = i % Cv;
+        long r = i / Cv;
         int w = r % W; r /= W;
         int h = r % H; long n = r / H;
-        float acc = 0.f;
+        float acc[8] = {};
         // output windows containing (h, w): ho*st <= h+pad <= ho*st+ks-1
         int ho_lo = (h + pad - ks + 1 + st - 1); ho_lo = ho_lo > 0 ? ho_lo / st : 0;
         int ho_hi = min((h + pad) / st, Ho - 1);
@@ -142,11 +161,21 @@ __global__ void k_maxpool_bwd(const bf16* __restrict__ dy,
             int kh = h + pad - ho * st;
             for (int wo = wo_lo; wo <= wo_hi; ++wo) {
                 int kw = w + pad - wo * st;
-                long o = ((n * Ho + ho) * Wo + wo) * C + c;
-                if (idx[o] == kh * ks + kw) acc += bf2f(dy[o]);
+                long o = ((n * Ho + ho) * Wo + wo) * Cv + c8;
+                unsigned char want = (unsigned char)(kh * ks + kw);
+                unsigned long long packed =
+                    reinterpret_cast<const unsigned long long*>(idx)[o];
+                s16x8 g = reinterpret_cast<const s16x8*>(dy)[o];
+                #pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    if (((packed >> (8 * j)) & 0xff) == want)
+                        acc[j] += us2f((unsigned short)g[j]);
             }
         }
-        dx[i] = f2bf(acc);
+        s16x8 o8;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) o8[j] = (short)f2us(acc[j]);
+        reinterpret_cast<s16x8*>(dx)[i] = o8;
     }
 }
 
@@ -196,18 +225,19 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long ks, long st, long pad) {
     int Ho = (H + 2 * (int)pad - (int)ks) / (int)st + 1;
     int Wo = (W + 2 * (int)pad - (int)ks) / (int)st + 1;
     TORCH_CHECK(ks * ks <= 256);
+    TORCH_CHECK(C % 8 == 0, "maxpool HIP path needs C % 8 == 0");
     auto y = at::empty({N, C, Ho, Wo},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
     auto idx = at::empty({N, Ho, Wo, C}, x.options().dtype(at::kByte));
-    long total = (long)N * Ho * Wo * C;
+    long total_v = (long)N * Ho * Wo * (C / 8);
     auto stream = at::hip::getCurrentHIPStream();
-    int blocks = std::min<long>(8192, ceil_div_i(total, 256));
+    int blocks = std::min<long>(8192, ceil_div_i(total_v, 256));
     hipLaunchKernelGGL(k_maxpool_fwd, dim3(blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
-                       idx.data_ptr<unsigned char>(), total, Ho, Wo, C, H, W,
-                       (int)ks, (int)st, (int)pad);
+                       idx.data_ptr<unsigned char>(), total_v, Ho, Wo, C / 8,
+                       H, W, (int)ks, (int)st, (int)pad);
     HIP_CHECK_LAST();
     return {y, idx};
 }
@@ -217,15 +247,15 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
     int N = dy.size(0), C = dy.size(1), Ho = dy.size(2), Wo = dy.size(3);
     auto dx = at::empty({N, C, (int)H, (int)W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
-    long total = (long)N * H * W * C;
+    long total_v = (long)N * H * W * (C / 8);
     auto stream = at::hip::getCurrentHIPStream();
-    int blocks = std::min<long>(8192, ceil_div_i(total, 256));
+    int blocks = std::min<long>(8192, ceil_div_i(total_v, 256));
     hipLaunchKernelGGL(k_maxpool_bwd, dim3(blocks), dim3(256), 0,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(dy.data_ptr()),
                        idx.data_ptr<unsigned char>(),
                        reinterpret_cast<bf16*>(dx.data_ptr()),
-                       total, Ho, Wo, C, (int)H, (int)W,
+                       total_v, Ho, Wo, C / 8, (int)H, (int)W,
                        (int)ks, (int)st, (int)pad);
     HIP_CHECK_LAST();
     return dx;
