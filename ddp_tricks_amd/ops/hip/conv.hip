@@ -27,6 +27,14 @@ struct ConvShape {
     int stride, pad;
     // magic-division decoders for the gathers (set on host)
     FastDiv fdQ, fdP, fdW, fdH, fdC, fdS, fdKo;
+    // MODE 2 (strided dgrad, sub-grid class (a,b)): only taps whose
+    // stride-residue matches the class contribute, so Kgemm shrinks to
+    // nr*ns*Ko and no MFMA work is spent on zero taps.
+    int cls_a, cls_b, nr, ns;
+    int taps_r[4], taps_s[4];   // contributing (r, s) values
+    int off_r[4], off_s[4];     // p = h' + off_r[ri], q = w' + off_s[si]
+    int Hfull, Wfull;           // original dx dims (epilogue scatter)
+    FastDiv fdNs;
 };
 
 static inline void init_fastdiv(ConvShape& cs) {
@@ -105,6 +113,21 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     int wcol = q * cs.stride + s - cs.pad;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
                         src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
+                } else if (MODE == 2) {
+                    // sub-grid class: gm -> (n, h', w') over the class's
+                    // Ha x Wb grid (fdH/fdW hold Ha/Wb); gk -> (ri,si,ko)
+                    unsigned rem = fd_div(gm, cs.fdW);
+                    int w1 = fd_mod(gm, cs.fdW, rem);
+                    unsigned n = fd_div(rem, cs.fdH);
+                    int h1 = fd_mod(rem, cs.fdH, n);
+                    unsigned rs2 = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs2);
+                    int ri = fd_div(rs2, cs.fdNs);
+                    int si = fd_mod(rs2, cs.fdNs, ri);
+                    int p = h1 + cs.off_r[ri];
+                    int q = w1 + cs.off_s[si];
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 } else if (STRIDE1) {
                     unsigned rem = fd_div(gm, cs.fdW);
                     int wcol = fd_mod(gm, cs.fdW, rem);
@@ -147,7 +170,18 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             int seg = pl_segp ^ (row & 7);
             int gn = n0 + row, gk = kt + seg * 8;
             const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
-            if (gn < Nout && gk < Kgemm)
+            if (MODE == 2) {
+                // full wt2 row is R*S*Ko wide; map the class k-index to it
+                if (gn < Nout && gk < Kgemm) {
+                    unsigned rs2 = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs2);
+                    int ri = fd_div(rs2, cs.fdNs);
+                    int si = fd_mod(rs2, cs.fdNs, ri);
+                    long gk_full = ((long)cs.taps_r[ri] * cs.S
+                                    + cs.taps_s[si]) * cs.Ko + ko;
+                    src = &Bsrc[(long)gn * (cs.R * cs.S * cs.Ko) + gk_full];
+                }
+            } else if (gn < Nout && gk < Kgemm)
                 src = &Bsrc[(long)gn * Kgemm + gk];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) unsigned short*)src,
@@ -197,7 +231,21 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             for (int r = 0; r < 4; ++r) {
                 int row = m0 + wr * WM + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= M) continue;
-                out[(long)row * Nout + col] = f2bf(acc[mi][ni][r] + badd);
+                if (MODE == 2) {
+                    // scatter the class sub-grid back into full dx:
+                    // (n, h', w') -> (a + h'*stride, b + w'*stride)
+                    unsigned rem = fd_div(row, cs.fdW);
+                    int w1 = fd_mod(row, cs.fdW, rem);
+                    unsigned n = fd_div(rem, cs.fdH);
+                    int h1 = fd_mod(rem, cs.fdH, n);
+                    long addr = (((long)n * cs.Hfull + cs.cls_a
+                                  + (long)h1 * cs.stride) * cs.Wfull
+                                 + cs.cls_b + (long)w1 * cs.stride) * Nout
+                                + col;
+                    out[addr] = f2bf(acc[mi][ni][r] + badd);
+                } else {
+                    out[(long)row * Nout + col] = f2bf(acc[mi][ni][r] + badd);
+                }
             }
         }
 }
@@ -842,28 +890,87 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     const bf16* dyp_ = reinterpret_cast<const bf16*>(dy.data_ptr());
     const bf16* wt2p = reinterpret_cast<const bf16*>(wt2.data_ptr());
     bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
-    if (cs.C >= 128) {
-        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
-        if (cs.stride == 1)
+    if (cs.stride == 1) {
+        if (cs.C >= 128) {
+            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
             hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
                                nullptr, dxp, cs, (int)M, Kgemm, cs.C);
-        else
-            hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, false>), grid,
-                               dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
-    } else {
-        dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
-        if (cs.stride == 1)
+        } else {
+            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
             hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
                                nullptr, dxp, cs, (int)M, Kgemm, cs.C);
-        else
-            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, false>), grid,
-                               dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+        }
+        HIP_CHECK_LAST();
+        return dx;
     }
-    HIP_CHECK_LAST();
+    // stride > 1: sub-grid class decomposition — one MODE-2 launch per
+    // (a, b) residue class, each summing ONLY its contributing taps
+    // (the divisibility-checked single launch wastes stride^2 x MFMA work
+    // on zero taps: measured 97 vs ~400 TFLOP/s against stride-1 shapes)
+    const int st = cs.stride;
+    // classes with no contributing taps leave their dx sub-grid untouched
+    // (e.g. 1x1/2: only class (0,0) exists) — pre-zero dx if any are empty
+    {
+        bool any_empty = false;
+        for (int a = 0; a < st && !any_empty; ++a) {
+            int nr = 0;
+            for (int r = 0; r < cs.R; ++r)
+                if (((a + cs.pad - r) % st + st) % st == 0) ++nr;
+            if (nr == 0) any_empty = true;
+        }
+        if (!any_empty)
+            for (int b = 0; b < st && !any_empty; ++b) {
+                int ns = 0;
+                for (int sidx = 0; sidx < cs.S; ++sidx)
+                    if (((b + cs.pad - sidx) % st + st) % st == 0) ++ns;
+                if (ns == 0) any_empty = true;
+            }
+        if (any_empty) dx.zero_();
+    }
+    for (int a = 0; a < st; ++a) {
+        for (int b = 0; b < st; ++b) {
+            ConvShape c2 = cs;
+            c2.cls_a = a; c2.cls_b = b;
+            c2.Hfull = cs.H; c2.Wfull = cs.W;
+            c2.nr = 0; c2.ns = 0;
+            for (int r = 0; r < cs.R; ++r)
+                if (((a + cs.pad - r) % st + st) % st == 0) {
+                    c2.taps_r[c2.nr] = r;
+                    // exact division by construction (C division truncates
+                    // toward zero, but the value is an exact multiple)
+                    c2.off_r[c2.nr] = (a + cs.pad - r) / st;
+                    ++c2.nr;
+                }
+            for (int sidx = 0; sidx < cs.S; ++sidx)
+                if (((b + cs.pad - sidx) % st + st) % st == 0) {
+                    c2.taps_s[c2.ns] = sidx;
+                    c2.off_s[c2.ns] = (b + cs.pad - sidx) / st;
+                    ++c2.ns;
+                }
+            if (c2.nr == 0 || c2.ns == 0) continue;  // dx stays... no: must zero
+            int Ha = (cs.H - a + st - 1) / st;
+            int Wb = (cs.W - b + st - 1) / st;
+            c2.H = Ha; c2.W = Wb;           // fdH/fdW decode the class grid
+            c2.fdH.init(Ha); c2.fdW.init(Wb);
+            c2.fdNs.init(c2.ns);
+            long M2 = (long)cs.N * Ha * Wb;
+            int K2 = c2.nr * c2.ns * cs.Ko;
+            if (cs.C >= 128) {
+                dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 128));
+                hipLaunchKernelGGL((k_conv_gemm<2, 128, 2, 2, true>), grid,
+                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
+                                   nullptr, dxp, c2, (int)M2, K2, cs.C);
+            } else {
+                dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 64));
+                hipLaunchKernelGGL((k_conv_gemm<2, 64, 4, 1, true>), grid,
+                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
+                                   nullptr, dxp, c2, (int)M2, K2, cs.C);
+            }
+            HIP_CHECK_LAST();
+        }
+    }
     return dx;
 }
 
