@@ -80,12 +80,50 @@ def bench_functional(shapes, reps):
         print(f"{name:8s} {'Ffb':6s} {us:9.1f} {3*flops/us/1e6:9.1f}")
 
 
+BN_SHAPES = [
+    ("stem", 256, 64, 112, 112),
+    ("l1", 256, 256, 56, 56),
+    ("l1b", 256, 64, 56, 56),
+    ("l2", 256, 512, 28, 28),
+    ("l3", 256, 1024, 14, 14),
+    ("l4", 256, 2048, 7, 7),
+    ("toy2", 1024, 128, 24, 24),
+]
+
+
+def bench_bn(reps):
+    ext = load_extension(required=True)
+    dev = torch.device("cuda:0")
+    print(f"{'shape':6s} {'op':10s} {'us':>9s} {'GB/s':>8s}")
+    for name, N, C, H, W in BN_SHAPES:
+        x = torch.randn(N, C, H, W, device=dev).to(torch.bfloat16).contiguous(memory_format=CL)
+        dy = torch.randn_like(x).contiguous(memory_format=CL)
+        g = torch.rand(C, device=dev) + 0.5
+        b = torch.randn(C, device=dev)
+        rm = torch.zeros(C, device=dev)
+        rv = torch.ones(C, device=dev)
+        nbytes = x.numel() * 2
+        us = time_op(lambda: ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5,
+                                              True, None), reps)
+        # fwd: read x twice (partial+apply) + write y + mask
+        print(f"{name:6s} {'fwd_train':10s} {us:9.1f} {(3.06*nbytes)/us/1e3:8.0f}")
+        y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True, None)
+        us = time_op(lambda: ext.bn_bwd(x, dy, g, sm, si, mask, True, False),
+                     reps)
+        # bwd: partial reads x,dy,mask; dx reads x,dy,mask writes dx
+        print(f"{name:6s} {'bwd':10s} {us:9.1f} {(5.13*nbytes)/us/1e3:8.0f}")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--resnet", action="store_true")
     ap.add_argument("--functional", action="store_true")
+    ap.add_argument("--bn", action="store_true")
     ap.add_argument("--reps", type=int, default=200)
     args = ap.parse_args()
+    if args.bn:
+        bench_bn(args.reps)
+        return
     if args.functional:
         load_extension(required=True)
         bench_functional(TOYNET + (RESNET if args.resnet else []), args.reps)
