@@ -638,6 +638,128 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
         }
 }
 
+// Pair-m wide wgrad: each thread stages TWO consecutive m's per j-group so
+// LDS writes are packed b32 (16 stores/operand/iter vs 32 conflicted b16) —
+// the stage phase of the wide kernel is store-issue bound.
+__global__ __launch_bounds__(256)
+void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
+                            const bf16* __restrict__ x,
+                            float* __restrict__ slab, ConvShape cs, long M,
+                            int Kgemm, int S) {
+    constexpr int DEPTH = 64;
+    __shared__ bf16 lds_a[128][DEPTH + 8];
+    __shared__ bf16 lds_b[128][DEPTH + 8];
+    const int ko0 = blockIdx.x * 128;
+    const int rc0 = blockIdx.y * 128;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[4][4] = {};
+    const int m2 = (tid & 31) * 2;          // m, m+1
+    const int jb = (tid >> 5) * 8;          // 8 groups x 8 = 64 j per pass
+    const long m_begin = (long)split * DEPTH;
+
+    for (long mt = m_begin; mt < M; mt += (long)S * DEPTH) {
+        long gm0 = mt + m2;
+        const bool v0 = gm0 < M, v1 = gm0 + 1 < M;
+        // decode both m's once (x gather bases)
+        unsigned q0 = 0, p0 = 0, n0_ = 0, q1 = 0, p1 = 0, n1_ = 0;
+        if (v0) {
+            unsigned rem = fd_div((unsigned)gm0, cs.fdQ);
+            q0 = fd_mod((unsigned)gm0, cs.fdQ, rem);
+            n0_ = fd_div(rem, cs.fdP);
+            p0 = fd_mod(rem, cs.fdP, n0_);
+        }
+        if (v1) {
+            unsigned rem = fd_div((unsigned)gm0 + 1, cs.fdQ);
+            q1 = fd_mod((unsigned)gm0 + 1, cs.fdQ, rem);
+            n1_ = fd_div(rem, cs.fdP);
+            p1 = fd_mod(rem, cs.fdP, n1_);
+        }
+        #pragma unroll
+        for (int half = 0; half < 2; ++half) {
+            const int j = jb + half * 64;
+            bf16x8_t a0 = {}, a1 = {};
+            if (ko0 + j < cs.Ko) {
+                if (v0) a0 = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[gm0 * cs.Ko + ko0 + j]);
+                if (v1) a1 = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[(gm0 + 1) * cs.Ko + ko0 + j]);
+            }
+            bf16x8_t b0 = {}, b1 = {};
+            int gk = rc0 + j;
+            if (gk < Kgemm) {
+                unsigned rs = fd_div(gk, cs.fdC);
+                int c = fd_mod(gk, cs.fdC, rs);
+                int r = fd_div(rs, cs.fdS);
+                int sx = fd_mod(rs, cs.fdS, r);
+                if (v0) {
+                    int h = (int)p0 * cs.stride + r - cs.pad;
+                    int wcol = (int)q0 * cs.stride + sx - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        b0 = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n0_ * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+                if (v1) {
+                    int h = (int)p1 * cs.stride + r - cs.pad;
+                    int wcol = (int)q1 * cs.stride + sx - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        b1 = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n1_ * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+            }
+            #pragma unroll
+            for (int jj = 0; jj < 8; ++jj) {
+                unsigned pa = (unsigned)(unsigned short)a0[jj]
+                    | ((unsigned)(unsigned short)a1[jj] << 16);
+                unsigned pb = (unsigned)(unsigned short)b0[jj]
+                    | ((unsigned)(unsigned short)b1[jj] << 16);
+                *reinterpret_cast<unsigned*>(&lds_a[j + jj][m2]) = pa;
+                *reinterpret_cast<unsigned*>(&lds_b[j + jj][m2]) = pb;
+            }
+        }
+        __syncthreads();
+
+        #pragma unroll
+        for (int ks = 0; ks < DEPTH; ks += 32) {
+            bf16x8_t af[4], bfr[4];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = rc0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
 // conv1 wgrad, fully specialized C==1 / 3x3 (the MNIST stem): each walker
 // wave sweeps a CONTIGUOUS output range with incremental (n,p,q) tracking —
 // no divisions in the inner loop — and a 3x3 sliding x-window in registers
@@ -1055,7 +1177,11 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
            M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
-    if (use_wide && wide64)
+    if (use_wide && wide64 && wv && wv[1] == 'p')
+        hipLaunchKernelGGL(k_conv_wgrad_wide_pair, dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_wide && wide64)
         hipLaunchKernelGGL((k_conv_wgrad_wide<64>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
