@@ -1177,7 +1177,10 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
            M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
-    if (use_wide && wide64 && wv && wv[1] == 'p')
+    // pair-store wide is the measured default (conv4 373 vs 461 us,
+    // conv3 165 vs 210); DDPX_WGRAD_V=w selects the scalar-store wide
+    const bool wide_pair = !wv || wv[1] == 'p' || wv[1] == 0;
+    if (use_wide && wide64 && wide_pair)
         hipLaunchKernelGGL(k_conv_wgrad_wide_pair, dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
