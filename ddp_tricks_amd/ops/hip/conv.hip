@@ -638,6 +638,115 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
         }
 }
 
+// Pair-m single-buffer 64x64 wgrad: thread halves split the two operands;
+// each thread stages its m-PAIR with packed b32 stores (conflict-free).
+__global__ __launch_bounds__(256)
+void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
+                          const bf16* __restrict__ x,
+                          float* __restrict__ slab, ConvShape cs, long M,
+                          int Kgemm, int S) {
+    __shared__ bf16 lds_a[WBM][SLDK];   // [ko][m]
+    __shared__ bf16 lds_b[WBN][SLDK];   // [rsc][m]
+    const int ko0 = blockIdx.x * WBM;
+    const int rc0 = blockIdx.y * WBN;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[2][2] = {};
+    const int m2 = (tid & 15) * 2;          // m, m+1 (covers 32 m)
+    const int jsel = tid >> 4;              // 0..15
+    const int j8 = (jsel & 7) * 8;          // 8 j-groups x 8 = 64 j
+    const bool is_b = jsel >= 8;            // thread half stages A or B
+
+    const long m_begin = (long)split * SBK;
+    for (long mt = m_begin; mt < M; mt += (long)S * SBK) {
+        long gm0 = mt + m2;
+        const bool v0 = gm0 < M, v1 = gm0 + 1 < M;
+        bf16x8_t t0 = {}, t1 = {};
+        if (!is_b) {
+            if (ko0 + j8 < cs.Ko) {
+                if (v0) t0 = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[gm0 * cs.Ko + ko0 + j8]);
+                if (v1) t1 = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[(gm0 + 1) * cs.Ko + ko0 + j8]);
+            }
+        } else {
+            int gk = rc0 + j8;
+            if (gk < Kgemm) {
+                unsigned rs = fd_div(gk, cs.fdC);
+                int c = fd_mod(gk, cs.fdC, rs);
+                int r = fd_div(rs, cs.fdS);
+                int sx = fd_mod(rs, cs.fdS, r);
+                if (v0) {
+                    unsigned rem = fd_div((unsigned)gm0, cs.fdQ);
+                    int q = fd_mod((unsigned)gm0, cs.fdQ, rem);
+                    unsigned n = fd_div(rem, cs.fdP);
+                    int pp = fd_mod(rem, cs.fdP, n);
+                    int h = pp * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + sx - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        t0 = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+                if (v1) {
+                    unsigned rem = fd_div((unsigned)gm0 + 1, cs.fdQ);
+                    int q = fd_mod((unsigned)gm0 + 1, cs.fdQ, rem);
+                    unsigned n = fd_div(rem, cs.fdP);
+                    int pp = fd_mod(rem, cs.fdP, n);
+                    int h = pp * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + sx - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        t1 = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                }
+            }
+        }
+        bf16* dst = is_b ? &lds_b[0][0] : &lds_a[0][0];
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj) {
+            unsigned pk = (unsigned)(unsigned short)t0[jj]
+                | ((unsigned)(unsigned short)t1[jj] << 16);
+            *reinterpret_cast<unsigned*>(&dst[(j8 + jj) * SLDK + m2]) = pk;
+        }
+        __syncthreads();
+
+        bf16x8_t af[2], bfr[2];
+        const int kcol = (lane >> 4) * 8;
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+            int col = rc0 + wc * 32 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
 // Pair-m wide wgrad: each thread stages TWO consecutive m's per j-group so
 // LDS writes are packed b32 (16 stores/operand/iter vs 32 conflicted b16) —
 // the stage phase of the wide kernel is store-issue bound.
@@ -1191,6 +1300,10 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     else if (use_wide)
         hipLaunchKernelGGL((k_conv_wgrad_wide<32>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_sb && !(wv && wv[1] == 'x'))
+        hipLaunchKernelGGL(k_conv_wgrad_sb_pair, dim3(gk, gr, S_), dim3(256),
+                           0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_sb)
         hipLaunchKernelGGL(k_conv_wgrad_sb, dim3(gk, gr, S_), dim3(256), 0,
