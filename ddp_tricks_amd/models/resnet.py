@@ -19,6 +19,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ..ops.functional import conv_bn
 from ..ops.modules import (
     AdaptiveAvgPool2d, BatchNorm2d, Conv2d, Identity, Linear, MaxPool2d,
 )
@@ -47,8 +48,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        return self.bn2(self.conv2(out), residual=identity)
+        out = conv_bn(self.conv1, self.bn1, x)
+        return conv_bn(self.conv2, self.bn2, out, residual=identity)
 
 
 class Bottleneck(nn.Module):
@@ -68,9 +69,9 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        return self.bn3(self.conv3(out), residual=identity)
+        out = conv_bn(self.conv1, self.bn1, x)
+        out = conv_bn(self.conv2, self.bn2, out)
+        return conv_bn(self.conv3, self.bn3, out, residual=identity)
 
 
 class ResNet(nn.Module):
@@ -116,7 +117,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*mods)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.maxpool(self.bn1(self.conv1(x)))
+        x = self.maxpool(conv_bn(self.conv1, self.bn1, x))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x)
         return self.fc(torch.flatten(x, 1))

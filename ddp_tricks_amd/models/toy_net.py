@@ -49,6 +49,15 @@ class Toy_Net(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        out = self.conv(x)
+        # walk the Sequentials explicitly so each conv->BN pair runs the
+        # fused stats path (parameter keys are unchanged — same children)
+        from ..ops.functional import conv_bn
+        c = self.conv
+        out = conv_bn(c[0], c[1], x)
+        out = conv_bn(c[3], c[4], out)
+        out = c[6](out)
+        out = conv_bn(c[7], c[8], out)
+        out = conv_bn(c[10], c[11], out)
+        out = c[13](out)
         out = self.dense(out)
         return out

@@ -115,6 +115,70 @@ def _use_pad8(C: int, H: int, W: int) -> bool:
     return C < 8 and H * W >= 1024
 
 
+def _conv_fwd_prep(ctx, x, weight, bias, stride, padding):
+    """Shared operand prep for the conv autograd functions; returns
+    (ext, xb_for_kernel, wb, bias_f) and fills ctx bookkeeping."""
+    ext = require_ext_for(x)
+    xb = _chlast(_to_bf16(x))
+    C_in = weight.shape[1]
+    if C_in < 8:
+        pad8_bwd = _use_pad8(C_in, xb.shape[2], xb.shape[3])
+        xp = _pad8(xb)
+        wb = weight_variant(weight, "nhwc_p8")
+        ctx.save_for_backward(xp if pad8_bwd else xb, weight)
+        ctx.meta = (stride, padding, x.dtype, weight.shape,
+                    bias is not None, pad8_bwd)
+        return ext, xp, wb, (bias if bias is None else bias.detach().float())
+    wb = weight_variant(weight, "nhwc")
+    ctx.save_for_backward(xb, weight)
+    ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
+                False)
+    return ext, xb, wb, (bias if bias is None else bias.detach().float())
+
+
+def _conv_bwd_impl(ctx, dy):
+    xb, weight = ctx.saved_tensors
+    stride, padding, x_dtype, w_shape, has_bias, pad8 = ctx.meta
+    ext = require_ext_for(dy)
+    dyb = _chlast(_to_bf16(dy))
+    K, C, R, S = w_shape
+    dx = dw = db = None
+    if ctx.needs_input_grad[0]:
+        wt2 = weight_variant(weight, "wt2_p8" if pad8 else "wt2")
+        dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], xb.shape[1],
+                              xb.shape[2], xb.shape[3], R, S,
+                              stride, padding)
+        if pad8:
+            dx = dx[:, :C]
+        if x_dtype == torch.float32:
+            dx = dx.float()
+    if ctx.needs_input_grad[1]:
+        dw = ext.conv2d_wgrad(dyb, xb, R, S, stride, padding)
+        if pad8:
+            dw = dw[:, :C].contiguous()
+    if has_bias and ctx.needs_input_grad[2]:
+        db = ext.col_sum(_nhwc_2d(dyb))
+    return dx, dw, db
+
+
+class _HIPConv2dStats(torch.autograd.Function):
+    """Conv forward that also emits the consuming BN's partial statistics
+    from the epilogue (stats output is non-differentiable)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding):
+        ext, xk, wb, bias_f = _conv_fwd_prep(ctx, x, weight, bias, stride,
+                                             padding)
+        y, stats = ext.conv2d_fwd_stats(xk, wb, bias_f, stride, padding)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
+
+    @staticmethod
+    def backward(ctx, dy, dstats):
+        dx, dw, db = _conv_bwd_impl(ctx, dy)
+        return dx, dw, db, None, None
+
+
 class _HIPConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
@@ -147,27 +211,7 @@ class _HIPConv2d(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        xb, weight = ctx.saved_tensors
-        stride, padding, x_dtype, w_shape, has_bias, pad8 = ctx.meta
-        ext = require_ext_for(dy)
-        dyb = _chlast(_to_bf16(dy))
-        K, C, R, S = w_shape
-        dx = dw = db = None
-        if ctx.needs_input_grad[0]:
-            wt2 = weight_variant(weight, "wt2_p8" if pad8 else "wt2")
-            dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], xb.shape[1],
-                                  xb.shape[2], xb.shape[3], R, S,
-                                  stride, padding)
-            if pad8:
-                dx = dx[:, :C]
-            if x_dtype == torch.float32:
-                dx = dx.float()
-        if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(dyb, xb, R, S, stride, padding)
-            if pad8:
-                dw = dw[:, :C].contiguous()
-        if has_bias and ctx.needs_input_grad[2]:
-            db = ext.col_sum(_nhwc_2d(dyb))
+        dx, dw, db = _conv_bwd_impl(ctx, dy)
         return dx, dw, db, None, None
 
 
@@ -239,14 +283,14 @@ class _HIPBatchNorm(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                training, momentum, eps, fuse_relu, residual):
+                training, momentum, eps, fuse_relu, residual, pre_stats=None):
         ext = require_ext_for(x)
         xb = _chlast(_to_bf16(x))
         rb = None if residual is None else _chlast(_to_bf16(residual))
         if training:
             y, save_mean, save_invstd, mask = ext.bn_fwd_train(
                 xb, weight.detach(), bias.detach(), running_mean, running_var,
-                momentum, eps, fuse_relu, rb)
+                momentum, eps, fuse_relu, rb, pre_stats)
             ctx.save_for_backward(xb, weight, save_mean, save_invstd, mask)
         else:
             y = ext.bn_fwd_eval(xb, weight.detach(), bias.detach(),
@@ -270,16 +314,20 @@ class _HIPBatchNorm(torch.autograd.Function):
         if ctx.x_dtype == torch.float32:
             dx = dx.float()
         return (dx, dweight, dbias, None, None, None, None, None, None,
-                dresid)
+                dresid, None)
 
 
 def batch_norm(x, running_mean, running_var, weight, bias,
-               training, momentum, eps, fuse_relu=False, residual=None):
+               training, momentum, eps, fuse_relu=False, residual=None,
+               pre_stats=None):
     """BN with optionally fused ReLU and residual add: relu(bn(x) + residual)
-    — the ResNet block epilogue in one kernel (skip grad returned in bwd)."""
+    — the ResNet block epilogue in one kernel (skip grad returned in bwd).
+    pre_stats: [2,C,S] partial sums emitted by the producing conv's epilogue
+    (conv2d_stats) — skips the BN statistics pass entirely."""
     if x.is_cuda and require_ext_for(x) is not None:
         return _HIPBatchNorm.apply(x, weight, bias, running_mean, running_var,
-                                   training, momentum, eps, fuse_relu, residual)
+                                   training, momentum, eps, fuse_relu,
+                                   residual, pre_stats)
     xf = x.float() if x.dtype != torch.float32 else x
     y = F.batch_norm(xf, running_mean, running_var, weight, bias,
                      training, momentum, eps)
@@ -405,3 +453,22 @@ def argmax_correct(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
         ext = require_ext_for(logits)
         return ext.argmax_correct(logits.contiguous(), target)
     return (torch.argmax(logits, dim=1) == target).sum()
+
+
+def conv_bn(conv_mod, bn_mod, x, residual=None):
+    """Fused producer/consumer: on the GPU training path the conv epilogue
+    emits the BN's per-channel partial statistics, and the BN forward skips
+    its stats pass (SURVEY §7 M5 fusion note).  Falls back to the plain
+    composition on CPU, in eval mode, or off the MFMA path."""
+    use_stats = (x.is_cuda and require_ext_for(x) is not None
+                 and bn_mod.training and getattr(conv_mod, "dilation",
+                                                 (1, 1)) in ((1, 1), 1))
+    if use_stats:
+        stride = conv_mod.stride[0] if isinstance(conv_mod.stride, tuple) \
+            else conv_mod.stride
+        padding = conv_mod.padding[0] if isinstance(conv_mod.padding, tuple) \
+            else conv_mod.padding
+        y, stats = _HIPConv2dStats.apply(x, conv_mod.weight, conv_mod.bias,
+                                         stride, padding)
+        return bn_mod(y, residual=residual, pre_stats=stats)
+    return bn_mod(conv_mod(x), residual=residual)

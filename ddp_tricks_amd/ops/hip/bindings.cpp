@@ -33,7 +33,8 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
                                      double eps, bool fuse_relu,
-                                     c10::optional<at::Tensor> residual);
+                                     c10::optional<at::Tensor> residual,
+                                     c10::optional<at::Tensor> pre_stats);
 at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        at::Tensor running_mean, at::Tensor running_var,
                        double eps, bool fuse_relu,
@@ -55,6 +56,9 @@ at::Tensor linear_wgrad(at::Tensor dy, at::Tensor x);
 // conv.hip
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias,
                       long stride, long pad);
+std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
+                                         c10::optional<at::Tensor> bias,
+                                         long stride, long pad);
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C, long H,
                         long W, long R, long S, long stride, long pad);
 at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
@@ -90,7 +94,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("x"), py::arg("gamma"), py::arg("beta"),
           py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
           py::arg("eps"), py::arg("fuse_relu") = false,
-          py::arg("residual") = c10::nullopt);
+          py::arg("residual") = c10::nullopt,
+          py::arg("pre_stats") = c10::nullopt);
     m.def("bn_fwd_eval", &bn_fwd_eval,
           py::arg("x"), py::arg("gamma"), py::arg("beta"),
           py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
@@ -108,6 +113,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("linear_dgrad", &linear_dgrad);
     m.def("linear_wgrad", &linear_wgrad);
     m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
+          py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
+          py::arg("pad") = 0);
+    m.def("conv2d_fwd_stats", &conv2d_fwd_stats, py::arg("x"), py::arg("w"),
           py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
           py::arg("pad") = 0);
     m.def("conv2d_dgrad", &conv2d_dgrad);

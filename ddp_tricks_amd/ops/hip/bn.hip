@@ -295,7 +295,8 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
                                      double eps, bool fuse_relu,
-                                     c10::optional<at::Tensor> residual) {
+                                     c10::optional<at::Tensor> residual,
+                                     c10::optional<at::Tensor> pre_stats) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0 && C / 8 <= 256, "bn HIP path needs C%8==0");
@@ -303,8 +304,12 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
     auto fopts = gamma.options().dtype(at::kFloat);
     int block = pick_block(C);
     int nw = block / (C / 8);
-    int S = bn_splits(M, nw);
-    auto slab = at::empty({2, C, S}, fopts);
+    // pre_stats: [2][C][S] partial sums already produced by the conv
+    // epilogue — the stats pass (a full re-read of x) is skipped.
+    int S = pre_stats.has_value() ? (int)pre_stats->size(2)
+                                  : bn_splits(M, nw);
+    auto slab = pre_stats.has_value() ? *pre_stats
+                                      : at::empty({2, C, S}, fopts);
     auto save_mean = at::empty({C}, fopts);
     auto save_invstd = at::empty({C}, fopts);
     auto scale = at::empty({C}, fopts);
@@ -315,9 +320,12 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
     const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
     int lds = 2 * nw * C * 4;
 
-    hipLaunchKernelGGL(k_bn_partial, dim3(S), dim3(block), lds,
-                       stream.stream(), xp, M, C, S, slab.data_ptr<float>());
-    HIP_CHECK_LAST();
+    if (!pre_stats.has_value()) {
+        hipLaunchKernelGGL(k_bn_partial, dim3(S), dim3(block), lds,
+                           stream.stream(), xp, M, C, S,
+                           slab.data_ptr<float>());
+        HIP_CHECK_LAST();
+    }
     hipLaunchKernelGGL(k_bn_combine, dim3(ceil_div_i(C, 4)), dim3(256), 0,
                        stream.stream(), slab.data_ptr<float>(), S, C, M,
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),

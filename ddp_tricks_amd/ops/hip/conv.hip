@@ -66,7 +66,8 @@ template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
-                 ConvShape cs, int M, int Kgemm, int Nout) {
+                 ConvShape cs, int M, int Kgemm, int Nout,
+                 float* __restrict__ stats = nullptr) {
     constexpr int WM = CBM / WAVES_M;
     constexpr int WN = TBN / WAVES_N;
     constexpr int MI = WM / 16;
@@ -220,6 +221,7 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
         __syncthreads();
     }
 
+    float psum[NI] = {}, psq[NI] = {};
     #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
         #pragma unroll
@@ -231,6 +233,11 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             for (int r = 0; r < 4; ++r) {
                 int row = m0 + wr * WM + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= M) continue;
+                if (MODE == 0 && stats) {
+                    float v = acc[mi][ni][r] + badd;
+                    psum[ni] += v;
+                    psq[ni] = fmaf(v, v, psq[ni]);
+                }
                 if (MODE == 2) {
                     // scatter the class sub-grid back into full dx:
                     // (n, h', w') -> (a + h'*stride, b + w'*stride)
@@ -248,6 +255,42 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                 }
             }
         }
+
+    if (MODE == 0 && stats) {
+        // Per-channel partial sum/sumsq of this block's 128 output rows —
+        // feeds the BN that consumes this conv (its stats pass never
+        // re-reads y).  slab layout matches k_bn_combine: [2][Nout][gridM].
+        __syncthreads();                       // done reading lds tiles
+        float* red = reinterpret_cast<float*>(&lds_a[0][0][0]);  // [WAVES_M][TBN][2]
+        #pragma unroll
+        for (int ni = 0; ni < NI; ++ni) {
+            float a = psum[ni], b = psq[ni];
+            a += __shfl_down(a, 32, 64);
+            a += __shfl_down(a, 16, 64);
+            b += __shfl_down(b, 32, 64);
+            b += __shfl_down(b, 16, 64);
+            if ((lane & 48) == 0) {            // lanes 0..15 hold the totals
+                int cl = wc * WN + ni * 16 + (lane & 15);
+                red[(wr * TBN + cl) * 2] = a;
+                red[(wr * TBN + cl) * 2 + 1] = b;
+            }
+        }
+        __syncthreads();
+        const int gridM = gridDim.x;
+        for (int cl = tid; cl < TBN; cl += 256) {
+            float a = 0.f, b = 0.f;
+            #pragma unroll
+            for (int w = 0; w < WAVES_M; ++w) {
+                a += red[(w * TBN + cl) * 2];
+                b += red[(w * TBN + cl) * 2 + 1];
+            }
+            int col = n0 + cl;
+            if (col < Nout) {
+                stats[(long)col * gridM + blockIdx.x] = a;
+                stats[(long)Nout * gridM + (long)col * gridM + blockIdx.x] = b;
+            }
+        }
+    }
 }
 
 // conv with Cin < 8 (e.g. the MNIST stem, Cin=1): direct VALU kernel,
@@ -1058,6 +1101,10 @@ static ConvShape make_shape(const at::Tensor& x, int Ko, int R, int S,
     return cs;
 }
 
+std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
+                                         c10::optional<at::Tensor> bias,
+                                         long stride, long pad);
+
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
                       c10::optional<at::Tensor> bias, long stride, long pad) {
     // x: NCHW logical / channels_last physical bf16; w: KCRS logical /
@@ -1321,4 +1368,41 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
                        dw.data_ptr<float>());
     HIP_CHECK_LAST();
     return dw;
+}
+
+
+std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
+                                         c10::optional<at::Tensor> bias,
+                                         long stride, long pad) {
+    // forward conv + per-channel partial stats for the consuming BN
+    // (SURVEY §7 "fuse normalisation work into the producing kernel")
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+    ConvShape cs = make_shape(x, w.size(0), w.size(2), w.size(3),
+                              (int)stride, (int)pad);
+    TORCH_CHECK(cs.C % 8 == 0, "stats path needs the MFMA conv (C%8==0)");
+    int Kgemm = cs.R * cs.S * cs.C;
+    long M = (long)cs.N * cs.P * cs.Q;
+    auto y = at::empty({cs.N, cs.Ko, cs.P, cs.Q},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = at::hip::getCurrentHIPStream();
+    const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+    const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+    const bf16* wp = reinterpret_cast<const bf16*>(w.data_ptr());
+    bf16* yp = reinterpret_cast<bf16*>(y.data_ptr());
+    int gridM = ceil_div_i(M, CBM);
+    auto slab = at::empty({2, (long)cs.Ko, (long)gridM},
+                          x.options().dtype(at::kFloat));
+    if (cs.Ko >= 128) {
+        dim3 grid(gridM, ceil_div_i(cs.Ko, 128));
+        hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid, dim3(256), 0,
+                           stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                           Kgemm, cs.Ko, slab.data_ptr<float>());
+    } else {
+        dim3 grid(gridM, ceil_div_i(cs.Ko, 64));
+        hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
+                           stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                           Kgemm, cs.Ko, slab.data_ptr<float>());
+    }
+    HIP_CHECK_LAST();
+    return {y, slab};
 }

@@ -26,7 +26,7 @@ class Linear(nn.Linear):
 
 
 class _BatchNormBase:
-    def _bn_forward(self, x, residual=None):
+    def _bn_forward(self, x, residual=None, pre_stats=None):
         self._check_input_dim(x)
         if self.training:
             if self.num_batches_tracked is not None:
@@ -36,7 +36,7 @@ class _BatchNormBase:
         return F_ops.batch_norm(
             x, self.running_mean, self.running_var, self.weight, self.bias,
             self.training or not self.track_running_stats, momentum, self.eps,
-            fuse_relu=fuse_relu, residual=residual)
+            fuse_relu=fuse_relu, residual=residual, pre_stats=pre_stats)
 
 
 class BatchNorm2d(nn.BatchNorm2d, _BatchNormBase):
@@ -44,10 +44,11 @@ class BatchNorm2d(nn.BatchNorm2d, _BatchNormBase):
         super().__init__(*args, **kwargs)
         self.fuse_relu = fuse_relu
 
-    def forward(self, x, residual=None):
+    def forward(self, x, residual=None, pre_stats=None):
         # residual: optional skip tensor added before the (fused) ReLU —
-        # relu(bn(x) + residual) in one kernel on GPU (ResNet blocks)
-        return self._bn_forward(x, residual)
+        # relu(bn(x) + residual) in one kernel on GPU (ResNet blocks);
+        # pre_stats: partial statistics from the producing conv's epilogue
+        return self._bn_forward(x, residual, pre_stats)
 
 
 class BatchNorm1d(nn.BatchNorm1d, _BatchNormBase):

@@ -502,3 +502,38 @@ def test_pack_conv_weight():
         assert nhwc.shape == ref_nhwc.shape
         assert torch.equal(nhwc.float(), ref_nhwc.float())
         assert torch.equal(wt2.float(), ref_wt2.float())
+
+
+def test_conv_bn_fused_stats_matches_plain():
+    """conv_bn (conv-epilogue partial stats -> BN skips its stats pass)
+    must match the unfused conv->bn composition exactly in both outputs
+    and running-stat updates."""
+    from ddp_tricks_amd.ops.functional import conv_bn
+    from ddp_tricks_amd.ops.modules import BatchNorm2d, Conv2d
+    torch.manual_seed(21)
+    for C, K, H, stride, pad, bias in [(64, 128, 24, 1, 0, True),
+                                       (3, 64, 32, 1, 1, False),
+                                       (64, 128, 32, 2, 1, False)]:
+        conv = Conv2d(C, K, 3, stride=stride, padding=pad, bias=bias).to(DEV)
+        bn_a = BatchNorm2d(K, fuse_relu=True).to(DEV)
+        bn_b = BatchNorm2d(K, fuse_relu=True).to(DEV)
+        bn_b.load_state_dict(bn_a.state_dict())
+        conv.train(); bn_a.train(); bn_b.train()
+        x = torch.randn(16, C, H, H, device=DEV).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        y_fused = conv_bn(conv, bn_a, x)
+        y_plain = bn_b(conv(x))
+        # identical kernels downstream of identical stats => near-equal;
+        # stats come from fp32 accum vs bf16 y: tiny quantization delta
+        _close(y_fused, y_plain, rel=1e-2, atol=2e-2,
+               name=f"conv_bn {C}->{K}")
+        _close(bn_a.running_mean, bn_b.running_mean, rel=1e-2, atol=1e-3,
+               name="rmean")
+        _close(bn_a.running_var, bn_b.running_var, rel=1e-2, atol=1e-3,
+               name="rvar")
+        # gradients flow through the fused pair
+        xg = x.float().requires_grad_(True)
+        out = conv_bn(conv, bn_a, xg)
+        out.sum().backward()
+        assert conv.weight.grad is not None and bn_a.weight.grad is not None
+        assert xg.grad is not None
