@@ -50,3 +50,37 @@ def test_train_e2e_cpu(tmp_path, monkeypatch, capsys):
                if os.path.isdir(os.path.join(logdir, d))}
     assert {"Loss_train", "Loss_valid", "Acc_train", "Acc_valid"} <= subdirs
     dist.destroy_process_group()
+
+
+def test_validate_tool(tmp_path, monkeypatch):
+    """tools/validate.py evaluates a saved best checkpoint."""
+    import subprocess
+    import sys
+    import types
+
+    import torch.distributed as dist
+
+    from ddp_tricks_amd import amp
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "256")
+    monkeypatch.setenv("DDPX_NO_TQDM", "1")
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29684")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    amp._state.__init__()
+    from ddp_tricks_amd.utils.train import train
+    args = types.SimpleNamespace(
+        exp_name="VAL", learning_rate=0.05, batch_size=64, epochs=1,
+        warmup_epochs=1, warmup_type="linear", seed_num=42,
+        data_path="/nonexistent", model_path=str(tmp_path), local_rank=0)
+    train(args)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, DDPX_SYNTH_SAMPLES="256")
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "validate.py"),
+         "-n", "VAL", "-p", str(tmp_path), "-d", "/nonexistent"],
+        capture_output=True, text=True, timeout=300, env=env, cwd=repo)
+    assert r.returncode == 0, r.stderr
+    assert "valid_acc" in r.stdout
