@@ -124,6 +124,34 @@ __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
                               unsigned char* __restrict__ mask) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
+    // 2-deep unroll: both 16 B loads in flight before either is consumed
+    // (single-load iterations leave HBM latency exposed — ~5 of 8 TB/s)
+    for (; i + stride < total_v; i += 2 * stride) {
+        long i2 = i + stride;
+        s16x8 v0 = reinterpret_cast<const s16x8*>(x)[i];
+        s16x8 v1 = reinterpret_cast<const s16x8*>(x)[i2];
+        s16x8 r0 = resid ? reinterpret_cast<const s16x8*>(resid)[i] : s16x8{};
+        s16x8 r1 = resid ? reinterpret_cast<const s16x8*>(resid)[i2] : s16x8{};
+        int cv0 = (i % Cv) * 8, cv1 = (i2 % Cv) * 8;
+        s16x8 o0, o1;
+        unsigned m0 = 0, m1 = 0;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = fmaf(us2f((unsigned short)v0[j]), scale[cv0 + j],
+                           shift[cv0 + j]);
+            if (resid) f += us2f((unsigned short)r0[j]);
+            if (relu) { if (f > 0.f) m0 |= 1u << j; f = fmaxf(f, 0.f); }
+            o0[j] = (short)f2us(f);
+            float g = fmaf(us2f((unsigned short)v1[j]), scale[cv1 + j],
+                           shift[cv1 + j]);
+            if (resid) g += us2f((unsigned short)r1[j]);
+            if (relu) { if (g > 0.f) m1 |= 1u << j; g = fmaxf(g, 0.f); }
+            o1[j] = (short)f2us(g);
+        }
+        reinterpret_cast<s16x8*>(y)[i] = o0;
+        reinterpret_cast<s16x8*>(y)[i2] = o1;
+        if (mask) { mask[i] = (unsigned char)m0; mask[i2] = (unsigned char)m1; }
+    }
     for (; i < total_v; i += stride) {
         int cv = (i % Cv) * 8;
         s16x8 v = reinterpret_cast<const s16x8*>(x)[i];
@@ -168,7 +196,29 @@ __global__ void k_bn_bwd_partial(const bf16* __restrict__ x,
     }
     float sum_dy[8] = {}, sum_dyx[8] = {};
     const int cpgv = cpg;
-    for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
+    const long rstep = (long)S * nw;
+    long r = (long)s * nw + walker;
+    for (; r + rstep < M; r += 2 * rstep) {
+        long r2 = r + rstep;
+        s16x8 vx0 = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg0 = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        s16x8 vx1 = reinterpret_cast<const s16x8*>(x + r2 * C)[c8];
+        s16x8 vg1 = reinterpret_cast<const s16x8*>(dy + r2 * C)[c8];
+        unsigned m0 = relu ? mask[r * cpgv + c8] : 0xffu;
+        unsigned m1 = relu ? mask[r2 * cpgv + c8] : 0xffu;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = (m0 >> j) & 1u ? us2f((unsigned short)vg0[j]) : 0.f;
+            float xh = (us2f((unsigned short)vx0[j]) - mean[j]) * invstd[j];
+            sum_dy[j] += g;
+            sum_dyx[j] = fmaf(g, xh, sum_dyx[j]);
+            float g1 = (m1 >> j) & 1u ? us2f((unsigned short)vg1[j]) : 0.f;
+            float xh1 = (us2f((unsigned short)vx1[j]) - mean[j]) * invstd[j];
+            sum_dy[j] += g1;
+            sum_dyx[j] = fmaf(g1, xh1, sum_dyx[j]);
+        }
+    }
+    for (; r < M; r += rstep) {
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
         unsigned m = relu ? mask[r * cpgv + c8] : 0xffu;
@@ -254,7 +304,35 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
     }
     long r0 = (long)blockIdx.x * nw + walker;
     long rstride = (long)gridDim.x * nw;
-    for (long r = r0; r < M; r += rstride) {
+    long r = r0;
+    for (; r + rstride < M; r += 2 * rstride) {
+        long r2 = r + rstride;
+        s16x8 vx0 = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg0 = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        s16x8 vx1 = reinterpret_cast<const s16x8*>(x + r2 * C)[c8];
+        s16x8 vg1 = reinterpret_cast<const s16x8*>(dy + r2 * C)[c8];
+        unsigned m0 = relu ? mask[r * cpg + c8] : 0xffu;
+        unsigned m1 = relu ? mask[r2 * cpg + c8] : 0xffu;
+        s16x8 o0, og0, o1, og1;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = (m0 >> j) & 1u ? us2f((unsigned short)vg0[j]) : 0.f;
+            float xh = (us2f((unsigned short)vx0[j]) - mean[j]) * invstd[j];
+            o0[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
+            og0[j] = (short)f2us(g);
+            float g1 = (m1 >> j) & 1u ? us2f((unsigned short)vg1[j]) : 0.f;
+            float xh1 = (us2f((unsigned short)vx1[j]) - mean[j]) * invstd[j];
+            o1[j] = (short)f2us(ca[j] * (g1 - cb[j] - xh1 * cc[j]));
+            og1[j] = (short)f2us(g1);
+        }
+        reinterpret_cast<s16x8*>(dx + r * C)[c8] = o0;
+        reinterpret_cast<s16x8*>(dx + r2 * C)[c8] = o1;
+        if (dresid) {
+            reinterpret_cast<s16x8*>(dresid + r * C)[c8] = og0;
+            reinterpret_cast<s16x8*>(dresid + r2 * C)[c8] = og1;
+        }
+    }
+    for (; r < M; r += rstride) {
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
         s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
         unsigned m = relu ? mask[r * cpg + c8] : 0xffu;
