@@ -13,7 +13,6 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include <climits>
 #include "common.h"
 
 typedef short bf16x8_t __attribute__((ext_vector_type(8)));
@@ -94,97 +93,71 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     const int pl_row = lane >> 3;
     const int pl_segp = lane & 7;          // physical segment (LDS-linear)
 
-    // Each thread's A-gather rows are FIXED across every k-tile (chunk
-    // assignment depends only on wid/lane), so the m-side decode runs ONCE
-    // per block into registers; the k-tile loop only decodes gk.
-    constexpr int A_PER = (A_CHUNKS + 3) / 4;
-    struct RowInfo { long nb; int u, v; };   // meaning depends on MODE
-    RowInfo rinfo[A_PER];
-    {
-        int idx = 0;
-        for (int ch = wid; ch < A_CHUNKS; ch += 4, ++idx) {
+    auto issue_tile = [&](int kt, int buf) {
+        // ---- A tile: chunks round-robined over the 4 waves ----
+        for (int ch = wid; ch < A_CHUNKS; ch += 4) {
             int row = ch * 8 + pl_row;
-            int gm = m0 + row;
-            rinfo[idx].u = INT_MIN;            // invalid marker
-            if (gm < M) {
+            int seg = pl_segp ^ (row & 7);     // logical k-segment
+            int gm = m0 + row, gk = kt + seg * 8;
+            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
+            if (gm < M && gk < Kgemm) {
                 if (MODE == 0) {
                     unsigned rem = fd_div(gm, cs.fdQ);
                     int q = fd_mod(gm, cs.fdQ, rem);
                     unsigned n = fd_div(rem, cs.fdP);
                     int p = fd_mod(rem, cs.fdP, n);
-                    rinfo[idx].nb = (long)n * cs.H;
-                    rinfo[idx].u = p * cs.stride - cs.pad;   // h0
-                    rinfo[idx].v = q * cs.stride - cs.pad;   // w0
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
+                    int h = p * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + s - cs.pad;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
                 } else if (MODE == 2) {
+                    // sub-grid class: gm -> (n, h', w') over the class's
+                    // Ha x Wb grid (fdH/fdW hold Ha/Wb); gk -> (ri,si,ko)
                     unsigned rem = fd_div(gm, cs.fdW);
                     int w1 = fd_mod(gm, cs.fdW, rem);
                     unsigned n = fd_div(rem, cs.fdH);
                     int h1 = fd_mod(rem, cs.fdH, n);
-                    rinfo[idx].nb = (long)n * cs.P;
-                    rinfo[idx].u = h1;
-                    rinfo[idx].v = w1;
+                    unsigned rs2 = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs2);
+                    int ri = fd_div(rs2, cs.fdNs);
+                    int si = fd_mod(rs2, cs.fdNs, ri);
+                    int p = h1 + cs.off_r[ri];
+                    int q = w1 + cs.off_s[si];
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
+                } else if (STRIDE1) {
+                    unsigned rem = fd_div(gm, cs.fdW);
+                    int wcol = fd_mod(gm, cs.fdW, rem);
+                    unsigned n = fd_div(rem, cs.fdH);
+                    int h = fd_mod(rem, cs.fdH, n);
+                    unsigned rs = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int s = fd_mod(rs, cs.fdS, r);
+                    int p = h + cs.pad - r;
+                    int q = wcol + cs.pad - s;
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 } else {
                     unsigned rem = fd_div(gm, cs.fdW);
                     int wcol = fd_mod(gm, cs.fdW, rem);
                     unsigned n = fd_div(rem, cs.fdH);
                     int h = fd_mod(rem, cs.fdH, n);
-                    rinfo[idx].nb = (long)n * cs.P;
-                    rinfo[idx].u = h + cs.pad;
-                    rinfo[idx].v = wcol + cs.pad;
-                }
-            }
-        }
-    }
-
-    auto issue_tile = [&](int kt, int buf) {
-        // ---- A tile: chunks round-robined over the 4 waves ----
-        int idx = 0;
-        for (int ch = wid; ch < A_CHUNKS; ch += 4, ++idx) {
-            int row = ch * 8 + pl_row;
-            int seg = pl_segp ^ (row & 7);     // logical k-segment
-            int gk = kt + seg * 8;
-            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
-            if (rinfo[idx].u != INT_MIN && gk < Kgemm) {
-                const RowInfo& R_ = rinfo[idx];
-                if (MODE == 0) {
-                    unsigned rs = fd_div(gk, cs.fdC);
-                    int c = fd_mod(gk, cs.fdC, rs);
-                    int r = fd_div(rs, cs.fdS);
-                    int s = fd_mod(rs, cs.fdS, r);
-                    int h = R_.u + r;
-                    int wcol = R_.v + s;
-                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        src = &Asrc[((R_.nb + h) * cs.W + wcol) * cs.C + c];
-                } else if (MODE == 2) {
-                    unsigned rs2 = fd_div(gk, cs.fdKo);
-                    int ko = fd_mod(gk, cs.fdKo, rs2);
-                    int ri = fd_div(rs2, cs.fdNs);
-                    int si = fd_mod(rs2, cs.fdNs, ri);
-                    int p = R_.u + cs.off_r[ri];
-                    int q = R_.v + cs.off_s[si];
-                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[((R_.nb + p) * cs.Q + q) * cs.Ko + ko];
-                } else if (STRIDE1) {
                     unsigned rs = fd_div(gk, cs.fdKo);
                     int ko = fd_mod(gk, cs.fdKo, rs);
                     int r = fd_div(rs, cs.fdS);
                     int s = fd_mod(rs, cs.fdS, r);
-                    int p = R_.u - r;
-                    int q = R_.v - s;
-                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[((R_.nb + p) * cs.Q + q) * cs.Ko + ko];
-                } else {
-                    unsigned rs = fd_div(gk, cs.fdKo);
-                    int ko = fd_mod(gk, cs.fdKo, rs);
-                    int r = fd_div(rs, cs.fdS);
-                    int s = fd_mod(rs, cs.fdS, r);
-                    int pn = R_.u - r;
-                    int qn = R_.v - s;
+                    int pn = h + cs.pad - r;
+                    int qn = wcol + cs.pad - s;
                     // strided conv: only taps where stride divides contribute
                     int p = pn / cs.stride, q = qn / cs.stride;
                     if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
                         qn == q * cs.stride && p < cs.P && q < cs.Q)
-                        src = &Asrc[((R_.nb + p) * cs.Q + q) * cs.Ko + ko];
+                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
                 }
             }
             __builtin_amdgcn_global_load_lds(
