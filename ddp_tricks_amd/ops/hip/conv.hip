@@ -13,6 +13,7 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <climits>
 #include "common.h"
 
 typedef short bf16x8_t __attribute__((ext_vector_type(8)));
@@ -93,71 +94,97 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
     const int pl_row = lane >> 3;
     const int pl_segp = lane & 7;          // physical segment (LDS-linear)
 
+    // Each thread's A-gather rows are FIXED across every k-tile (the chunk
+    // assignment depends only on wid/lane), so the m-side decode runs ONCE
+    // into registers; loops are fully unrolled so the per-row array keeps
+    // CONSTANT indices (a runtime-indexed version spilled to scratch and
+    // ran 40-100% slower — kept here as the measured counterexample).
+    constexpr int A_PER = A_CHUNKS / 4;
+    long ri_nb[A_PER];
+    int ri_u[A_PER], ri_v[A_PER];
+    #pragma unroll
+    for (int ci = 0; ci < A_PER; ++ci) {
+        const int ch = wid + ci * 4;
+        const int row = ch * 8 + pl_row;
+        const int gm = m0 + row;
+        ri_u[ci] = INT_MIN;
+        if (gm < M) {
+            if (MODE == 0) {
+                unsigned rem = fd_div(gm, cs.fdQ);
+                int q = fd_mod(gm, cs.fdQ, rem);
+                unsigned n = fd_div(rem, cs.fdP);
+                int p = fd_mod(rem, cs.fdP, n);
+                ri_nb[ci] = (long)n * cs.H;
+                ri_u[ci] = p * cs.stride - cs.pad;   // h0
+                ri_v[ci] = q * cs.stride - cs.pad;   // w0
+            } else if (MODE == 2) {
+                unsigned rem = fd_div(gm, cs.fdW);
+                int w1 = fd_mod(gm, cs.fdW, rem);
+                unsigned n = fd_div(rem, cs.fdH);
+                int h1 = fd_mod(rem, cs.fdH, n);
+                ri_nb[ci] = (long)n * cs.P;
+                ri_u[ci] = h1;
+                ri_v[ci] = w1;
+            } else {
+                unsigned rem = fd_div(gm, cs.fdW);
+                int wcol = fd_mod(gm, cs.fdW, rem);
+                unsigned n = fd_div(rem, cs.fdH);
+                int h = fd_mod(rem, cs.fdH, n);
+                ri_nb[ci] = (long)n * cs.P;
+                ri_u[ci] = h + cs.pad;
+                ri_v[ci] = wcol + cs.pad;
+            }
+        }
+    }
+
     auto issue_tile = [&](int kt, int buf) {
         // ---- A tile: chunks round-robined over the 4 waves ----
-        for (int ch = wid; ch < A_CHUNKS; ch += 4) {
+        #pragma unroll
+        for (int ci = 0; ci < A_PER; ++ci) {
+            const int ch = wid + ci * 4;
             int row = ch * 8 + pl_row;
             int seg = pl_segp ^ (row & 7);     // logical k-segment
-            int gm = m0 + row, gk = kt + seg * 8;
+            int gk = kt + seg * 8;
             const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
-            if (gm < M && gk < Kgemm) {
+            if (ri_u[ci] != INT_MIN && gk < Kgemm) {
                 if (MODE == 0) {
-                    unsigned rem = fd_div(gm, cs.fdQ);
-                    int q = fd_mod(gm, cs.fdQ, rem);
-                    unsigned n = fd_div(rem, cs.fdP);
-                    int p = fd_mod(rem, cs.fdP, n);
                     unsigned rs = fd_div(gk, cs.fdC);
                     int c = fd_mod(gk, cs.fdC, rs);
                     int r = fd_div(rs, cs.fdS);
-                    int s = fd_mod(rs, cs.fdS, r);
-                    int h = p * cs.stride + r - cs.pad;
-                    int wcol = q * cs.stride + s - cs.pad;
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int h = ri_u[ci] + r;
+                    int wcol = ri_v[ci] + sx;
                     if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        src = &Asrc[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c];
+                        src = &Asrc[((ri_nb[ci] + h) * cs.W + wcol) * cs.C + c];
                 } else if (MODE == 2) {
-                    // sub-grid class: gm -> (n, h', w') over the class's
-                    // Ha x Wb grid (fdH/fdW hold Ha/Wb); gk -> (ri,si,ko)
-                    unsigned rem = fd_div(gm, cs.fdW);
-                    int w1 = fd_mod(gm, cs.fdW, rem);
-                    unsigned n = fd_div(rem, cs.fdH);
-                    int h1 = fd_mod(rem, cs.fdH, n);
                     unsigned rs2 = fd_div(gk, cs.fdKo);
                     int ko = fd_mod(gk, cs.fdKo, rs2);
-                    int ri = fd_div(rs2, cs.fdNs);
-                    int si = fd_mod(rs2, cs.fdNs, ri);
-                    int p = h1 + cs.off_r[ri];
-                    int q = w1 + cs.off_s[si];
+                    int ti = fd_div(rs2, cs.fdNs);
+                    int si = fd_mod(rs2, cs.fdNs, ti);
+                    int p = ri_u[ci] + cs.off_r[ti];
+                    int q = ri_v[ci] + cs.off_s[si];
                     if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
                 } else if (STRIDE1) {
-                    unsigned rem = fd_div(gm, cs.fdW);
-                    int wcol = fd_mod(gm, cs.fdW, rem);
-                    unsigned n = fd_div(rem, cs.fdH);
-                    int h = fd_mod(rem, cs.fdH, n);
                     unsigned rs = fd_div(gk, cs.fdKo);
                     int ko = fd_mod(gk, cs.fdKo, rs);
                     int r = fd_div(rs, cs.fdS);
-                    int s = fd_mod(rs, cs.fdS, r);
-                    int p = h + cs.pad - r;
-                    int q = wcol + cs.pad - s;
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int p = ri_u[ci] - r;
+                    int q = ri_v[ci] - sx;
                     if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
                 } else {
-                    unsigned rem = fd_div(gm, cs.fdW);
-                    int wcol = fd_mod(gm, cs.fdW, rem);
-                    unsigned n = fd_div(rem, cs.fdH);
-                    int h = fd_mod(rem, cs.fdH, n);
                     unsigned rs = fd_div(gk, cs.fdKo);
                     int ko = fd_mod(gk, cs.fdKo, rs);
                     int r = fd_div(rs, cs.fdS);
-                    int s = fd_mod(rs, cs.fdS, r);
-                    int pn = h + cs.pad - r;
-                    int qn = wcol + cs.pad - s;
-                    // strided conv: only taps where stride divides contribute
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int pn = ri_u[ci] - r;
+                    int qn = ri_v[ci] - sx;
                     int p = pn / cs.stride, q = qn / cs.stride;
                     if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
                         qn == q * cs.stride && p < cs.P && q < cs.Q)
-                        src = &Asrc[(((long)n * cs.P + p) * cs.Q + q) * cs.Ko + ko];
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
                 }
             }
             __builtin_amdgcn_global_load_lds(
