@@ -63,26 +63,27 @@ static inline void init_fastdiv(ConvShape& cs) {
 // that leaves LDS deterministic).
 __device__ __align__(16) unsigned char g_zero16[16];
 
-template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true>
+template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
+          int CBM_T = CBM>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
                  ConvShape cs, int M, int Kgemm, int Nout,
                  float* __restrict__ stats = nullptr) {
-    constexpr int WM = CBM / WAVES_M;
+    constexpr int WM = CBM_T / WAVES_M;
     constexpr int WN = TBN / WAVES_N;
     constexpr int MI = WM / 16;
     constexpr int NI = WN / 16;
-    constexpr int A_CHUNKS = CBM / 8;      // 1 KiB glds chunks (8 rows)
+    constexpr int A_CHUNKS = CBM_T / 8;    // 1 KiB glds chunks (8 rows)
     constexpr int B_CHUNKS = TBN / 8;
     // Single-buffer glds staging.  A 2-buffer prefetch ring was tried and
     // REVERTED: doubling LDS (64 KB) halved resident blocks/CU and measured
     // 10-30% SLOWER at every conv shape — this kernel hides HBM latency
     // with block-level parallelism (grids >> 256 workgroups), unlike the
     // guide's 256^2-tile GEMM that runs ~1 block/CU.
-    __shared__ bf16 lds_a[1][CBM][CBK];    // unpadded: glds dest is linear
+    __shared__ bf16 lds_a[1][CBM_T][CBK];  // unpadded: glds dest is linear
     __shared__ bf16 lds_b[1][TBN][CBK];
-    const int m0 = blockIdx.x * CBM;
+    const int m0 = blockIdx.x * CBM_T;
     const int n0 = blockIdx.y * TBN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1164,10 +1165,20 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
                                stream.stream(), xp, wp, bp, yp, cs, (int)M,
                                Kgemm, cs.Ko);
         } else {
-            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
-            hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
-                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
-                               Kgemm, cs.Ko);
+            // narrow-output (Ko<128) fwd: 256-row tile so the A-gather
+            // decode amortizes over more MFMA work per block
+            static const char* cv = getenv("DDPX_CONV64_V");
+            if (!(cv && cv[0] == 'n') && M >= 256 * 64) {
+                dim3 grid(ceil_div_i(M, 256), ceil_div_i(cs.Ko, 64));
+                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1, true, 256>),
+                                   grid, dim3(256), 0, stream.stream(), xp,
+                                   wp, bp, yp, cs, (int)M, Kgemm, cs.Ko);
+            } else {
+                dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
+                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid,
+                                   dim3(256), 0, stream.stream(), xp, wp, bp,
+                                   yp, cs, (int)M, Kgemm, cs.Ko);
+            }
         }
     }
     HIP_CHECK_LAST();
