@@ -1212,6 +1212,11 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
             hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
                                nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+        } else if (M >= 256 * 64) {
+            dim3 grid(ceil_div_i(M, 256), ceil_div_i(cs.C, 64));
+            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true, 256>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
         } else {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
             hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
