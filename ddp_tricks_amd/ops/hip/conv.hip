@@ -1165,20 +1165,14 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
                                stream.stream(), xp, wp, bp, yp, cs, (int)M,
                                Kgemm, cs.Ko);
         } else {
-            // narrow-output (Ko<128) fwd: 256-row tile so the A-gather
-            // decode amortizes over more MFMA work per block
-            static const char* cv = getenv("DDPX_CONV64_V");
-            if (!(cv && cv[0] == 'n') && M >= 256 * 64) {
-                dim3 grid(ceil_div_i(M, 256), ceil_div_i(cs.Ko, 64));
-                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1, true, 256>),
-                                   grid, dim3(256), 0, stream.stream(), xp,
-                                   wp, bp, yp, cs, (int)M, Kgemm, cs.Ko);
-            } else {
-                dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
-                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid,
-                                   dim3(256), 0, stream.stream(), xp, wp, bp,
-                                   yp, cs, (int)M, Kgemm, cs.Ko);
-            }
+            // NOTE: a 256-row tile for the narrow (Ko<128) case was tried
+            // and REVERTED — its 40 KB LDS drops residency 6 -> 4 blocks/CU
+            // and measured 20-45% slower (same lesson as the 2-buf glds
+            // ring: this kernel lives on block-level parallelism).
+            dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
+            hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
+                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                               Kgemm, cs.Ko);
         }
     }
     HIP_CHECK_LAST();
@@ -1210,11 +1204,6 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
         if (cs.C >= 128) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
             hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
-                               dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
-        } else if (M >= 256 * 64) {
-            dim3 grid(ceil_div_i(M, 256), ceil_div_i(cs.C, 64));
-            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true, 256>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
                                nullptr, dxp, cs, (int)M, Kgemm, cs.C);
         } else {
