@@ -537,3 +537,44 @@ def test_conv_bn_fused_stats_matches_plain():
         out.sum().backward()
         assert conv.weight.grad is not None and bn_a.weight.grad is not None
         assert xg.grad is not None
+
+
+def test_conv_shape_fuzz():
+    """Sweep a spread of conv shapes (odd spatial, strides, pads, channel
+    widths) through fwd/dgrad/wgrad vs torch fp32 — guards shapes no named
+    test pins down (ResNet-34 variants, future models)."""
+    import random
+    rng = random.Random(1234)
+    torch.manual_seed(99)
+    for trial in range(12):
+        C = rng.choice([8, 16, 24, 64, 96, 128])
+        K = rng.choice([8, 32, 64, 128, 192, 256])
+        R = rng.choice([1, 3, 5])
+        stride = rng.choice([1, 2])
+        pad = rng.choice([0, 1, 2]) if R > 1 else 0
+        H = rng.choice([7, 9, 14, 17, 28, 33])
+        if H + 2 * pad < R:
+            continue
+        N = rng.choice([2, 3, 8])
+        x = torch.randn(N, C, H, H, device=DEV).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        w = torch.randn(K, C, R, R, device=DEV).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        P = (H + 2 * pad - R) // stride + 1
+        if P < 1:
+            continue
+        tag = f"fuzz{trial} N{N} C{C} K{K} R{R} s{stride} p{pad} H{H}"
+        y = ext.conv2d_fwd(x, w, None, stride, pad)
+        _close(y, _conv_ref(x, w, None, stride, pad), name=f"{tag} fwd")
+        dy = torch.randn(N, K, P, P, device=DEV).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        wt2 = w.permute(1, 2, 3, 0).reshape(C, R * R * K).contiguous()
+        dx = ext.conv2d_dgrad(dy, wt2, N, C, H, H, R, R, stride, pad)
+        ref = torch.nn.grad.conv2d_input((N, C, H, H), w.float(), dy.float(),
+                                         stride=stride, padding=pad)
+        _close(dx, ref, name=f"{tag} dgrad")
+        dw = ext.conv2d_wgrad(dy, x, R, R, stride, pad)
+        refw = torch.nn.grad.conv2d_weight(x.float(), (K, C, R, R),
+                                           dy.float(), stride=stride,
+                                           padding=pad)
+        _close(dw, refw, rel=2e-2, atol=1.0, name=f"{tag} wgrad")
