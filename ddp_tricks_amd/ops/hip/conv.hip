@@ -977,8 +977,11 @@ __global__ void k_wgrad_c1_r3(const bf16* __restrict__ dy,
                                        q * cs.stride + s - cs.pad);
         };
         fill();
+        bf16 gnext = dy[gm * cs.Ko + ko];
         for (;;) {
-            float g = bf2f(dy[gm * cs.Ko + ko]);
+            float g = bf2f(gnext);
+            if (gm + 1 < gm_end)               // prefetch next row's dy
+                gnext = dy[(gm + 1) * cs.Ko + ko];
             #pragma unroll
             for (int k = 0; k < 9; ++k) acc[k] = fmaf(g, xw[k], acc[k]);
             if (++gm >= gm_end) break;
@@ -1300,7 +1303,7 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     if (cs.C < 8) {
         int rsc_total = Kgemm;
         if (rsc_total <= 16 && cs.Ko <= 64) {
-            int S_ = (int)std::max<long>(1, std::min<long>(2048, M / 8));
+            int S_ = (int)std::max<long>(1, std::min<long>(8192, M / 32));
             auto slab = at::empty({rsc_total * cs.Ko, S_},
                                   x.options().dtype(at::kFloat));
             if (cs.C == 1 && cs.R == 3 && cs.S == 3)
