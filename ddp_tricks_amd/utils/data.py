@@ -244,7 +244,85 @@ class CudaPrefetcher:
 
         _preload()
         while next_batch is not None:
-            torch.cuda.current_stream(self.device).wait_stream(self.stream)
+            main = torch.cuda.current_stream(self.device)
+            main.wait_stream(self.stream)
             batch = next_batch
+            # tensors were allocated on the copy stream; mark their use on
+            # the main stream so the caching allocator doesn't recycle them
+            # under in-flight kernels (torch stream-semantics rule)
+            batch[0].record_stream(main)
+            batch[1].record_stream(main)
             _preload()
             yield batch
+
+
+class FastBatchLoader:
+    """Vectorized batch loader for in-memory tensor datasets (MNIST/CIFAR):
+    one ``index_select`` per batch instead of per-item ``__getitem__`` +
+    collation — the Python loader was the epoch wall-time bound once the
+    GPU step dropped to ~3.7 ms.  Preserves DataLoader semantics: iterates
+    (images, labels) batches in the order produced by ``sampler`` (or
+    sequentially when sampler is None), honors ``set_epoch`` through the
+    sampler, optional pinned-memory staging for async H2D.
+    """
+
+    SLOTS = 4   # pinned-staging rotation depth (per-slot event guards reuse)
+
+    def __init__(self, dataset, batch_size: int, sampler=None,
+                 pin_memory: bool = False, device=None):
+        self.dataset = dataset
+        self.batch_size = batch_size
+        self.sampler = sampler
+        self.device = device
+        self.gpu = (device is not None and device.type == "cuda"
+                    and torch.cuda.is_available())
+        self.pin = pin_memory and self.gpu
+
+    def __len__(self):
+        n = len(self.sampler) if self.sampler is not None else len(self.dataset)
+        return (n + self.batch_size - 1) // self.batch_size
+
+    def __iter__(self):
+        if self.sampler is not None:
+            idx = torch.as_tensor(list(self.sampler), dtype=torch.long)
+        else:
+            idx = torch.arange(len(self.dataset), dtype=torch.long)
+        images = self.dataset.images
+        labels = self.dataset.labels
+        B = self.batch_size
+        if not self.gpu:
+            for s0 in range(0, idx.numel(), B):
+                sel = idx[s0:s0 + B]
+                yield (torch.index_select(images, 0, sel),
+                       torch.index_select(labels, 0, sel))
+            return
+        # GPU path: own the pinned staging AND the H2D (one-ahead on a copy
+        # stream).  Slot reuse is guarded by a host event-sync: the host can
+        # run many steps ahead of the device, so "old enough" is not enough.
+        S = self.SLOTS
+        copy_stream = torch.cuda.Stream(device=self.device)
+        pin_img = [torch.empty((B,) + images.shape[1:],
+                               dtype=images.dtype).pin_memory()
+                   for _ in range(S)]
+        pin_lbl = [torch.empty((B,), dtype=labels.dtype).pin_memory()
+                   for _ in range(S)]
+        events = [None] * S
+        main = torch.cuda.current_stream(self.device)
+        for bi, s0 in enumerate(range(0, idx.numel(), B)):
+            k = bi % S
+            if events[k] is not None:
+                events[k].synchronize()   # prior H2D from this slot done
+            sel = idx[s0:s0 + B]
+            n = sel.numel()
+            torch.index_select(images, 0, sel, out=pin_img[k][:n])
+            torch.index_select(labels, 0, sel, out=pin_lbl[k][:n])
+            with torch.cuda.stream(copy_stream):
+                dev_i = pin_img[k][:n].to(self.device, non_blocking=True)
+                dev_l = pin_lbl[k][:n].to(self.device, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record(copy_stream)
+            events[k] = ev
+            main.wait_stream(copy_stream)
+            dev_i.record_stream(main)
+            dev_l.record_stream(main)
+            yield dev_i, dev_l

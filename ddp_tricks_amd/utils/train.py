@@ -28,7 +28,7 @@ from ..models import build_model
 from ..ops.optim import FusedSGD
 from ..parallel.ddp import DistributedDataParallel as DDP
 from .callbacks import EarlyStopping, same_seeds
-from .data import DATASETS, DistributedSampler, CudaPrefetcher
+from .data import DATASETS, DistributedSampler, CudaPrefetcher, FastBatchLoader
 from .engine import iterate_loader
 from .lookahead import Lookahead
 from .schedulers import ReduceLROnPlateau, WarmupLambdaLR
@@ -63,15 +63,30 @@ def train(args):
     train_set = Dataset(root=args.data_path, train=True, download=True)
     train_sampler = DistributedSampler(train_set)
     same_seeds(args.seed_num)
-    train_loader = DataLoader(train_set, batch_size=args.batch_size,
-                              shuffle=False, pin_memory=device.type == "cuda",
-                              sampler=train_sampler)
+    pin = device.type == "cuda"
+    dev_arg = device if pin else None
+    if hasattr(train_set, "images"):   # in-memory tensors: vectorized path
+        train_loader = FastBatchLoader(train_set, args.batch_size,
+                                       sampler=train_sampler, pin_memory=pin,
+                                       device=dev_arg)
+    else:
+        train_loader = DataLoader(train_set, batch_size=args.batch_size,
+                                  shuffle=False, pin_memory=pin,
+                                  sampler=train_sampler)
     valid_set = Dataset(root=args.data_path, train=False, download=True)
-    valid_loader = DataLoader(valid_set, batch_size=args.batch_size,
-                              shuffle=False, pin_memory=device.type == "cuda")
+    if hasattr(valid_set, "images"):
+        valid_loader = FastBatchLoader(valid_set, args.batch_size,
+                                       pin_memory=pin, device=dev_arg)
+    else:
+        valid_loader = DataLoader(valid_set, batch_size=args.batch_size,
+                                  shuffle=False, pin_memory=pin)
     if device.type == "cuda":
-        train_loader = CudaPrefetcher(train_loader, device)
-        valid_loader = CudaPrefetcher(valid_loader, device)
+        # FastBatchLoader already yields device tensors with its own
+        # one-ahead copy stream; only wrap generic DataLoaders
+        if not isinstance(train_loader, FastBatchLoader):
+            train_loader = CudaPrefetcher(train_loader, device)
+        if not isinstance(valid_loader, FastBatchLoader):
+            valid_loader = CudaPrefetcher(valid_loader, device)
 
     print(f"Now Training: {args.exp_name}")
 

@@ -82,3 +82,30 @@ def test_synthetic_templates_shared_across_splits():
             assert sum(same) / len(same) > 0.9, same
     finally:
         del os.environ["DDPX_SYNTH_SAMPLES"]
+
+
+def test_fast_batch_loader_matches_dataloader():
+    import os
+
+    import torch
+    from torch.utils.data import DataLoader
+
+    from ddp_tricks_amd.utils.data import (MNIST, DistributedSampler,
+                                           FastBatchLoader)
+    os.environ["DDPX_SYNTH_SAMPLES"] = "100"
+    try:
+        ds = MNIST(root="/nonexistent", train=True)
+        smp1 = DistributedSampler(ds, num_replicas=2, rank=0)
+        smp2 = DistributedSampler(ds, num_replicas=2, rank=0)
+        smp1.set_epoch(3)
+        smp2.set_epoch(3)
+        ref = DataLoader(ds, batch_size=16, shuffle=False, sampler=smp1)
+        fast = FastBatchLoader(ds, 16, sampler=smp2)
+        ref_b = list(ref)
+        fast_b = list(fast)
+        assert len(ref_b) == len(fast_b) == len(fast)
+        for (ri, rl), (fi, fl) in zip(ref_b, fast_b):
+            assert torch.equal(ri, fi)
+            assert torch.equal(rl, fl)
+    finally:
+        del os.environ["DDPX_SYNTH_SAMPLES"]
