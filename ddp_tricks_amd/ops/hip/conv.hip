@@ -278,11 +278,53 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                                  + cs.cls_b + (long)w1 * cs.stride) * Nout
                                 + col;
                     out[addr] = f2bf(acc[mi][ni][r] + badd);
-                } else {
-                    out[(long)row * Nout + col] = f2bf(acc[mi][ni][r] + badd);
                 }
             }
         }
+
+    if (MODE != 2) {
+        // Coalesced epilogue: the fragment layout's direct stores are
+        // scalar bf16 scattered across rows (32 B effective segments);
+        // round-trip the tile through LDS and write full 16 B chunks of
+        // consecutive columns instead.
+        __syncthreads();                    // tiles are dead; reuse as stage
+        // [CBM_T][TBN] image split across the two tile arrays (each holds
+        // CBM_T*CBK elements; the image may need both)
+        bf16* stageA = &lds_a[0][0][0];
+        bf16* stageB = &lds_b[0][0][0];
+        constexpr int HALF = CBM_T * CBK;
+        auto stage_at = [&](int i) -> bf16& {
+            return i < HALF ? stageA[i] : stageB[i - HALF];
+        };
+        #pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < NI; ++ni) {
+                int cl = wc * WN + ni * 16 + (lane & 15);
+                float badd2 = (bias && n0 + cl < Nout) ? bias[n0 + cl] : 0.f;
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int rl = wr * WM + mi * 16 + (lane >> 4) * 4 + r;
+                    stage_at(rl * TBN + cl) = f2bf(acc[mi][ni][r] + badd2);
+                }
+            }
+        __syncthreads();
+        // write phase: each 16-lane group streams one row's TBN columns
+        constexpr int ROWS_PER_PASS = 256 / (TBN / 8);
+        const int rsub = tid / (TBN / 8);
+        const int csub = (tid % (TBN / 8)) * 8;
+        for (int r0_ = 0; r0_ < CBM_T; r0_ += ROWS_PER_PASS) {
+            int rl = r0_ + rsub;
+            long row = (long)m0 + rl;
+            int col = n0 + csub;
+            if (rl < CBM_T && row < M && col < Nout) {
+                int i = rl * TBN + csub;    // 16 B chunk within one half
+                const bf16* src16 = i < HALF ? &stageA[i] : &stageB[i - HALF];
+                *reinterpret_cast<bf16x8_t*>(&out[row * Nout + col]) =
+                    *reinterpret_cast<const bf16x8_t*>(src16);
+            }
+        }
+    }
 
     if (MODE == 0 && stats) {
         // Per-channel partial sum/sumsq of this block's 128 output rows —
