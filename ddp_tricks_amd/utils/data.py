@@ -155,8 +155,14 @@ class SyntheticImageNet(Dataset):
     def __getitem__(self, idx: int):
         g = torch.Generator().manual_seed((idx << 1) | (1 if self.train else 0))
         label = int(torch.randint(0, 1000, (1,), generator=g))
-        img = torch.rand(3, 224, 224, generator=g)
-        return img, label
+        # class-dependent learnable pattern: a label-seeded 14x14 patch tiled
+        # to 224x224 (materializing 1000 full-res templates would be 600 MB),
+        # plus index noise — shared across train/valid by construction
+        gt = torch.Generator().manual_seed(100000 + label)
+        patch = torch.rand(3, 14, 14, generator=gt)
+        img = 0.6 * patch.repeat_interleave(16, 1).repeat_interleave(16, 2) \
+            + 0.4 * torch.rand(3, 224, 224, generator=g)
+        return img.clamp_(0, 1), label
 
 
 DATASETS = {"mnist": MNIST, "cifar10": CIFAR10,
