@@ -137,32 +137,34 @@ class CIFAR10(Dataset):
 
 class SyntheticImageNet(Dataset):
     """Synthetic ImageNet-shaped data ([3,224,224], 1000 classes) for the
-    ResNet-50 config (BASELINE.json config 5) — generated per-index so the
-    full 1.28M-image epoch never materializes in host memory."""
+    ResNet-50 config (BASELINE.json config 5).  Class identity comes from a
+    shared 14x14 patch bank tiled up to 224x224 (learnable; full-res
+    templates would be 600 MB); images are materialized VECTORIZED in
+    __init__ (per-item generation measured seconds-slow) so the in-memory
+    FastBatchLoader path applies.  Size defaults are modest — host RAM is
+    ~600 KB/image fp32."""
 
     def __init__(self, root: str = "./datasets/", train: bool = True,
                  download: bool = False, num_samples: Optional[int] = None):
-        n_default = 10240 if train else 2048
+        n_default = 4096 if train else 1024
         env_cap = os.environ.get("DDPX_SYNTH_SAMPLES")
         if num_samples is None and env_cap:
-            num_samples = int(env_cap)
-        self.n = num_samples or n_default
-        self.train = train
+            num_samples = min(int(env_cap), n_default)
+        n = num_samples or n_default
+        gt = torch.Generator().manual_seed(55555)      # shared patch bank
+        patches = torch.rand(1000, 3, 14, 14, generator=gt)
+        g = torch.Generator().manual_seed(606 if train else 607)
+        self.labels = torch.randint(0, 1000, (n,), generator=g)
+        base = patches[self.labels]
+        base = base.repeat_interleave(16, 2).repeat_interleave(16, 3)
+        noise = torch.rand(n, 3, 224, 224, generator=g)
+        self.images = (0.6 * base + 0.4 * noise).clamp_(0, 1)
 
     def __len__(self) -> int:
-        return self.n
+        return self.labels.shape[0]
 
     def __getitem__(self, idx: int):
-        g = torch.Generator().manual_seed((idx << 1) | (1 if self.train else 0))
-        label = int(torch.randint(0, 1000, (1,), generator=g))
-        # class-dependent learnable pattern: a label-seeded 14x14 patch tiled
-        # to 224x224 (materializing 1000 full-res templates would be 600 MB),
-        # plus index noise — shared across train/valid by construction
-        gt = torch.Generator().manual_seed(100000 + label)
-        patch = torch.rand(3, 14, 14, generator=gt)
-        img = 0.6 * patch.repeat_interleave(16, 1).repeat_interleave(16, 2) \
-            + 0.4 * torch.rand(3, 224, 224, generator=g)
-        return img.clamp_(0, 1), label
+        return self.images[idx], int(self.labels[idx])
 
 
 DATASETS = {"mnist": MNIST, "cifar10": CIFAR10,
