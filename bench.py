@@ -41,6 +41,8 @@ MODEL_CONFIGS = {
     # r18 2048: 79.7k vs 73.6k img/s at 1024; r50 512: 6.55k vs 6.01k at 256
     "resnet18": ({"num_classes": 10, "cifar_stem": True}, (3, 32, 32), 10,
                  2048, "CIFAR-10(synthetic)"),
+    "resnet34": ({"num_classes": 10, "cifar_stem": True}, (3, 32, 32), 10,
+                 2048, "CIFAR-10(synthetic)"),
     "resnet50": ({"num_classes": 1000}, (3, 224, 224), 1000, 512,
                  "ImageNet(synthetic)"),
 }
