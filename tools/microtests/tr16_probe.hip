@@ -17,15 +17,17 @@ __global__ void probe(unsigned short* out, int mode) {
     else if (mode == 1)  addr = &lds[(l & 15) * 4];               // T10: own 8B chunk
     else if (mode == 2)  addr = &lds[(l & 15) + (l >> 4) * 64];   // pattern base
     else                 addr = &lds[l * 4];                      // linear 8B
-    u32 r0, r1;
-    asm volatile("ds_read_b64_tr_b16 %0, %2\n\ts_waitcnt lgkmcnt(0)"
-                 : "=v"(r0), "=v"(r1)
-                 : "v"(addr)
+    typedef __attribute__((address_space(3))) unsigned short* lds_ptr;
+    lds_ptr lp = (lds_ptr)addr;
+    unsigned long long r;
+    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(r)
+                 : "v"(lp)
                  : "memory");
-    out[(long)mode * 256 + l * 4 + 0] = (unsigned short)(r0 & 0xffff);
-    out[(long)mode * 256 + l * 4 + 1] = (unsigned short)(r0 >> 16);
-    out[(long)mode * 256 + l * 4 + 2] = (unsigned short)(r1 & 0xffff);
-    out[(long)mode * 256 + l * 4 + 3] = (unsigned short)(r1 >> 16);
+    out[(long)mode * 256 + l * 4 + 0] = (unsigned short)(r & 0xffff);
+    out[(long)mode * 256 + l * 4 + 1] = (unsigned short)((r >> 16) & 0xffff);
+    out[(long)mode * 256 + l * 4 + 2] = (unsigned short)((r >> 32) & 0xffff);
+    out[(long)mode * 256 + l * 4 + 3] = (unsigned short)((r >> 48) & 0xffff);
 }
 
 int main() {
