@@ -860,6 +860,151 @@ void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
         }
 }
 
+// Hardware-transpose wide wgrad: operands stored in NATURAL [m][ko]/[m][rsc]
+// LDS images (16 B ds_write_b128, no software transpose), fragments read
+// with gfx950's ds_read_b64_tr_b16.  Measured semantics (tools/microtests/
+// tr16_probe.hip): within a 16-lane group, lane g's j-th element comes from
+// element (g&3) of the 8-byte chunk at source-lane (4j + (g>>2))'s address;
+// addressing lane g at img[M0 + (g>>2)][col + (g&3)*4] therefore delivers
+// img[M0 + j][col + g] — the exact MFMA fragment.  Row stride 144 elements
+// (128 + 16 pad) makes the 16 addresses bank-disjoint (stride 72 dwords
+// = 8 mod 64).
+constexpr int TRS = 144;           // padded ko/rsc row stride (elements)
+
+typedef __attribute__((address_space(3))) const unsigned short* lds_cptr;
+
+DEV_INLINE unsigned long long tr16_read(lds_cptr a) {
+    unsigned long long r;
+    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(a));
+    return r;
+}
+DEV_INLINE unsigned long long tr16_read_off1152(lds_cptr a) {
+    unsigned long long r;   // +4 m-rows (4 * TRS * 2 bytes = 1152)
+    asm volatile("ds_read_b64_tr_b16 %0, %1 offset:1152" : "=v"(r) : "v"(a));
+    return r;
+}
+
+__global__ __launch_bounds__(256)
+void k_conv_wgrad_wide_tr(const bf16* __restrict__ dy,
+                          const bf16* __restrict__ x,
+                          float* __restrict__ slab, ConvShape cs, long M,
+                          int Kgemm, int S) {
+    constexpr int DEPTH = 64;
+    __shared__ bf16 img_a[DEPTH][TRS];   // [m][ko]   (natural layout)
+    __shared__ bf16 img_b[DEPTH][TRS];   // [m][rsc]
+    const int ko0 = blockIdx.x * 128;
+    const int rc0 = blockIdx.y * 128;
+    const int split = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
+
+    f32x4 acc[4][4] = {};
+    const int mloc = tid & 63;
+    const int jq = (tid >> 6) * 8;       // 4 quarters: j = jq + 32h
+
+    const long m_begin = (long)split * DEPTH;
+    for (long mt = m_begin; mt < M; mt += (long)S * DEPTH) {
+        long gm = mt + mloc;
+        const bool valid = gm < M;
+        // ---- A image: dy rows, natural 16 B writes ----
+        #pragma unroll
+        for (int h = 0; h < 4; ++h) {
+            const int j = jq + h * 32;
+            bf16x8_t va = {};
+            if (valid && ko0 + j < cs.Ko)
+                va = *reinterpret_cast<const bf16x8_t*>(
+                    &dy[gm * cs.Ko + ko0 + j]);
+            *reinterpret_cast<bf16x8_t*>(&img_a[mloc][j]) = va;
+        }
+        // ---- B image: im2col(x) gather, natural 16 B writes ----
+        if (valid) {
+            unsigned rem = fd_div((unsigned)gm, cs.fdQ);
+            int q = fd_mod((unsigned)gm, cs.fdQ, rem);
+            unsigned n = fd_div(rem, cs.fdP);
+            int p = fd_mod(rem, cs.fdP, n);
+            #pragma unroll
+            for (int h = 0; h < 4; ++h) {
+                const int j = jq + h * 32;
+                bf16x8_t vb = {};
+                int gk = rc0 + j;
+                if (gk < Kgemm) {
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int hh = p * cs.stride + r - cs.pad;
+                    int wcol = q * cs.stride + sx - cs.pad;
+                    if (hh >= 0 && hh < cs.H && wcol >= 0 && wcol < cs.W)
+                        vb = *reinterpret_cast<const bf16x8_t*>(
+                            &x[(((long)n * cs.H + hh) * cs.W + wcol) * cs.C + c]);
+                }
+                *reinterpret_cast<bf16x8_t*>(&img_b[mloc][j]) = vb;
+            }
+        } else {
+            #pragma unroll
+            for (int h = 0; h < 4; ++h)
+                *reinterpret_cast<bf16x8_t*>(&img_b[mloc][jq + h * 32]) =
+                    bf16x8_t{};
+        }
+        __syncthreads();
+
+        const int g = lane & 15;
+        const int msub = (lane >> 4) * 8 + (g >> 2);   // lane's address row
+        const int csub = (g & 3) * 4;                  // lane's address col
+        #pragma unroll
+        for (int ks = 0; ks < DEPTH; ks += 32) {
+            // issue all 16 transpose reads, wait once, then reinterpret —
+            // asm results are NOT valid until the explicit lgkmcnt wait
+            unsigned long long ra[4][2], rb[4][2];
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi) {
+                lds_cptr a = (lds_cptr)&img_a[ks + msub]
+                                            [wr * 64 + mi * 16 + csub];
+                ra[mi][0] = tr16_read(a);
+                ra[mi][1] = tr16_read_off1152(a);
+            }
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                lds_cptr b = (lds_cptr)&img_b[ks + msub]
+                                            [wc * 64 + ni * 16 + csub];
+                rb[ni][0] = tr16_read(b);
+                rb[ni][1] = tr16_read_off1152(b);
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            bf16x8_t af[4], bfr[4];
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                __builtin_memcpy(&af[mi], &ra[mi][0], 16);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                __builtin_memcpy(&bfr[ni], &rb[ni][0], 16);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            int col = rc0 + wc * 64 + ni * 16 + (lane & 15);
+            if (col >= Kgemm) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = ko0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= cs.Ko) continue;
+                slab[((long)split * cs.Ko + row) * Kgemm + col] =
+                    acc[mi][ni][r];
+            }
+        }
+}
+
 // Pair-m wide wgrad: each thread stages TWO consecutive m's per j-group so
 // LDS writes are packed b32 (16 stores/operand/iter vs 32 conflicted b16) —
 // the stage phase of the wide kernel is store-issue bound.
@@ -1415,7 +1560,11 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     // pair-store wide is the measured default (conv4 373 vs 461 us,
     // conv3 165 vs 210); DDPX_WGRAD_V=w selects the scalar-store wide
     const bool wide_pair = !wv || wv[1] == 'p' || wv[1] == 0;
-    if (use_wide && wide64 && wide_pair)
+    if (use_wide && wide64 && wv && wv[1] == 't')
+        hipLaunchKernelGGL(k_conv_wgrad_wide_tr, dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_wide && wide64 && wide_pair)
         hipLaunchKernelGGL(k_conv_wgrad_wide_pair, dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
