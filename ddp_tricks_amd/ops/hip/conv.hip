@@ -971,20 +971,44 @@ void k_conv_wgrad_wide_tr(const bf16* __restrict__ dy,
                 rb[ni][0] = tr16_read(b);
                 rb[ni][1] = tr16_read_off1152(b);
             }
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            // counted waits: start the ni-th MFMA column as soon as its
+            // B fragment lands; the waited values are threaded THROUGH the
+            // asm ("+v") so the compiler cannot reorder their uses above it
             bf16x8_t af[4], bfr[4];
+            asm volatile("s_waitcnt lgkmcnt(6)"
+                         : "+v"(ra[0][0]), "+v"(ra[0][1]), "+v"(ra[1][0]),
+                           "+v"(ra[1][1]), "+v"(ra[2][0]), "+v"(ra[2][1]),
+                           "+v"(ra[3][0]), "+v"(ra[3][1]), "+v"(rb[0][0]),
+                           "+v"(rb[0][1]));
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 __builtin_memcpy(&af[mi], &ra[mi][0], 16);
-            #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
-                __builtin_memcpy(&bfr[ni], &rb[ni][0], 16);
+            __builtin_memcpy(&bfr[0], &rb[0][0], 16);
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
-                #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
-                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+                acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[0], acc[mi][0], 0, 0, 0);
+            asm volatile("s_waitcnt lgkmcnt(4)"
+                         : "+v"(rb[1][0]), "+v"(rb[1][1]));
+            __builtin_memcpy(&bfr[1], &rb[1][0], 16);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[1], acc[mi][1], 0, 0, 0);
+            asm volatile("s_waitcnt lgkmcnt(2)"
+                         : "+v"(rb[2][0]), "+v"(rb[2][1]));
+            __builtin_memcpy(&bfr[2], &rb[2][0], 16);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                acc[mi][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[2], acc[mi][2], 0, 0, 0);
+            asm volatile("s_waitcnt lgkmcnt(0)"
+                         : "+v"(rb[3][0]), "+v"(rb[3][1]));
+            __builtin_memcpy(&bfr[3], &rb[3][0], 16);
+            #pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                acc[mi][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi], bfr[3], acc[mi][3], 0, 0, 0);
         }
         __syncthreads();
     }
