@@ -578,3 +578,36 @@ def test_conv_shape_fuzz():
                                            dy.float(), stride=stride,
                                            padding=pad)
         _close(dw, refw, rel=2e-2, atol=1.0, name=f"{tag} wgrad")
+
+
+def test_bn_shape_fuzz():
+    """BN train fwd+bwd across randomized (N,C,H,W) incl. odd spatial and
+    the C/8==256 ceiling — guards shapes outside the named tests."""
+    import random
+    rng = random.Random(77)
+    for trial in range(8):
+        C = rng.choice([8, 24, 64, 256, 1024, 2048])
+        N = rng.choice([2, 5, 16])
+        H = rng.choice([1, 3, 7, 13, 28])
+        x = torch.randn(N, C, H, H, device=DEV).to(torch.bfloat16)\
+            .contiguous(memory_format=CL)
+        g = torch.rand(C, device=DEV) + 0.5
+        b = torch.randn(C, device=DEV)
+        rm = torch.zeros(C, device=DEV)
+        rv = torch.ones(C, device=DEV)
+        y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True)
+        xf = x.float().requires_grad_(True)
+        g2 = g.clone().requires_grad_(True)
+        b2 = b.clone().requires_grad_(True)
+        ref = torch.nn.functional.batch_norm(
+            xf, torch.zeros(C, device=DEV), torch.ones(C, device=DEV),
+            g2, b2, True, 0.1, 1e-5).relu()
+        tag = f"bnfuzz{trial} N{N} C{C} H{H}"
+        _close(y, ref, name=f"{tag} fwd")
+        dy = torch.randn_like(ref).to(torch.bfloat16)
+        ref.backward(dy.float())
+        dx, dg, db = ext.bn_bwd(x, dy.contiguous(memory_format=CL), g, sm,
+                                si, mask, True)
+        _close(dg, g2.grad, rel=2e-2, atol=0.2, name=f"{tag} dgamma")
+        _close(db, b2.grad, rel=2e-2, atol=0.2, name=f"{tag} dbeta")
+        _close(dx, xf.grad, rel=5e-2, atol=5e-2, name=f"{tag} dx")
