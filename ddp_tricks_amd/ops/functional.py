@@ -102,8 +102,10 @@ def _nhwc_2d(x: torch.Tensor) -> torch.Tensor:
 
 def _pad8(xb: torch.Tensor) -> torch.Tensor:
     """Zero-pad channels to 8 (channels_last) so C<8 inputs can take the
-    MFMA implicit-GEMM path — worth it at ResNet-stem scale where the
-    direct VALU kernel is compute-bound."""
+    MFMA implicit-GEMM path — one fused kernel on GPU."""
+    ext = require_ext_for(xb)
+    if ext is not None:
+        return ext.pad8_channels(xb)
     N, C, H, W = xb.shape
     xp = torch.zeros(N, 8, H, W, dtype=xb.dtype, device=xb.device)
     xp = xp.contiguous(memory_format=CL)

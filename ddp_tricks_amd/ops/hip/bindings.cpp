@@ -10,6 +10,7 @@ void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 void fused_lookahead(std::vector<at::Tensor> fast, std::vector<at::Tensor> slow,
                      double alpha, c10::optional<at::Tensor> found_inf);
 at::Tensor cast_to_bf16(at::Tensor x);
+at::Tensor pad8_channels(at::Tensor x);
 std::vector<at::Tensor> pack_conv_weight(at::Tensor w, bool pad8,
                                          bool want_wt2);
 
@@ -79,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("fast"), py::arg("slow"), py::arg("alpha"),
           py::arg("found_inf") = c10::nullopt);
     m.def("cast_to_bf16", &cast_to_bf16);
+    m.def("pad8_channels", &pad8_channels);
     m.def("pack_conv_weight", &pack_conv_weight, py::arg("w"),
           py::arg("pad8") = false, py::arg("want_wt2") = true);
     m.def("ce_fwd", &ce_fwd);

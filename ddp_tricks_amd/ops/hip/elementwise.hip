@@ -316,3 +316,34 @@ std::vector<at::Tensor> pack_conv_weight(at::Tensor w, bool pad8,
     if (want_wt2) return {nhwc, wt2};
     return {nhwc};
 }
+
+// Zero-pad channels to 8 in ONE pass (replaces at::zeros + slice copy_ —
+// two ATen launches per step on the C<8 stem path).
+__global__ void k_pad8(const bf16* __restrict__ x, bf16* __restrict__ xp,
+                       long total_px, int C) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total_px; i += stride) {
+        s16x8 o = {};
+        const bf16* src = x + i * C;
+        for (int c = 0; c < C; ++c) o[c] = *(const short*)&src[c];
+        reinterpret_cast<s16x8*>(xp)[i] = o;
+    }
+}
+
+at::Tensor pad8_channels(at::Tensor x) {
+    // x: [N,C<8,H,W] channels_last bf16 -> [N,8,H,W] channels_last
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+    TORCH_CHECK(C < 8);
+    auto xp = at::empty({N, 8, H, W},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    long total = (long)N * H * W;
+    auto stream = at::hip::getCurrentHIPStream();
+    int blocks = std::min<long>(4096, ceil_div_i(total, 256));
+    hipLaunchKernelGGL(k_pad8, dim3(blocks), dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<bf16*>(xp.data_ptr()), total, C);
+    HIP_CHECK_LAST();
+    return xp;
+}
