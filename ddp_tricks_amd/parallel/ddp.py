@@ -218,8 +218,13 @@ class DistributedDataParallel(nn.Module):
         torch._foreach_zero_([b.flat for b in self._buckets])
 
     def forward(self, *args, **kwargs):
+        # NOTE: broadcast in EVAL forwards too (torch-DDP semantics): ranks'
+        # BN running stats drift apart during training (rank-local updates
+        # land after each pre-forward sync), and rank-identical validation
+        # metrics are what keep the un-collectivized EarlyStopping/plateau
+        # decisions in lockstep.
         if (self.broadcast_buffers and self.world_size > 1
-                and self.module.training and self._buf_flats):
+                and self._buf_flats):
             if self._rccl is not None:
                 self._ext.rccl_group_start()
                 for flat in self._buf_flats:

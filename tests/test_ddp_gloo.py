@@ -140,3 +140,37 @@ def test_singleproc_ddp_grad_views():
         assert any(p.grad.data_ptr() >= f.data_ptr()
                    and p.grad.data_ptr() < f.data_ptr() + f.numel() * 4
                    for f in flats)
+
+
+def _worker_eval_buffer_sync(rank, port, results):
+    _init(rank, WORLD, port)
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    torch.manual_seed(11)
+    model = TinyNet()
+    ddp = DDP(model)
+    # diverge the BN running stats rank-locally (post-broadcast updates)
+    with torch.no_grad():
+        model.bn.running_mean.add_(float(rank + 1))
+    ddp.eval()
+    with torch.no_grad():
+        ddp(torch.zeros(4, 8))   # eval forward must re-sync buffers
+    if rank == 1:
+        results.put(model.bn.running_mean.numpy().copy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_eval_forward_resyncs_buffers():
+    """Eval forwards broadcast buffers (rank-identical valid metrics keep
+    EarlyStopping in lockstep — SURVEY Appendix A.7)."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_eval_buffer_sync,
+                         args=(r, 29613, results)) for r in range(WORLD)]
+    [p.start() for p in procs]
+    rm1 = results.get(timeout=110)
+    [p.join(timeout=60) for p in procs]
+    assert all(p.exitcode == 0 for p in procs)
+    # rank 1 must hold rank 0's buffers (running_mean += 1.0, not += 2.0)
+    assert abs(float(rm1.mean()) - 1.0) < 1e-5, rm1
