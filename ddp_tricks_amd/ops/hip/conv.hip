@@ -752,14 +752,18 @@ void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
 }
 
 // Pair-m single-buffer 64x64 wgrad: thread halves split the two operands;
-// each thread stages its m-PAIR with packed b32 stores (conflict-free).
+// each thread stages m-PAIRs with packed b32 stores (conflict-free).
+// DEPTH = staged contraction (32 or 64; 64 halves the barrier rate).
+template <int DEPTH>
 __global__ __launch_bounds__(256)
 void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
                           const bf16* __restrict__ x,
                           float* __restrict__ slab, ConvShape cs, long M,
                           int Kgemm, int S) {
-    __shared__ bf16 lds_a[WBM][SLDK];   // [ko][m]
-    __shared__ bf16 lds_b[WBN][SLDK];   // [rsc][m]
+    __shared__ bf16 lds_a[WBM][DEPTH + 8];   // [ko][m]
+    __shared__ bf16 lds_b[WBN][DEPTH + 8];   // [rsc][m]
+    constexpr int SLD = DEPTH + 8;
+    constexpr int PAIRS = DEPTH / 2;         // m-pairs per row
     const int ko0 = blockIdx.x * WBM;
     const int rc0 = blockIdx.y * WBN;
     const int split = blockIdx.z;
@@ -768,79 +772,95 @@ void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
     const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
 
     f32x4 acc[2][2] = {};
-    const int m2 = (tid & 15) * 2;          // m, m+1 (covers 32 m)
-    const int jsel = tid >> 4;              // 0..15
-    const int j8 = (jsel & 7) * 8;          // 8 j-groups x 8 = 64 j
-    const bool is_b = jsel >= 8;            // thread half stages A or B
+    const int m2 = (tid & (PAIRS - 1)) * 2;
+    const int jsel = tid / PAIRS;            // 256/PAIRS j-slots
+    constexpr int JSLOTS = 256 / PAIRS;      // 16 (D32) or 8 (D64)
+    const bool is_b = jsel >= JSLOTS / 2;    // thread half stages A or B
+    const int jb = (jsel % (JSLOTS / 2)) * 8;
+    constexpr int JSTEP = (JSLOTS / 2) * 8;  // j covered per pass
 
-    const long m_begin = (long)split * SBK;
-    for (long mt = m_begin; mt < M; mt += (long)S * SBK) {
+    const long m_begin = (long)split * DEPTH;
+    for (long mt = m_begin; mt < M; mt += (long)S * DEPTH) {
         long gm0 = mt + m2;
         const bool v0 = gm0 < M, v1 = gm0 + 1 < M;
-        bf16x8_t t0 = {}, t1 = {};
-        if (!is_b) {
-            if (ko0 + j8 < cs.Ko) {
-                if (v0) t0 = *reinterpret_cast<const bf16x8_t*>(
-                    &dy[gm0 * cs.Ko + ko0 + j8]);
-                if (v1) t1 = *reinterpret_cast<const bf16x8_t*>(
-                    &dy[(gm0 + 1) * cs.Ko + ko0 + j8]);
+        // decode once per m-pair (B threads only need it)
+        unsigned q0 = 0, p0 = 0, n0_ = 0, q1 = 0, p1 = 0, n1_ = 0;
+        if (is_b) {
+            if (v0) {
+                unsigned rem = fd_div((unsigned)gm0, cs.fdQ);
+                q0 = fd_mod((unsigned)gm0, cs.fdQ, rem);
+                n0_ = fd_div(rem, cs.fdP);
+                p0 = fd_mod(rem, cs.fdP, n0_);
             }
-        } else {
-            int gk = rc0 + j8;
-            if (gk < Kgemm) {
-                unsigned rs = fd_div(gk, cs.fdC);
-                int c = fd_mod(gk, cs.fdC, rs);
-                int r = fd_div(rs, cs.fdS);
-                int sx = fd_mod(rs, cs.fdS, r);
-                if (v0) {
-                    unsigned rem = fd_div((unsigned)gm0, cs.fdQ);
-                    int q = fd_mod((unsigned)gm0, cs.fdQ, rem);
-                    unsigned n = fd_div(rem, cs.fdP);
-                    int pp = fd_mod(rem, cs.fdP, n);
-                    int h = pp * cs.stride + r - cs.pad;
-                    int wcol = q * cs.stride + sx - cs.pad;
-                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        t0 = *reinterpret_cast<const bf16x8_t*>(
-                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
-                }
-                if (v1) {
-                    unsigned rem = fd_div((unsigned)gm0 + 1, cs.fdQ);
-                    int q = fd_mod((unsigned)gm0 + 1, cs.fdQ, rem);
-                    unsigned n = fd_div(rem, cs.fdP);
-                    int pp = fd_mod(rem, cs.fdP, n);
-                    int h = pp * cs.stride + r - cs.pad;
-                    int wcol = q * cs.stride + sx - cs.pad;
-                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        t1 = *reinterpret_cast<const bf16x8_t*>(
-                            &x[(((long)n * cs.H + h) * cs.W + wcol) * cs.C + c]);
-                }
+            if (v1) {
+                unsigned rem = fd_div((unsigned)gm0 + 1, cs.fdQ);
+                q1 = fd_mod((unsigned)gm0 + 1, cs.fdQ, rem);
+                n1_ = fd_div(rem, cs.fdP);
+                p1 = fd_mod(rem, cs.fdP, n1_);
             }
         }
-        bf16* dst = is_b ? &lds_b[0][0] : &lds_a[0][0];
         #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {
-            unsigned pk = (unsigned)(unsigned short)t0[jj]
-                | ((unsigned)(unsigned short)t1[jj] << 16);
-            *reinterpret_cast<unsigned*>(&dst[(j8 + jj) * SLDK + m2]) = pk;
+        for (int j8 = jb; j8 < 64; j8 += JSTEP) {
+            bf16x8_t t0 = {}, t1 = {};
+            if (!is_b) {
+                if (ko0 + j8 < cs.Ko) {
+                    if (v0) t0 = *reinterpret_cast<const bf16x8_t*>(
+                        &dy[gm0 * cs.Ko + ko0 + j8]);
+                    if (v1) t1 = *reinterpret_cast<const bf16x8_t*>(
+                        &dy[(gm0 + 1) * cs.Ko + ko0 + j8]);
+                }
+            } else {
+                int gk = rc0 + j8;
+                if (gk < Kgemm) {
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    if (v0) {
+                        int h = (int)p0 * cs.stride + r - cs.pad;
+                        int wcol = (int)q0 * cs.stride + sx - cs.pad;
+                        if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                            t0 = *reinterpret_cast<const bf16x8_t*>(
+                                &x[(((long)n0_ * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                    }
+                    if (v1) {
+                        int h = (int)p1 * cs.stride + r - cs.pad;
+                        int wcol = (int)q1 * cs.stride + sx - cs.pad;
+                        if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                            t1 = *reinterpret_cast<const bf16x8_t*>(
+                                &x[(((long)n1_ * cs.H + h) * cs.W + wcol) * cs.C + c]);
+                    }
+                }
+            }
+            bf16* dst = is_b ? &lds_b[0][0] : &lds_a[0][0];
+            #pragma unroll
+            for (int jj = 0; jj < 8; ++jj) {
+                unsigned pk = (unsigned)(unsigned short)t0[jj]
+                    | ((unsigned)(unsigned short)t1[jj] << 16);
+                *reinterpret_cast<unsigned*>(&dst[(j8 + jj) * SLD + m2]) = pk;
+            }
         }
         __syncthreads();
 
-        bf16x8_t af[2], bfr[2];
-        const int kcol = (lane >> 4) * 8;
         #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
-            af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
-        #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
-            bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
-        #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+        for (int ks = 0; ks < DEPTH; ks += 32) {
+            bf16x8_t af[2], bfr[2];
+            const int kcol = ks + (lane >> 4) * 8;
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
             #pragma unroll
             for (int ni = 0; ni < 2; ++ni)
-                acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+                bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
+                    &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 2; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        }
         __syncthreads();
     }
 
@@ -1600,9 +1620,13 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
         hipLaunchKernelGGL((k_conv_wgrad_wide<32>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_sb && wv && wv[1] == '6')
+        hipLaunchKernelGGL((k_conv_wgrad_sb_pair<64>), dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_sb && !(wv && wv[1] == 'x'))
-        hipLaunchKernelGGL(k_conv_wgrad_sb_pair, dim3(gk, gr, S_), dim3(256),
-                           0, stream.stream(), dyp, xp,
+        hipLaunchKernelGGL((k_conv_wgrad_sb_pair<32>), dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_sb)
         hipLaunchKernelGGL(k_conv_wgrad_sb, dim3(gk, gr, S_), dim3(256), 0,
