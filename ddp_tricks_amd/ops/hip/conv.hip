@@ -18,8 +18,7 @@
 
 typedef short bf16x8_t __attribute__((ext_vector_type(8)));
 
-constexpr int CBM = 128, CBN = 128, CBK = 64;
-constexpr int CLDK = CBK + 8;
+constexpr int CBM = 128, CBK = 64;
 
 struct ConvShape {
     int N, H, W, C;     // input
