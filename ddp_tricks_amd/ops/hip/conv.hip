@@ -1392,7 +1392,8 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
                            stream.stream(), xp, wp, bp, yp, cs, M);
     } else {
         TORCH_CHECK(cs.C % 8 == 0, "conv fwd needs C % 8 == 0");
-        if (cs.Ko >= 128) {
+        static const char* tbn_env = getenv("DDPX_CONV_TBN64");
+        if (cs.Ko >= 128 && !tbn_env) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 128));
             hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid, dim3(256), 0,
                                stream.stream(), xp, wp, bp, yp, cs, (int)M,
