@@ -1051,12 +1051,12 @@ void k_conv_wgrad_wide_tr(const bf16* __restrict__ dy,
 // Pair-m wide wgrad: each thread stages TWO consecutive m's per j-group so
 // LDS writes are packed b32 (16 stores/operand/iter vs 32 conflicted b16) —
 // the stage phase of the wide kernel is store-issue bound.
+template <int DEPTH = 64>
 __global__ __launch_bounds__(256)
 void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
                             const bf16* __restrict__ x,
                             float* __restrict__ slab, ConvShape cs, long M,
                             int Kgemm, int S) {
-    constexpr int DEPTH = 64;
     __shared__ bf16 lds_a[128][DEPTH + 8];
     __shared__ bf16 lds_b[128][DEPTH + 8];
     const int ko0 = blockIdx.x * 128;
@@ -1067,8 +1067,9 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
     const int wid = tid >> 6, wr = wid >> 1, wc = wid & 1;
 
     f32x4 acc[4][4] = {};
-    const int m2 = (tid & 31) * 2;          // m, m+1
-    const int jb = (tid >> 5) * 8;          // 8 groups x 8 = 64 j per pass
+    const int m2 = (tid & (DEPTH / 2 - 1)) * 2;   // m, m+1
+    const int jb = (tid / (DEPTH / 2)) * 8;
+    constexpr int JSTEP2 = (256 / (DEPTH / 2)) * 8;  // j per pass
     const long m_begin = (long)split * DEPTH;
 
     for (long mt = m_begin; mt < M; mt += (long)S * DEPTH) {
@@ -1089,8 +1090,7 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
             p1 = fd_mod(rem, cs.fdP, n1_);
         }
         #pragma unroll
-        for (int half = 0; half < 2; ++half) {
-            const int j = jb + half * 64;
+        for (int j = jb; j < 128; j += JSTEP2) {
             bf16x8_t a0 = {}, a1 = {};
             if (ko0 + j < cs.Ko) {
                 if (v0) a0 = *reinterpret_cast<const bf16x8_t*>(
@@ -1608,8 +1608,12 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
         hipLaunchKernelGGL(k_conv_wgrad_wide_tr, dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
+    else if (use_wide && wv && wv[1] == 'q')     // wq: pair at depth 32
+        hipLaunchKernelGGL((k_conv_wgrad_wide_pair<32>), dim3(gk, gr, S_),
+                           dim3(256), 0, stream.stream(), dyp, xp,
+                           slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_wide && wide64 && wide_pair)
-        hipLaunchKernelGGL(k_conv_wgrad_wide_pair, dim3(gk, gr, S_),
+        hipLaunchKernelGGL((k_conv_wgrad_wide_pair<64>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
     else if (use_wide && wide64)
