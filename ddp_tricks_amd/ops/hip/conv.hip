@@ -1573,20 +1573,24 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     }
 
     TORCH_CHECK(cs.C % 8 == 0 && cs.Ko % 8 == 0);
-    // Variant choice is MEASURED (profiles/, tools/bench_kernels.py): the
-    // 128x128 wide tile wins when it fits (4x MFMA work per barrier), the
-    // single-buffer 32-deep 64x64 tile beats the double-buffered pipelined
-    // one everywhere else.  DDPX_WGRAD_V = wide|sb|pipe overrides.
+    // Variant choice is MEASURED (profiles/, tools/bench_kernels.py):
+    //   * 128x128 pair-store wide tile, 64-deep, for big-Kgemm convs with
+    //     enough tiles (conv3 165, conv4 373 us);
+    //   * the same tile 32-DEEP for small-Kgemm or tile-starved-but-deep-M
+    //     shapes (conv2 308 vs 345 sb; [1024ch 1x1] 76 vs 95 us) — the
+    //     shallower stage runs at occupancy 4 vs 3;
+    //   * 64x64 pair-store sb otherwise.  DDPX_WGRAD_V overrides
+    //     (w/wq/w3/wx/wt/s/s6/sx/p).
     static const char* wv = getenv("DDPX_WGRAD_V");
     const char sel = wv ? wv[0] : 0;
     const bool can_wide = cs.Ko >= 128 && Kgemm >= 128;
-    // wide also needs enough 128x128 tiles to fill the machine (measured:
-    // conv2 with 1x5 tiles is slower wide than sb, conv3/conv4 faster)
     const bool enough = can_wide &&
-        ceil_div_i(cs.Ko, 128) * ceil_div_i(Kgemm, 128) >= 8;
+        (ceil_div_i(cs.Ko, 128) * ceil_div_i(Kgemm, 128) >= 8
+         || M >= 262144);
     const bool use_wide = (sel ? sel == 'w' : enough) && can_wide;
     const bool use_sb = sel ? (sel == 's' || (sel == 'w' && !can_wide))
                             : !enough;
+    const bool depth64_auto = Kgemm >= 1152;
     const bool wide64 = use_wide && !(wv && wv[1] == '3');  // w32 forces 32
     const int depth = use_wide ? (wide64 ? 64 : SBK) : (use_sb ? SBK : WBK);
     const int tm = use_wide ? 128 : WBM, tn = use_wide ? 128 : WBN;
@@ -1608,7 +1612,7 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
         hipLaunchKernelGGL(k_conv_wgrad_wide_tr, dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
-    else if (use_wide && wv && wv[1] == 'q')     // wq: pair at depth 32
+    else if (use_wide && (wv ? wv[1] == 'q' : !depth64_auto))
         hipLaunchKernelGGL((k_conv_wgrad_wide_pair<32>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
