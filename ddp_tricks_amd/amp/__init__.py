@@ -61,6 +61,7 @@ class _AmpState:
         self.found_inf = None        # per-step flag, read by the fused
         #                              optimizer kernels (device-side skip)
         self.overflow_count = None   # running count, host-synced per epoch
+        self.async_steps = 0         # backward count since last scaler sync
         self.async_mode = False
 
 
@@ -88,10 +89,14 @@ def maybe_sync_scaler() -> None:
     if not _state.async_mode or _state.overflow_count is None:
         return
     n = int(_state.overflow_count.item())
+    good = max(0, _state.async_steps - n)
+    for _ in range(n):
+        _state.scaler.update(found_inf=True)
+    for _ in range(good):          # keep the x2-after-growth_interval policy
+        _state.scaler.update(found_inf=False)
     if n > 0:
-        for _ in range(n):
-            _state.scaler.update(found_inf=True)
         _state.overflow_count.zero_()
+    _state.async_steps = 0
 
 
 def state_dict() -> dict:
@@ -213,6 +218,7 @@ def scale_loss(loss, optimizer, delay_unscale: bool = False):
         ext = load_extension(required=False)
         if ext is not None:
             _state.async_mode = True
+            _state.async_steps += 1
             fi = _device_buffers(tensors[0].device)
             fi.zero_()  # stream-ordered: prior step's kernels already read it
             ext.multi_tensor_unscale(tensors, fi, inv)
