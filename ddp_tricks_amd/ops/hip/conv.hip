@@ -1601,7 +1601,7 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     // TFLOP/s at the old S<=64 cap on [Ko=64,C=64] wgrad
     // deepen splits until ~1024 blocks, but keep >=8 contraction steps per
     // block (measured: S=512 at M=65k quadrupled a 55 us wgrad)
-    while (gk * gr * S_ < 1024 && S_ < 512 &&
+    while (gk * gr * S_ < 1024 && S_ < 2048 &&
            M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
