@@ -1601,7 +1601,9 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
     // TFLOP/s at the old S<=64 cap on [Ko=64,C=64] wgrad
     // deepen splits until ~1024 blocks, but keep >=8 contraction steps per
     // block (measured: S=512 at M=65k quadrupled a 55 us wgrad)
-    while (gk * gr * S_ < 1024 && S_ < 2048 &&
+    // (cap 2048 measured worse: the S-sweep in the combine pass dominates
+    // — 281 vs 174 us at [Ko=64,C=64,M=800k])
+    while (gk * gr * S_ < 1024 && S_ < 512 &&
            M / ((long)S_ * 2 * depth) >= 8) S_ *= 2;
     auto slab = at::empty({S_, (long)cs.Ko, (long)Kgemm},
                           x.options().dtype(at::kFloat));
