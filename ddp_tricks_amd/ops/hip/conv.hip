@@ -1630,7 +1630,9 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
         hipLaunchKernelGGL((k_conv_wgrad_wide<32>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
-    else if (use_sb && wv && wv[1] == '6')
+    else if (use_sb && (wv ? wv[1] == '6' : (gk == 1 && gr <= 8)))
+        // single-ko-tile shapes: 64-deep staging measured faster (1x1
+        // [64x64,M=800k] 166 vs 174 us; 7x7 stem 209 vs 215)
         hipLaunchKernelGGL((k_conv_wgrad_sb_pair<64>), dim3(gk, gr, S_),
                            dim3(256), 0, stream.stream(), dyp, xp,
                            slab.data_ptr<float>(), cs, M, Kgemm, S_);
