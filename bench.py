@@ -37,13 +37,14 @@ BASELINE_IMG_PER_SEC = 12000.0  # BASELINE.md implied train throughput
 # model -> (ctor kwargs, input CHW, n classes, per-GPU batch, dataset label)
 MODEL_CONFIGS = {
     "toy_net": ({}, (1, 28, 28), 10, 1024, "MNIST(synthetic)"),
-    # per-GPU batches picked by measurement (288 GB HBM3E leaves room):
-    # r18 2048: 79.7k vs 73.6k img/s at 1024; r50 512: 6.55k vs 6.01k at 256
+    # per-GPU batches picked by measurement (288 GB HBM3E leaves room;
+    # sweep: r18 97.5k@4096 vs 93k@2048; r50 8.75k@1024 vs 8.2k@512; the
+    # next doubling gains <2% while doubling pinned-host staging)
     "resnet18": ({"num_classes": 10, "cifar_stem": True}, (3, 32, 32), 10,
-                 2048, "CIFAR-10(synthetic)"),
+                 4096, "CIFAR-10(synthetic)"),
     "resnet34": ({"num_classes": 10, "cifar_stem": True}, (3, 32, 32), 10,
                  2048, "CIFAR-10(synthetic)"),
-    "resnet50": ({"num_classes": 1000}, (3, 224, 224), 1000, 512,
+    "resnet50": ({"num_classes": 1000}, (3, 224, 224), 1000, 1024,
                  "ImageNet(synthetic)"),
 }
 
