@@ -12,6 +12,8 @@ ops in fp32, doubling as the numerics oracle the GPU tests compare against.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -147,13 +149,23 @@ def _conv_bwd_impl(ctx, dy):
     dx = dw = db = None
     if ctx.needs_input_grad[0]:
         wt2 = weight_variant(weight, "wt2_p8" if pad8 else "wt2")
-        dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], xb.shape[1],
-                              xb.shape[2], xb.shape[3], R, S,
-                              stride, padding)
-        if pad8:
-            dx = dx[:, :C]
-        if x_dtype == torch.float32:
-            dx = dx.float()
+        box = getattr(ctx, "bn_box", None)
+        if (box is not None and not pad8 and stride == 1
+                and x_dtype != torch.float32):
+            dx, slab = ext.conv2d_dgrad_bn(
+                dyb, wt2, xb.shape[0], xb.shape[1], xb.shape[2], xb.shape[3],
+                R, S, padding, box["x"], box["mask"], box["mean"],
+                box["invstd"])
+            box["slab"] = slab
+            box["dx_ref"] = dx
+        else:
+            dx = ext.conv2d_dgrad(dyb, wt2, xb.shape[0], xb.shape[1],
+                                  xb.shape[2], xb.shape[3], R, S,
+                                  stride, padding)
+            if pad8:
+                dx = dx[:, :C]
+            if x_dtype == torch.float32:
+                dx = dx.float()
     if ctx.needs_input_grad[1]:
         dw = ext.conv2d_wgrad(dyb, xb, R, S, stride, padding)
         if pad8:
@@ -209,6 +221,10 @@ class _HIPConv2d(torch.autograd.Function):
         ctx.save_for_backward(xb, weight)
         ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
                     False)
+        if stride == 1 and xb is x \
+                and os.environ.get("DDPX_NO_BNFUSE", "0") != "1":
+            # producer-side BN-backward fusion (attr set by batch_norm)
+            ctx.bn_box = getattr(x, "_ddpx_bnbwd", None)
         return y
 
     @staticmethod
@@ -294,6 +310,13 @@ class _HIPBatchNorm(torch.autograd.Function):
                 xb, weight.detach(), bias.detach(), running_mean, running_var,
                 momentum, eps, fuse_relu, rb, pre_stats)
             ctx.save_for_backward(xb, weight, save_mean, save_invstd, mask)
+            # producer-side bwd fusion: the conv that consumes y can emit
+            # this BN's backward partials from its dgrad epilogue; the box
+            # carries what it needs and brings the result back (used only
+            # when dy's identity proves a single consumer — see backward)
+            ctx.bnbwd_box = {"x": xb, "mask": mask, "mean": save_mean,
+                             "invstd": save_invstd, "slab": None,
+                             "dx_ref": None}
         else:
             y = ext.bn_fwd_eval(xb, weight.detach(), bias.detach(),
                                 running_mean, running_var, eps, fuse_relu, rb)
@@ -308,9 +331,18 @@ class _HIPBatchNorm(torch.autograd.Function):
         assert ctx.training, "backward through eval-mode BN is unsupported"
         xb, weight, save_mean, save_invstd, mask = ctx.saved_tensors
         ext = require_ext_for(dy)
+        box = getattr(ctx, "bnbwd_box", None)
+        pre_slab = None
+        if (box is not None and box["slab"] is not None
+                and dy is box["dx_ref"]):
+            # dy IS the consuming conv's dgrad output (single consumer —
+            # a multi-consumer sum would be a different tensor object), so
+            # the partials it emitted are exactly this BN's
+            pre_slab = box["slab"]
         dyb = _chlast(_to_bf16(dy))
         out = ext.bn_bwd(xb, dyb, weight.detach(), save_mean,
-                         save_invstd, mask, ctx.fuse_relu, ctx.has_residual)
+                         save_invstd, mask, ctx.fuse_relu, ctx.has_residual,
+                         pre_slab)
         dx, dweight, dbias = out[0], out[1], out[2]
         dresid = out[3] if ctx.has_residual else None
         if ctx.x_dtype == torch.float32:
@@ -327,9 +359,17 @@ def batch_norm(x, running_mean, running_var, weight, bias,
     pre_stats: [2,C,S] partial sums emitted by the producing conv's epilogue
     (conv2d_stats) — skips the BN statistics pass entirely."""
     if x.is_cuda and require_ext_for(x) is not None:
-        return _HIPBatchNorm.apply(x, weight, bias, running_mean, running_var,
-                                   training, momentum, eps, fuse_relu,
-                                   residual, pre_stats)
+        y = _HIPBatchNorm.apply(x, weight, bias, running_mean, running_var,
+                                training, momentum, eps, fuse_relu,
+                                residual, pre_stats)
+        # surface the producer-fusion box on the tensor object so the
+        # consuming conv can find it (object-attribute, not data_ptr —
+        # immune to allocator reuse); ctx attributes are visible on grad_fn
+        if training and y.grad_fn is not None:
+            box = getattr(y.grad_fn, "bnbwd_box", None)
+            if box is not None:
+                y._ddpx_bnbwd = box
+        return y
     xf = x.float() if x.dtype != torch.float32 else x
     y = F.batch_norm(xf, running_mean, running_var, weight, bias,
                      training, momentum, eps)

@@ -43,7 +43,8 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
                                at::Tensor mask, bool fuse_relu,
-                               bool want_dresid);
+                               bool want_dresid,
+                               c10::optional<at::Tensor> pre_slab);
 
 // gemm.hip
 at::Tensor gemm_tn(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> bias,
@@ -62,6 +63,12 @@ std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
                                          long stride, long pad);
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C, long H,
                         long W, long R, long S, long stride, long pad);
+std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor dy, at::Tensor wt2,
+                                        long N, long C, long H, long W,
+                                        long R, long S, long pad,
+                                        at::Tensor bn_x, at::Tensor bn_mask,
+                                        at::Tensor bn_mean,
+                                        at::Tensor bn_invstd);
 at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
                         long stride, long pad);
 
@@ -105,7 +112,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_bwd", &bn_bwd,
           py::arg("x"), py::arg("dy"), py::arg("gamma"), py::arg("save_mean"),
           py::arg("save_invstd"), py::arg("mask"), py::arg("fuse_relu") = false,
-          py::arg("want_dresid") = false);
+          py::arg("want_dresid") = false, py::arg("pre_slab") = c10::nullopt);
     m.def("gemm_tn", &gemm_tn, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
     m.def("transpose_bf16", &transpose_bf16);
@@ -121,5 +128,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
           py::arg("pad") = 0);
     m.def("conv2d_dgrad", &conv2d_dgrad);
+    m.def("conv2d_dgrad_bn", &conv2d_dgrad_bn);
     m.def("conv2d_wgrad", &conv2d_wgrad);
 }

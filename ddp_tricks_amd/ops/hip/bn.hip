@@ -471,7 +471,8 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
                                at::Tensor mask, bool fuse_relu,
-                               bool want_dresid) {
+                               bool want_dresid,
+                               c10::optional<at::Tensor> pre_slab) {
     long M; int C;
     shape_mc(x, M, C);
     TORCH_CHECK(C % 8 == 0 && C / 8 <= 256);
@@ -479,8 +480,12 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
     auto fopts = gamma.options().dtype(at::kFloat);
     int block = pick_block(C);
     int nw = block / (C / 8);
-    int S = bn_splits(M, nw);
-    auto slab = at::empty({2, C, S}, fopts);
+    // pre_slab: backward partials emitted by the downstream conv's dgrad
+    // epilogue (producer-side fusion) — the partial pass is skipped.
+    int S = pre_slab.has_value() ? (int)pre_slab->size(2)
+                                 : bn_splits(M, nw);
+    auto slab = pre_slab.has_value() ? *pre_slab
+                                     : at::empty({2, C, S}, fopts);
     auto dgamma = at::empty({C}, fopts);
     auto dbeta = at::empty({C}, fopts);
     auto ca = at::empty({C}, fopts);
@@ -497,12 +502,14 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
         ? mask.data_ptr<unsigned char>() : nullptr;
     int lds = 2 * nw * C * 4;
 
-    hipLaunchKernelGGL(k_bn_bwd_partial, dim3(S), dim3(block), lds,
-                       stream.stream(), xp, dyp, yp,
-                       save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>(), M, C, S, fuse_relu,
-                       slab.data_ptr<float>());
-    HIP_CHECK_LAST();
+    if (!pre_slab.has_value()) {
+        hipLaunchKernelGGL(k_bn_bwd_partial, dim3(S), dim3(block), lds,
+                           stream.stream(), xp, dyp, yp,
+                           save_mean.data_ptr<float>(),
+                           save_invstd.data_ptr<float>(), M, C, S, fuse_relu,
+                           slab.data_ptr<float>());
+        HIP_CHECK_LAST();
+    }
     hipLaunchKernelGGL(k_bn_bwd_combine, dim3(ceil_div_i(C, 4)), dim3(256), 0,
                        stream.stream(), slab.data_ptr<float>(), S, C, M,
                        gamma.data_ptr<float>(), save_invstd.data_ptr<float>(),

@@ -62,13 +62,27 @@ static inline void init_fastdiv(ConvShape& cs) {
 // that leaves LDS deterministic).
 __device__ __align__(16) unsigned char g_zero16[16];
 
+// Producer-side BN-backward fusion (MODE 1): the dgrad output dx IS the
+// upstream BN's dy; while the staged epilogue streams dx out, it also
+// accumulates that BN's backward partial sums (sum(dy_eff) and
+// sum(dy_eff * xhat)) so the BN's own partial pass — a full re-read of
+// x and dy — is skipped.
+struct BnFuse {
+    const bf16* x;                  // BN input (bf16 channels_last)
+    const unsigned char* mask;      // packed relu mask (byte per 8 ch)
+    const float* mean;
+    const float* invstd;
+    float* slab;                    // [2][C][gridM]
+};
+
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
           int CBM_T = CBM>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
                  ConvShape cs, int M, int Kgemm, int Nout,
-                 float* __restrict__ stats = nullptr) {
+                 float* __restrict__ stats = nullptr,
+                 BnFuse bn = BnFuse{}) {
     constexpr int WM = CBM_T / WAVES_M;
     constexpr int WN = TBN / WAVES_N;
     constexpr int MI = WM / 16;
@@ -308,10 +322,22 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                 }
             }
         __syncthreads();
-        // write phase: each 16-lane group streams one row's TBN columns
+        // write phase: each 16-lane group streams one row's TBN columns.
+        // With bn.slab set (MODE 1), the same sweep accumulates the
+        // upstream BN's backward partials from the in-flight dx chunks.
         constexpr int ROWS_PER_PASS = 256 / (TBN / 8);
         const int rsub = tid / (TBN / 8);
         const int csub = (tid % (TBN / 8)) * 8;
+        const bool do_bn = MODE == 1 && bn.slab != nullptr;
+        float bsum[8] = {}, bsx[8] = {};
+        float bmean[8], binv[8];
+        if (do_bn && n0 + csub < Nout) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                bmean[j] = bn.mean[n0 + csub + j];
+                binv[j] = bn.invstd[n0 + csub + j];
+            }
+        }
         for (int r0_ = 0; r0_ < CBM_T; r0_ += ROWS_PER_PASS) {
             int rl = r0_ + rsub;
             long row = (long)m0 + rl;
@@ -319,8 +345,49 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             if (rl < CBM_T && row < M && col < Nout) {
                 int i = rl * TBN + csub;    // 16 B chunk within one half
                 const bf16* src16 = i < HALF ? &stageA[i] : &stageB[i - HALF];
-                *reinterpret_cast<bf16x8_t*>(&out[row * Nout + col]) =
-                    *reinterpret_cast<const bf16x8_t*>(src16);
+                bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(src16);
+                *reinterpret_cast<bf16x8_t*>(&out[row * Nout + col]) = v;
+                if (do_bn) {
+                    unsigned m8 = bn.mask
+                        ? bn.mask[row * (Nout >> 3) + (csub >> 3)] : 0xffu;
+                    s16x8 xv = *reinterpret_cast<const s16x8*>(
+                        &bn.x[row * Nout + col]);
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        float g = (m8 >> j) & 1u
+                            ? us2f((unsigned short)((s16x8&)v)[j]) : 0.f;
+                        float xh = (us2f((unsigned short)xv[j]) - bmean[j])
+                                   * binv[j];
+                        bsum[j] += g;
+                        bsx[j] = fmaf(g, xh, bsx[j]);
+                    }
+                }
+            }
+        }
+        if (do_bn) {
+            // reduce over the ROWS_PER_PASS rsub groups sharing each csub
+            __syncthreads();                  // stage reads done; reuse LDS
+            float* red = reinterpret_cast<float*>(stageA);  // [rsub][TBN][2]
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                red[(rsub * TBN + csub + j) * 2] = bsum[j];
+                red[(rsub * TBN + csub + j) * 2 + 1] = bsx[j];
+            }
+            __syncthreads();
+            const int gridM = gridDim.x;
+            for (int cl = tid; cl < TBN; cl += 256) {
+                float a = 0.f, b2 = 0.f;
+                #pragma unroll 4
+                for (int w = 0; w < ROWS_PER_PASS; ++w) {
+                    a += red[(w * TBN + cl) * 2];
+                    b2 += red[(w * TBN + cl) * 2 + 1];
+                }
+                int col = n0 + cl;
+                if (col < Nout) {
+                    bn.slab[(long)col * gridM + blockIdx.x] = a;
+                    bn.slab[(long)Nout * gridM + (long)col * gridM
+                            + blockIdx.x] = b2;
+                }
             }
         }
     }
@@ -1516,6 +1583,56 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
         }
     }
     return dx;
+}
+
+std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor dy, at::Tensor wt2,
+                                        long N, long C, long H, long W,
+                                        long R, long S, long pad,
+                                        at::Tensor bn_x, at::Tensor bn_mask,
+                                        at::Tensor bn_mean,
+                                        at::Tensor bn_invstd) {
+    // stride-1 dgrad that ALSO emits the upstream BN's backward partial
+    // sums ([2][C][gridM], fed to bn_bwd's pre_slab) — the BN partial pass
+    // never re-reads x/dy.
+    ConvShape cs;
+    cs.N = N; cs.C = C; cs.H = H; cs.W = W;
+    cs.Ko = dy.size(1); cs.P = dy.size(2); cs.Q = dy.size(3);
+    cs.R = R; cs.S = S; cs.stride = 1; cs.pad = pad;
+    init_fastdiv(cs);
+    TORCH_CHECK(cs.Ko % 8 == 0 && cs.C % 8 == 0);
+    int Kgemm = cs.R * cs.S * cs.Ko;
+    long M = (long)cs.N * cs.H * cs.W;
+    auto dx = at::empty({(long)cs.N, (long)cs.C, (long)cs.H, (long)cs.W},
+                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = at::hip::getCurrentHIPStream();
+    int gridM = ceil_div_i(M, CBM);
+    auto slab = at::empty({2, (long)cs.C, (long)gridM},
+                          dy.options().dtype(at::kFloat));
+    BnFuse bn;
+    bn.x = reinterpret_cast<const bf16*>(bn_x.data_ptr());
+    bn.mask = bn_mask.numel()
+        ? bn_mask.data_ptr<unsigned char>() : nullptr;
+    bn.mean = bn_mean.data_ptr<float>();
+    bn.invstd = bn_invstd.data_ptr<float>();
+    bn.slab = slab.data_ptr<float>();
+    const bf16* dyp_ = reinterpret_cast<const bf16*>(dy.data_ptr());
+    const bf16* wt2p = reinterpret_cast<const bf16*>(wt2.data_ptr());
+    bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
+    if (cs.C >= 128) {
+        dim3 grid(gridM, ceil_div_i(cs.C, 128));
+        hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
+                           dim3(256), 0, stream.stream(), dyp_, wt2p,
+                           nullptr, dxp, cs, (int)M, Kgemm, cs.C, nullptr,
+                           bn);
+    } else {
+        dim3 grid(gridM, ceil_div_i(cs.C, 64));
+        hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
+                           dim3(256), 0, stream.stream(), dyp_, wt2p,
+                           nullptr, dxp, cs, (int)M, Kgemm, cs.C, nullptr,
+                           bn);
+    }
+    HIP_CHECK_LAST();
+    return {dx, slab};
 }
 
 at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,

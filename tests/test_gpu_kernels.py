@@ -611,3 +611,65 @@ def test_bn_shape_fuzz():
         _close(dg, g2.grad, rel=2e-2, atol=0.2, name=f"{tag} dgamma")
         _close(db, b2.grad, rel=2e-2, atol=0.2, name=f"{tag} dbeta")
         _close(dx, xf.grad, rel=5e-2, atol=5e-2, name=f"{tag} dx")
+
+
+def test_producer_side_bn_bwd_fusion():
+    """conv dgrad emits the upstream BN's backward partials; gradients must
+    match the unfused path, and a residual junction (multi-consumer BN
+    output) must transparently fall back — exercised end-to-end."""
+    import os
+
+    from ddp_tricks_amd.ops.functional import batch_norm, conv2d
+
+    def chain(fuse):
+        os.environ["DDPX_NO_BNFUSE"] = "0" if fuse else "1"
+        try:
+            torch.manual_seed(31)
+            x = torch.randn(8, 64, 20, 20, device=DEV).to(torch.bfloat16)\
+                .contiguous(memory_format=CL).float().requires_grad_(True)
+            g = torch.nn.Parameter(torch.rand(64, device=DEV) + 0.5)
+            b = torch.nn.Parameter(torch.randn(64, device=DEV))
+            w = torch.nn.Parameter(torch.randn(128, 64, 3, 3, device=DEV)
+                                   .to(torch.bfloat16).float())
+            rm = torch.zeros(64, device=DEV)
+            rv = torch.ones(64, device=DEV)
+            y = batch_norm(x, rm, rv, g, b, True, 0.1, 1e-5, fuse_relu=True)
+            z = conv2d(y, w, None, stride=1, padding=1)
+            z.float().pow(2).mean().backward()
+            return (x.grad.clone(), g.grad.clone(), b.grad.clone(),
+                    w.grad.clone())
+        finally:
+            os.environ["DDPX_NO_BNFUSE"] = "0"
+
+    fused = chain(True)
+    plain = chain(False)
+    for a, c, name in zip(fused, plain, ("dx", "dgamma", "dbeta", "dw")):
+        _close(a, c, rel=1e-2, atol=1e-2, name=f"bnfuse {name}")
+
+    # residual junction: BN output feeds conv AND a skip add — the sum
+    # tensor differs from the conv's dx, so the fusion must NOT be used
+    # (wrong partials would corrupt dgamma/dbeta); verify grads match the
+    # unfused path exactly in that topology too
+    def junction(fuse):
+        os.environ["DDPX_NO_BNFUSE"] = "0" if fuse else "1"
+        try:
+            torch.manual_seed(32)
+            x = torch.randn(8, 64, 16, 16, device=DEV).to(torch.bfloat16)\
+                .contiguous(memory_format=CL).float().requires_grad_(True)
+            g = torch.nn.Parameter(torch.rand(64, device=DEV) + 0.5)
+            b = torch.nn.Parameter(torch.randn(64, device=DEV))
+            w = torch.nn.Parameter(torch.randn(64, 64, 3, 3, device=DEV)
+                                   .to(torch.bfloat16).float())
+            rm = torch.zeros(64, device=DEV)
+            rv = torch.ones(64, device=DEV)
+            y = batch_norm(x, rm, rv, g, b, True, 0.1, 1e-5, fuse_relu=True)
+            z = conv2d(y, w, None, stride=1, padding=1) + y  # two consumers
+            z.float().pow(2).mean().backward()
+            return (x.grad.clone(), g.grad.clone(), b.grad.clone())
+        finally:
+            os.environ["DDPX_NO_BNFUSE"] = "0"
+
+    fj = junction(True)
+    pj = junction(False)
+    for a, c, name in zip(fj, pj, ("dx", "dgamma", "dbeta")):
+        _close(a, c, rel=1e-3, atol=1e-3, name=f"junction {name}")
