@@ -137,6 +137,10 @@ def _conv_fwd_prep(ctx, x, weight, bias, stride, padding):
     ctx.save_for_backward(xb, weight)
     ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
                 False)
+    if stride == 1 and xb is x \
+            and os.environ.get("DDPX_NO_BNFUSE", "0") != "1":
+        # producer-side BN-backward fusion (attr set by batch_norm)
+        ctx.bn_box = getattr(x, "_ddpx_bnbwd", None)
     return ext, xb, wb, (bias if bias is None else bias.detach().float())
 
 
