@@ -138,8 +138,12 @@ def _conv_fwd_prep(ctx, x, weight, bias, stride, padding):
     ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
                 False)
     if stride == 1 and xb is x \
-            and os.environ.get("DDPX_NO_BNFUSE", "0") != "1":
-        # producer-side BN-backward fusion (attr set by batch_norm)
+            and os.environ.get("DDPX_BNFUSE", "0") == "1":
+        # producer-side BN-backward fusion (attr set by batch_norm).
+        # OPT-IN: measured net-negative at ResNet-50 (8.38k vs 8.72k
+        # img/s) — the epilogue's +32 VGPR (partial accumulators +
+        # coefficient cache) pushes the dgrad to the 512-reg cliff while
+        # the skipped BN partial pass only saves ~1 activation read.
         ctx.bn_box = getattr(x, "_ddpx_bnbwd", None)
     return ext, xb, wb, (bias if bias is None else bias.detach().float())
 
@@ -226,8 +230,8 @@ class _HIPConv2d(torch.autograd.Function):
         ctx.meta = (stride, padding, x.dtype, weight.shape, bias is not None,
                     False)
         if stride == 1 and xb is x \
-                and os.environ.get("DDPX_NO_BNFUSE", "0") != "1":
-            # producer-side BN-backward fusion (attr set by batch_norm)
+                and os.environ.get("DDPX_BNFUSE", "0") == "1":
+            # producer-side BN-backward fusion (opt-in; see _conv_fwd_prep)
             ctx.bn_box = getattr(x, "_ddpx_bnbwd", None)
         return y
 

@@ -622,7 +622,7 @@ def test_producer_side_bn_bwd_fusion():
     from ddp_tricks_amd.ops.functional import batch_norm, conv2d
 
     def chain(fuse):
-        os.environ["DDPX_NO_BNFUSE"] = "0" if fuse else "1"
+        os.environ["DDPX_BNFUSE"] = "1" if fuse else "0"
         try:
             torch.manual_seed(31)
             x = torch.randn(8, 64, 20, 20, device=DEV).to(torch.bfloat16)\
@@ -639,7 +639,7 @@ def test_producer_side_bn_bwd_fusion():
             return (x.grad.clone(), g.grad.clone(), b.grad.clone(),
                     w.grad.clone())
         finally:
-            os.environ["DDPX_NO_BNFUSE"] = "0"
+            os.environ["DDPX_BNFUSE"] = "0"
 
     fused = chain(True)
     plain = chain(False)
@@ -651,7 +651,7 @@ def test_producer_side_bn_bwd_fusion():
     # (wrong partials would corrupt dgamma/dbeta); verify grads match the
     # unfused path exactly in that topology too
     def junction(fuse):
-        os.environ["DDPX_NO_BNFUSE"] = "0" if fuse else "1"
+        os.environ["DDPX_BNFUSE"] = "1" if fuse else "0"
         try:
             torch.manual_seed(32)
             x = torch.randn(8, 64, 16, 16, device=DEV).to(torch.bfloat16)\
@@ -667,7 +667,7 @@ def test_producer_side_bn_bwd_fusion():
             z.float().pow(2).mean().backward()
             return (x.grad.clone(), g.grad.clone(), b.grad.clone())
         finally:
-            os.environ["DDPX_NO_BNFUSE"] = "0"
+            os.environ["DDPX_BNFUSE"] = "0"
 
     fj = junction(True)
     pj = junction(False)
