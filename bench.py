@@ -69,7 +69,11 @@ class SyntheticData:
     copy stream (mirrors the training pipeline's prefetch)."""
 
     def __init__(self, batch: int, device, shape=(1, 28, 28), classes=10,
-                 n_buffers: int = 4, h2d: bool = True):
+                 n_buffers: int = None, h2d: bool = True):
+        if n_buffers is None:
+            # ImageNet-sized staging is ~600 MB/buffer at B=1024: keep the
+            # pinned footprint sane for 8 ranks on one host
+            n_buffers = 2 if shape[-1] >= 128 else 4
         g = torch.Generator().manual_seed(1234)
         self.h2d = h2d and device.type == "cuda"
         self.device = device
