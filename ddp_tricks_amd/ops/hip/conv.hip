@@ -1124,8 +1124,12 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
                             const bf16* __restrict__ x,
                             float* __restrict__ slab, ConvShape cs, long M,
                             int Kgemm, int S) {
-    __shared__ bf16 lds_a[128][DEPTH + 8];
-    __shared__ bf16 lds_b[128][DEPTH + 8];
+    // row stride DEPTH+24: PMC showed 2.8-4.6k LDS-conflict stall cycles
+    // per wave at +8 padding; +24 keeps the 16-lane b128 fragment reads
+    // bank-disjoint (stride 44/28 dwords, gcd 4, all 16 starts distinct)
+    constexpr int SLD = DEPTH + 24;
+    __shared__ bf16 lds_a[128][SLD];
+    __shared__ bf16 lds_b[128][SLD];
     const int ko0 = blockIdx.x * 128;
     const int rc0 = blockIdx.y * 128;
     const int split = blockIdx.z;
