@@ -1127,7 +1127,13 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
     // row stride DEPTH+24: PMC showed 2.8-4.6k LDS-conflict stall cycles
     // per wave at +8 padding; +24 keeps the 16-lane b128 fragment reads
     // bank-disjoint (stride 44/28 dwords, gcd 4, all 16 starts distinct)
-    constexpr int SLD = DEPTH + 24;
+    // NOTE: row-stride padding (+8 → +24) measured ZERO effect on the
+    // LDS conflict counter (2848/4608 cycles/wave both ways) — the
+    // conflicts are between the (lane>>4) k-sub-groups of the b128
+    // fragment reads, which a row-stride change cannot separate.  The
+    // XOR 8-block swizzle below (phys m-block = m ^ ((row&3)<<3), same
+    // idiom as the fwd glds swizzle) is the fix that addresses it.
+    constexpr int SLD = DEPTH + 8;
     __shared__ bf16 lds_a[128][SLD];
     __shared__ bf16 lds_b[128][SLD];
     const int ko0 = blockIdx.x * 128;
@@ -1197,8 +1203,10 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
                     | ((unsigned)(unsigned short)a1[jj] << 16);
                 unsigned pb = (unsigned)(unsigned short)b0[jj]
                     | ((unsigned)(unsigned short)b1[jj] << 16);
-                *reinterpret_cast<unsigned*>(&lds_a[j + jj][m2]) = pa;
-                *reinterpret_cast<unsigned*>(&lds_b[j + jj][m2]) = pb;
+                const int row = j + jj;
+                const int mph = m2 ^ ((row & 3) << 3);   // 8-block swizzle
+                *reinterpret_cast<unsigned*>(&lds_a[row][mph]) = pa;
+                *reinterpret_cast<unsigned*>(&lds_b[row][mph]) = pb;
             }
         }
         __syncthreads();
@@ -1208,13 +1216,17 @@ void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
             bf16x8_t af[4], bfr[4];
             const int kcol = ks + (lane >> 4) * 8;
             #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < 4; ++mi) {
+                const int row = wr * 64 + mi * 16 + (lane & 15);
                 af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[wr * 64 + mi * 16 + (lane & 15)][kcol]);
+                    &lds_a[row][kcol ^ ((row & 3) << 3)]);
+            }
             #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < 4; ++ni) {
+                const int row = wc * 64 + ni * 16 + (lane & 15);
                 bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[wc * 64 + ni * 16 + (lane & 15)][kcol]);
+                    &lds_b[row][kcol ^ ((row & 3) << 3)]);
+            }
             #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 #pragma unroll
