@@ -903,7 +903,9 @@ void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
             for (int jj = 0; jj < 8; ++jj) {
                 unsigned pk = (unsigned)(unsigned short)t0[jj]
                     | ((unsigned)(unsigned short)t1[jj] << 16);
-                *reinterpret_cast<unsigned*>(&dst[(j8 + jj) * SLD + m2]) = pk;
+                const int row = j8 + jj;
+                const int mph = m2 ^ ((row & 3) << 3);   // 8-block swizzle
+                *reinterpret_cast<unsigned*>(&dst[row * SLD + mph]) = pk;
             }
         }
         __syncthreads();
@@ -913,13 +915,17 @@ void k_conv_wgrad_sb_pair(const bf16* __restrict__ dy,
             bf16x8_t af[2], bfr[2];
             const int kcol = ks + (lane >> 4) * 8;
             #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < 2; ++mi) {
+                const int row = wr * 32 + mi * 16 + (lane & 15);
                 af[mi] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_a[wr * 32 + mi * 16 + (lane & 15)][kcol]);
+                    &lds_a[row][kcol ^ ((row & 3) << 3)]);
+            }
             #pragma unroll
-            for (int ni = 0; ni < 2; ++ni)
+            for (int ni = 0; ni < 2; ++ni) {
+                const int row = wc * 32 + ni * 16 + (lane & 15);
                 bfr[ni] = *reinterpret_cast<const bf16x8_t*>(
-                    &lds_b[wc * 32 + ni * 16 + (lane & 15)][kcol]);
+                    &lds_b[row][kcol ^ ((row & 3) << 3)]);
+            }
             #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
                 #pragma unroll
