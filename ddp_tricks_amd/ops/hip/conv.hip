@@ -622,6 +622,33 @@ __global__ void k_wgrad_combine(const float* __restrict__ slab, int S,
     }
 }
 
+// Stage-1 combine for deep splits: sum CH consecutive s-slices per
+// (chunk, i) into slab2[chunk][i] — identity layout, fixed order
+// (deterministic).  Two passes beat one when S is deep: the single-pass
+// combine is latency-bound (e.g. conv2's S=128 slab gives only 288
+// blocks, each thread chewing 128 dependent 294-KB-strided loads —
+// measured 40.7 us avg where ~38 MB of reads should take <6); this pass
+// runs chunks× more waves and leaves the final KCRS scatter a shallow
+// S/CH-deep sum.
+__global__ void k_wgrad_combine_stage(const float* __restrict__ slab, int S,
+                                      int CH, long total,
+                                      float* __restrict__ slab2) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    const int c0 = blockIdx.y * CH;
+    const int ce = min(S, c0 + CH);
+    // two independent accumulator chains (fixed even/odd order — still
+    // deterministic) so two loads are in flight per iteration
+    float v0 = 0.f, v1 = 0.f;
+    int s = c0;
+    for (; s + 1 < ce; s += 2) {
+        v0 += slab[(long)s * total + i];
+        v1 += slab[(long)(s + 1) * total + i];
+    }
+    if (s < ce) v0 += slab[(long)s * total + i];
+    slab2[(long)blockIdx.y * total + i] = v0 + v1;
+}
+
 // Single-buffer 32-deep wgrad variant (the pre-pipelining shape; kept for
 // A/B selection via DDPX_WGRAD_V=sb — the profiler decides, not intuition).
 constexpr int SBK = 32;
@@ -1789,10 +1816,26 @@ at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
                            cs, M, Kgemm, S_);
     HIP_CHECK_LAST();
     long total = (long)cs.Ko * Kgemm;
+    at::Tensor slab_f = slab;
+    int S_c = S_;
+    if (S_ > 16) {
+        constexpr int CH = 8;
+        const int chunks = ceil_div_i(S_, CH);
+        auto slab2 = at::empty({chunks, total},
+                               x.options().dtype(at::kFloat));
+        const int blocks1 = (int)((total + 255) / 256);
+        hipLaunchKernelGGL(k_wgrad_combine_stage, dim3(blocks1, chunks),
+                           dim3(256), 0, stream.stream(),
+                           slab.data_ptr<float>(), S_, CH, total,
+                           slab2.data_ptr<float>());
+        HIP_CHECK_LAST();
+        slab_f = slab2;
+        S_c = chunks;
+    }
     int blocks = std::min<long>(4096, ceil_div_i(total, 256));
     hipLaunchKernelGGL(k_wgrad_combine, dim3(blocks), dim3(256), 0,
-                       stream.stream(), slab.data_ptr<float>(), S_, cs, Kgemm,
-                       dw.data_ptr<float>());
+                       stream.stream(), slab_f.data_ptr<float>(), S_c, cs,
+                       Kgemm, dw.data_ptr<float>());
     HIP_CHECK_LAST();
     return dw;
 }
