@@ -470,6 +470,30 @@ def relu(x, inplace=False):
     return F.relu(x, inplace=inplace)
 
 
+class _HIPFlattenNHWC(torch.autograd.Function):
+    """[N,C,H,W] channels_last bf16 -> [N, C*H*W] in NCHW semantic order
+    (the reference fc1 weight layout, reference utils/model.py:23-25) with
+    coalesced HIP transposes both ways — replaces ATen's strided
+    permute-copies at the conv->dense junction."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_ext_for(x)
+        ctx.chw = (x.shape[1], x.shape[2], x.shape[3])
+        return ext.nhwc_flatten(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext_for(dy)
+        return ext.nhwc_unflatten(dy.contiguous(), *ctx.chw)
+
+
+def nhwc_flatten(x):
+    if require_ext_for(x) is not None:
+        return _HIPFlattenNHWC.apply(x)
+    return torch.flatten(x, 1)
+
+
 # ------------------------------------------------------------ cross entropy ---
 
 class _HIPCrossEntropy(torch.autograd.Function):

@@ -11,6 +11,8 @@ void fused_lookahead(std::vector<at::Tensor> fast, std::vector<at::Tensor> slow,
                      double alpha, c10::optional<at::Tensor> found_inf);
 at::Tensor cast_to_bf16(at::Tensor x);
 at::Tensor pad8_channels(at::Tensor x);
+at::Tensor nhwc_flatten(at::Tensor x);
+at::Tensor nhwc_unflatten(at::Tensor dy, long C, long H, long W);
 std::vector<at::Tensor> pack_conv_weight(at::Tensor w, bool pad8,
                                          bool want_wt2);
 
@@ -88,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("found_inf") = c10::nullopt);
     m.def("cast_to_bf16", &cast_to_bf16);
     m.def("pad8_channels", &pad8_channels);
+    m.def("nhwc_flatten", &nhwc_flatten);
+    m.def("nhwc_unflatten", &nhwc_unflatten);
     m.def("pack_conv_weight", &pack_conv_weight, py::arg("w"),
           py::arg("pad8") = false, py::arg("want_wt2") = true);
     m.def("ce_fwd", &ce_fwd);

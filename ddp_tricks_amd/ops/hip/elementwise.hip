@@ -347,3 +347,92 @@ at::Tensor pad8_channels(at::Tensor x) {
     HIP_CHECK_LAST();
     return xp;
 }
+
+// NHWC flatten boundary (Toy_Net conv->dense junction): [N,C,H,W]
+// channels_last bf16 -> [N, C*H*W] in NCHW semantic order (the order the
+// reference's fc1 weight layout expects; reference utils/model.py:23-25),
+// plus the inverse for backward.  Replaces two ~23 us strided ATen
+// permute-copies per step: here the fwd gather is coalesced across the
+// c-threads (consecutive c -> consecutive addresses within each s row)
+// and each thread writes its HW-contiguous chunk with s16x8 stores.
+__global__ void k_nhwc_flatten(const bf16* __restrict__ x,
+                               bf16* __restrict__ y, int C, int HW) {
+    const int c = blockIdx.y * blockDim.x + threadIdx.x;
+    const long b = blockIdx.x;
+    if (c >= C) return;
+    const bf16* src = x + (long)b * HW * C + c;
+    bf16* dst = y + ((long)b * C + c) * HW;
+    // NO runtime-indexed staging array (it would spill to scratch — the
+    // HW bound is runtime); fixed-8 inner unroll keeps the gather in
+    // registers and the store a single s16x8.
+    if ((HW & 7) == 0) {
+        for (int s8 = 0; s8 < HW; s8 += 8) {
+            s16x8 v;
+            #pragma unroll
+            for (int k = 0; k < 8; ++k)
+                v[k] = *(const short*)&src[(long)(s8 + k) * C];
+            *reinterpret_cast<s16x8*>(&dst[s8]) = v;
+        }
+    } else {
+        for (int s = 0; s < HW; ++s)
+            *(short*)&dst[s] = *(const short*)&src[(long)s * C];
+    }
+}
+
+__global__ void k_nhwc_unflatten(const bf16* __restrict__ dy,
+                                 bf16* __restrict__ dx, int C, int HW) {
+    const int c = blockIdx.y * blockDim.x + threadIdx.x;
+    const long b = blockIdx.x;
+    if (c >= C) return;
+    const bf16* src = dy + ((long)b * C + c) * HW;   // contiguous chunk
+    bf16* dst = dx + (long)b * HW * C + c;
+    if ((HW & 7) == 0) {
+        for (int s8 = 0; s8 < HW; s8 += 8) {
+            const s16x8 v = *reinterpret_cast<const s16x8*>(&src[s8]);
+            #pragma unroll
+            for (int k = 0; k < 8; ++k)
+                *(short*)&dst[(long)(s8 + k) * C] = v[k];
+        }
+    } else {
+        for (int s = 0; s < HW; ++s)
+            *(short*)&dst[(long)s * C] = *(const short*)&src[s];
+    }
+}
+
+at::Tensor nhwc_flatten(at::Tensor x) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+                x.dim() == 4, "nhwc_flatten: 4D cuda bf16 expected");
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "nhwc_flatten: channels_last expected");
+    const int N = x.size(0), C = x.size(1);
+    const int HW = x.size(2) * x.size(3);
+    TORCH_CHECK(HW <= 64, "nhwc_flatten: HW too large");
+    auto y = at::empty({(long)N, (long)C * HW}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(k_nhwc_flatten,
+                       dim3(N, ceil_div_i(C, 256)), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()), C, HW);
+    HIP_CHECK_LAST();
+    return y;
+}
+
+at::Tensor nhwc_unflatten(at::Tensor dy, long C, long H, long W) {
+    TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
+                dy.dim() == 2 && dy.is_contiguous(),
+                "nhwc_unflatten: 2D contiguous cuda bf16 expected");
+    const long N = dy.size(0);
+    const int HW = (int)(H * W);
+    TORCH_CHECK(dy.size(1) == C * HW && HW <= 64);
+    auto dx = at::empty({N, C, H, W},
+                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(k_nhwc_unflatten,
+                       dim3(N, ceil_div_i((int)C, 256)), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dy.data_ptr()),
+                       reinterpret_cast<bf16*>(dx.data_ptr()), (int)C, HW);
+    HIP_CHECK_LAST();
+    return dx;
+}

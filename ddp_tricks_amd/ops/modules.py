@@ -9,6 +9,8 @@ reference semantics.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -79,7 +81,21 @@ class AdaptiveAvgPool2d(nn.AdaptiveAvgPool2d):
 
 
 class Flatten(nn.Flatten):
-    pass
+    """flatten(start_dim=1) with a dedicated kernel at the NHWC conv→dense
+    junction: ``nn.Flatten`` on a channels_last tensor materialises the
+    NCHW semantic order through ATen's strided permute-copy (~23 µs each
+    way per Toy_Net step); the HIP pair does the same transpose with
+    coalesced access both directions (reference utils/model.py:23)."""
+
+    def forward(self, x):
+        if (self.start_dim == 1 and self.end_dim == -1 and x.dim() == 4
+                and x.is_cuda and x.dtype == torch.bfloat16
+                and 1 < x.size(2) * x.size(3) <= 64 and x.size(1) > 1
+                and not x.is_contiguous()  # NCHW-contig flatten is a view
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and os.environ.get("DDPX_NHWC_FLATTEN", "1") == "1"):
+            return F_ops.nhwc_flatten(x)
+        return super().forward(x)
 
 
 class Identity(nn.Identity):

@@ -673,3 +673,23 @@ def test_producer_side_bn_bwd_fusion():
     pj = junction(False)
     for a, c, name in zip(fj, pj, ("dx", "dgamma", "dbeta")):
         _close(a, c, rel=1e-3, atol=1e-3, name=f"junction {name}")
+
+
+@pytest.mark.gpu
+def test_nhwc_flatten_matches_torch():
+    """nhwc_flatten == torch.flatten on the NCHW semantic order, both ways
+    (the Toy_Net conv->dense junction; reference utils/model.py:23)."""
+    from ddp_tricks_amd.ops.modules import Flatten
+    torch.manual_seed(5)
+    for shape in [(64, 512, 4, 4), (17, 96, 3, 5), (8, 300, 1, 16)]:
+        x = torch.randn(*shape, device="cuda").to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        y = Flatten()(x)
+        x_ref = x.detach().clone().requires_grad_(True)
+        y_ref = torch.flatten(x_ref, 1)
+        assert y.shape == y_ref.shape
+        assert torch.equal(y.float(), y_ref.float())
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        y_ref.backward(dy)
+        assert torch.equal(x.grad.float(), x_ref.grad.float())
