@@ -101,3 +101,13 @@ def test_train_resnet18_cifar10_cpu(tmp_path, monkeypatch):
     import torch
     sd = torch.load(os.path.join(tmp_path, "R18.pt"), weights_only=True)
     assert "layer4.1.bn2.running_var" in sd and sd["fc.weight"].shape == (10, 512)
+
+
+def test_flatten_cpu_fallback_semantics():
+    """Flatten's CPU path (and any non-dispatch GPU case) must equal
+    torch.flatten on channels_last input — NCHW semantic order."""
+    import torch
+    from ddp_tricks_amd.ops.modules import Flatten
+    x = torch.randn(3, 5, 2, 4).contiguous(memory_format=torch.channels_last)
+    y = Flatten()(x)
+    assert torch.equal(y, torch.flatten(x, 1))
