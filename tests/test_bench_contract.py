@@ -42,3 +42,32 @@ def test_bench_json_contract():
     assert cfg["model"] == "Toy_Net" and cfg["parallelism"] == "dp1"
     assert cfg["global_batch"] == 8
     assert rec["data"] == "synthetic"
+
+
+@pytest.mark.timeout(300)
+def test_bench_world2_rank0_prints_once():
+    """The driver's N>1 launch shape: one process per rank with
+    RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* env — only rank 0 prints the JSON
+    line and it reports the whole-job aggregate (n_gpus=2)."""
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29678",
+                    "RANK": str(rank), "LOCAL_RANK": str(rank),
+                    "WORLD_SIZE": "2", "DDPX_NO_TQDM": "1"})
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(REPO, "bench.py"),
+             "--steps", "2", "--warmup", "1", "--batch-size", "8"],
+            env=env, cwd=REPO, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE, text=True))
+    outs = [p.communicate(timeout=280) for p in procs]
+    assert all(p.returncode == 0 for p in procs), \
+        [o[1][-1000:] for o in outs]
+    json_lines = [ln for o, _ in outs for ln in o.splitlines()
+                  if ln.startswith("{")]
+    assert len(json_lines) == 1, json_lines   # rank 0 only
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 16      # whole-job aggregate
+    assert rec["config"]["per_gpu_batch"] == 8
