@@ -234,3 +234,34 @@ def master_params(optimizer):
     for group in optimizer.param_groups:
         for p in group["params"]:
             yield p
+
+
+def register_float_function(mod, name: str) -> None:
+    """apex-API parity (reference README.md L11 'Apex' §4: registered
+    functions run in fp32 under amp).  Wraps ``mod.name`` to cast floating
+    tensor args up to fp32 while amp is enabled.  Unlike apex this may be
+    called before OR after ``initialize`` (the wrapper checks at call
+    time)."""
+    _register_cast(mod, name, torch.float32)
+
+
+def register_half_function(mod, name: str) -> None:
+    """apex-API parity: run ``mod.name`` in the compute dtype (bf16 here —
+    MFMA-native on CDNA4, where apex used fp16)."""
+    _register_cast(mod, name, torch.bfloat16)
+
+
+def _register_cast(mod, name: str, dtype) -> None:
+    fn = getattr(mod, name)
+    if getattr(fn, "_amp_registered", False):
+        return
+
+    def wrapped(*args, **kwargs):
+        if _state.enabled:
+            args = tuple(a.to(dtype)
+                         if torch.is_tensor(a) and a.is_floating_point()
+                         else a for a in args)
+        return fn(*args, **kwargs)
+
+    wrapped._amp_registered = True
+    setattr(mod, name, wrapped)

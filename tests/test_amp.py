@@ -68,3 +68,30 @@ def test_o0_passthrough():
     with amp.scale_loss(loss, opt) as scaled:
         scaled.backward()
     assert torch.allclose(p.grad, torch.ones(2))
+
+
+def test_register_float_function_casts_up():
+    """apex-API parity: registered functions see fp32 inputs while amp is
+    enabled (reference README 'Apex' §4, commented L11)."""
+    import types
+    from ddp_tricks_amd import amp
+
+    seen = {}
+
+    def probe(x):
+        seen["dtype"] = x.dtype
+        return x
+
+    ns = types.SimpleNamespace(probe=probe)
+    amp.register_float_function(ns, "probe")
+    amp._state.enabled = True
+    try:
+        ns.probe(torch.zeros(2, dtype=torch.bfloat16))
+        assert seen["dtype"] == torch.float32
+    finally:
+        amp._state.enabled = False
+    ns.probe(torch.zeros(2, dtype=torch.bfloat16))
+    assert seen["dtype"] == torch.bfloat16   # passthrough when disabled
+    # idempotent registration
+    amp.register_float_function(ns, "probe")
+    assert ns.probe._amp_registered
