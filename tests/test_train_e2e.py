@@ -40,6 +40,12 @@ def test_train_e2e_cpu(tmp_path, monkeypatch, capsys):
     out = capsys.readouterr().out
     # epoch 0 trains at LR 0
     assert "learning_rate: 0.0," in out
+    # epoch line format conformance (reference utils/train.py:101)
+    import re
+    assert re.search(
+        r"epoch: \d{3}/3, time: \d+\.\d{2}s, learning_rate: [\d.e-]+, "
+        r"train_loss: \d+\.\d{4}, train_acc: \d+\.\d{4}, "
+        r"valid_loss: \d+\.\d{4}, valid_acc: \d+\.\d{4}", out), out
     # checkpoint written with the reference naming scheme
     assert os.path.exists(os.path.join(tmp_path, "TEST_run.pt"))
     sd = torch.load(os.path.join(tmp_path, "TEST_run.pt"))
