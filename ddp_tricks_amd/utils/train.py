@@ -20,7 +20,6 @@ import os
 import time
 
 import torch
-import torch.distributed as torch_dist
 from torch.utils.data import DataLoader
 
 from .. import amp
