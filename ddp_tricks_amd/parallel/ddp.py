@@ -154,6 +154,7 @@ class DistributedDataParallel(nn.Module):
             size += nbytes
         if bucket.params:
             self._buckets.append(bucket)
+        self._param_view = {}
         for b in self._buckets:
             total = sum(p.numel() for p in b.params)
             b.flat = torch.zeros(total, dtype=torch.float32, device=self._device)
@@ -162,6 +163,7 @@ class DistributedDataParallel(nn.Module):
                 view = b.flat.narrow(0, off, p.numel()).view_as(p)
                 p.grad = view
                 b.views.append(view)
+                self._param_view[p] = view
                 off += p.numel()
             b.expected = len(b.params)
             self._param_bucket.update({p: b for p in b.params})
@@ -173,6 +175,15 @@ class DistributedDataParallel(nn.Module):
     # ----------------------------------------------------------- runtime ---
 
     def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
+        # Self-heal: if something dropped the bucket view (a
+        # zero_grad(set_to_none=True) from user code), autograd accumulated
+        # this step's gradient into a FRESH tensor — fold it back into the
+        # bucket and re-point, or the all-reduce would reduce zeros.
+        view = self._param_view[p]
+        if p.grad is not view:
+            with torch.no_grad():
+                view.copy_(p.grad)
+            p.grad = view
         b = self._param_bucket[p]
         b.ready += 1
         if b.ready < b.expected:

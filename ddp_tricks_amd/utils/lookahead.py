@@ -87,7 +87,13 @@ class Lookahead(Optimizer):
                 group["counter"] = 0
         return loss
 
-    def zero_grad(self, set_to_none: bool = True):
+    def zero_grad(self, set_to_none: bool = False):
+        # Default False, NOT torch's True: gradients are views into the DDP
+        # bucket flats — set_to_none would drop the views, autograd would
+        # accumulate into fresh tensors, and the bucket all-reduce would
+        # reduce zeros (silent rank desync at world>1).  The DDP hook also
+        # self-heals re-pointed grads, but keeping the views is the fast
+        # path.
         return self.optimizer.zero_grad(set_to_none=set_to_none)
 
     def state_dict(self):

@@ -174,3 +174,61 @@ def test_eval_forward_resyncs_buffers():
     assert all(p.exitcode == 0 for p in procs)
     # rank 1 must hold rank 0's buffers (running_mean += 1.0, not += 2.0)
     assert abs(float(rm1.mean()) - 1.0) < 1e-5, rm1
+
+
+def _worker_full_stack_sync(rank, port, results):
+    """The production wiring (amp O1 + FusedSGD + Lookahead): rank grads and
+    params must stay identical after steps — regression for the
+    zero_grad(set_to_none=True) bucket-view drop (Lookahead's old default
+    detached grads from the bucket flats, so the all-reduce reduced zeros
+    and ranks silently desynced; the DDP hook now self-heals re-pointed
+    grads and Lookahead defaults to set_to_none=False)."""
+    _init(rank, WORLD, port)
+    from ddp_tricks_amd import amp
+    from ddp_tricks_amd.ops.optim import FusedSGD
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    from ddp_tricks_amd.utils.lookahead import Lookahead
+    amp._state.__init__()
+    torch.manual_seed(7)
+    model = TinyNet()
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, nesterov=True)
+    la = Lookahead(opt, k=2, alpha=0.5)
+    model, apex_opt = amp.initialize(model, la, "O1")
+    ddp = DDP(model)
+    torch.manual_seed(500 + rank)        # DIFFERENT data per rank
+    sigs = []
+    for step in range(3):
+        apex_opt.zero_grad(set_to_none=(step == 1))   # worst case mid-run
+        x = torch.randn(4, 8)
+        t = torch.randn(4, 4)
+        ddp.train()
+        loss = ((ddp(x) - t) ** 2).mean()
+        with amp.scale_loss(loss, apex_opt) as sl:
+            sl.backward()
+        gsig = sorted(float(p.grad.double().sum())
+                      for p in model.parameters())
+        apex_opt.step()
+        psig = sorted(float(p.detach().double().sum())
+                      for p in model.parameters())
+        sigs.append((gsig, psig))
+    amp._state.__init__()
+    results.put((rank, sigs))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_full_stack_ranks_stay_identical():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_full_stack_sync,
+                         args=(r, 29614, results)) for r in range(WORLD)]
+    [p.start() for p in procs]
+    got = dict(results.get(timeout=110) for _ in range(WORLD))
+    [p.join(timeout=60) for p in procs]
+    assert all(p.exitcode == 0 for p in procs)
+    for step in range(3):
+        g0, p0 = got[0][step]
+        g1, p1 = got[1][step]
+        assert g0 == pytest.approx(g1, abs=1e-12), f"grads differ at step {step}"
+        assert p0 == pytest.approx(p1, abs=1e-12), f"params differ at step {step}"
