@@ -39,6 +39,11 @@ def main():
         kwargs = {"num_classes": 10, "cifar_stem": True}
     model = build_model(args.model, **kwargs).to(device)
     ckpt = os.path.join(args.model_path, f"{args.exp_name}.pt")
+    if not os.path.exists(ckpt):
+        raise SystemExit(
+            f"validate.py: no checkpoint at {ckpt} — train first "
+            f"(run.py -n={args.exp_name} -p={args.model_path}) or pass "
+            f"-n/-p for an existing experiment")
     sd = torch.load(ckpt, map_location=device, weights_only=True)
     model.load_state_dict(sd)
     model.eval()
