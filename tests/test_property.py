@@ -106,3 +106,71 @@ def test_warmup_schedule_matches_reference_formula(warmup, epochs, base_lr):
     for epoch, lr in enumerate(seen):
         expect = base_lr * min(epoch / warmup, 1.0)
         assert abs(lr - expect) < 1e-12, (epoch, lr, expect)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    k=st.integers(min_value=1, max_value=7),
+    alpha=st.floats(min_value=0.1, max_value=0.9, allow_nan=False),
+    steps=st.integers(min_value=1, max_value=20),
+    seed=st.integers(min_value=0, max_value=1000),
+)
+def test_lookahead_matches_reference_algorithm(k, alpha, steps, seed):
+    """slow += alpha*(fast-slow); fast <- slow every k inner steps
+    (reference utils/lookahead.py:19-41), over random (k, alpha, steps)."""
+    from ddp_tricks_amd.utils.lookahead import Lookahead
+    torch.manual_seed(seed)
+    p_ours = torch.nn.Parameter(torch.randn(6))
+    p_ref = torch.nn.Parameter(p_ours.detach().clone())
+    grads = [torch.randn(6) for _ in range(steps)]
+
+    la = Lookahead(torch.optim.SGD([p_ours], lr=0.1), k=k, alpha=alpha)
+    # literal transcription of the reference step (utils/lookahead.py:33-41):
+    # inner step first; slow-update while counter==0 (slow lazily snapshot
+    # at its first use, i.e. AFTER the first inner step); then counter++
+    # and reset at k
+    opt_ref = torch.optim.SGD([p_ref], lr=0.1)
+    slow = None
+    counter = 0
+    for g in grads:
+        p_ours.grad = g.clone()
+        la.step()
+        p_ref.grad = g.clone()
+        opt_ref.step()
+        if counter == 0:
+            with torch.no_grad():
+                if slow is None:
+                    slow = p_ref.detach().clone()
+                slow += alpha * (p_ref.detach() - slow)
+                p_ref.copy_(slow)
+        counter += 1
+        if counter >= k:
+            counter = 0
+    assert torch.allclose(p_ours.detach(), p_ref.detach(), atol=1e-6), \
+        (k, alpha, steps)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    seq=st.lists(st.floats(min_value=0.0, max_value=5.0, allow_nan=False),
+                 min_size=1, max_size=40),
+    patience=st.integers(min_value=0, max_value=5),
+    factor=st.sampled_from([0.1, 0.5]),
+)
+def test_plateau_matches_torch(seq, patience, factor):
+    """ReduceLROnPlateau re-implementation vs torch over random valid-loss
+    sequences (reference wires torch's at utils/train.py:53)."""
+    from ddp_tricks_amd.utils.schedulers import ReduceLROnPlateau
+    p1 = torch.nn.Parameter(torch.zeros(1))
+    o1 = torch.optim.SGD([p1], lr=0.1)
+    p2 = torch.nn.Parameter(torch.zeros(1))
+    o2 = torch.optim.SGD([p2], lr=0.1)
+    ours = ReduceLROnPlateau(optimizer=o1, mode="min", factor=factor,
+                             patience=patience, verbose=False)
+    ref = torch.optim.lr_scheduler.ReduceLROnPlateau(
+        o2, mode="min", factor=factor, patience=patience)
+    for v in seq:
+        ours.step(v)
+        ref.step(v)
+        assert abs(o1.param_groups[0]["lr"]
+                   - o2.param_groups[0]["lr"]) < 1e-12
