@@ -174,3 +174,35 @@ def test_plateau_matches_torch(seq, patience, factor):
         ref.step(v)
         assert abs(o1.param_groups[0]["lr"]
                    - o2.param_groups[0]["lr"]) < 1e-12
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    momentum=st.sampled_from([0.0, 0.5, 0.9]),
+    wd=st.sampled_from([0.0, 1e-4, 1e-2]),
+    nesterov=st.booleans(),
+    lr=st.sampled_from([0.1, 0.01]),
+    steps=st.integers(min_value=1, max_value=8),
+    seed=st.integers(min_value=0, max_value=500),
+)
+def test_fused_sgd_cpu_matches_torch(momentum, wd, nesterov, lr, steps, seed):
+    """FusedSGD's CPU math vs torch.optim.SGD across the config grid
+    (reference wires SGD(momentum=0.9, nesterov=True) at utils/train.py:41)."""
+    if nesterov and momentum == 0.0:
+        return
+    from ddp_tricks_amd.ops.optim import FusedSGD
+    torch.manual_seed(seed)
+    p1 = torch.nn.Parameter(torch.randn(10))
+    p2 = torch.nn.Parameter(p1.detach().clone())
+    o1 = FusedSGD([p1], lr=lr, momentum=momentum, weight_decay=wd,
+                  nesterov=nesterov)
+    o2 = torch.optim.SGD([p2], lr=lr, momentum=momentum, weight_decay=wd,
+                         nesterov=nesterov)
+    for _ in range(steps):
+        g = torch.randn(10)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        o1.step()
+        o2.step()
+    assert torch.allclose(p1.detach(), p2.detach(), atol=1e-6), \
+        (momentum, wd, nesterov, lr, steps)
