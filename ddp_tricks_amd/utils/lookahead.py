@@ -98,8 +98,19 @@ class Lookahead(Optimizer):
 
     def state_dict(self):
         fast_state_dict = self.optimizer.state_dict()
+        # Same {fast_state, slow_state, param_groups} split as the reference
+        # (utils/lookahead.py:43-55), but slow entries are keyed by PARAM
+        # INDEX (torch's convention) instead of the reference's id(k):
+        # raw Python ids cannot be remapped onto a new process's params, so
+        # the reference's format silently drops the slow weights on load —
+        # harmless there (its training path never loads a Lookahead state,
+        # SURVEY §5.4) but fatal to our --resume extension, where resumed
+        # slow weights must continue the straight-run trajectory exactly
+        # (tests/test_resume.py::test_resume_bitwise_equals_straight_run).
+        index = {p: i for i, p in enumerate(
+            p for g in self.param_groups for p in g["params"])}
         slow_state = {
-            (id(k) if isinstance(k, torch.Tensor) else k): v
+            (index[k] if isinstance(k, torch.Tensor) else k): v
             for k, v in self.state.items()
         }
         return {
@@ -109,16 +120,24 @@ class Lookahead(Optimizer):
         }
 
     def load_state_dict(self, state_dict):
-        slow_state_dict = {
-            "state": state_dict["slow_state"],
-            "param_groups": state_dict["param_groups"],
-        }
         fast_state_dict = {
             "state": state_dict["fast_state"],
             "param_groups": state_dict["param_groups"],
         }
-        super().load_state_dict(slow_state_dict)
         self.optimizer.load_state_dict(fast_state_dict)
+        # the inner load REPLACES its param_groups list — re-share it (and
+        # the per-group counters it restored) through the wrapper
+        self.param_groups = self.optimizer.param_groups
+        for group in self.param_groups:
+            group.setdefault("counter", 0)
+        params = [p for g in self.param_groups for p in g["params"]]
+        self.state = defaultdict(dict)
+        for k, v in state_dict["slow_state"].items():
+            if isinstance(k, int) and 0 <= k < len(params):
+                self.state[params[k]] = v
+            # id(k)-keyed entries (reference-format checkpoints) cannot be
+            # mapped to live params; they are dropped, matching what the
+            # reference's own load achieves
         self.fast_state = self.optimizer.state
 
     def add_param_group(self, param_group):

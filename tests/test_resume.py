@@ -59,3 +59,34 @@ def test_no_resume_flag_ignores_sidecar(tmp_path, monkeypatch, dist_env, capsys)
     amp._state.__init__()
     train(_args(tmp_path, epochs=1, resume=False))
     assert not os.path.exists(os.path.join(tmp_path, "RES.resume.pt"))
+
+
+def test_resume_bitwise_equals_straight_run(tmp_path, monkeypatch, dist_env):
+    """Strong conformance: 2 epochs + resume for 2 more must produce the
+    SAME final weights as 4 straight epochs — every piece of state the
+    sidecar carries (model/momentum/schedulers/early-stop/scaler) and the
+    per-epoch determinism (sampler set_epoch seeding) must line up."""
+    from ddp_tricks_amd.utils.train import train
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "256")
+    monkeypatch.setenv("DDPX_NO_TQDM", "1")
+
+    amp._state.__init__()
+    a = _args(tmp_path, epochs=4, resume=True)
+    a.exp_name = "STRAIGHT"
+    train(a)
+    straight = torch.load(os.path.join(tmp_path, "STRAIGHT.resume.pt"),
+                          weights_only=False)["model"]
+
+    amp._state.__init__()
+    b = _args(tmp_path, epochs=2, resume=True)
+    b.exp_name = "SPLIT"
+    train(b)
+    amp._state.__init__()
+    c = _args(tmp_path, epochs=4, resume=True)
+    c.exp_name = "SPLIT"
+    train(c)
+    split = torch.load(os.path.join(tmp_path, "SPLIT.resume.pt"),
+                       weights_only=False)["model"]
+
+    for k in straight:
+        assert torch.equal(straight[k], split[k]), k
