@@ -90,3 +90,33 @@ def test_validate_tool(tmp_path, monkeypatch):
         capture_output=True, text=True, timeout=300, env=env, cwd=repo)
     assert r.returncode == 0, r.stderr
     assert "valid_acc" in r.stdout
+
+
+@pytest.mark.timeout(300)
+def test_two_runs_same_seed_identical(tmp_path, monkeypatch):
+    """The reference's de-facto verification strategy:
+    same seed, same command => identical trained weights."""
+    import types
+
+    import torch.distributed as dist
+
+    from ddp_tricks_amd.utils.train import train
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "256")
+    monkeypatch.setenv("DDPX_NO_TQDM", "1")
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29656")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    sds = []
+    for run in range(2):
+        amp._state.__init__()
+        a = types.SimpleNamespace(
+            exp_name=f"DET{run}", learning_rate=0.1, batch_size=64, epochs=2,
+            warmup_epochs=1, warmup_type="linear", seed_num=42,
+            data_path="/nonexistent", model_path=str(tmp_path), local_rank=0,
+            resume=True)
+        train(a)
+        sds.append(torch.load(os.path.join(tmp_path, f"DET{run}.resume.pt"),
+                              weights_only=False)["model"])
+    for k in sds[0]:
+        assert torch.equal(sds[0][k], sds[1][k]), k
