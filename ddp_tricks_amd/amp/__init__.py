@@ -61,6 +61,10 @@ class _AmpState:
         self.found_inf = None        # per-step flag, read by the fused
         #                              optimizer kernels (device-side skip)
         self.overflow_count = None   # running count, host-synced per epoch
+        self.first_of_step = None    # 1-based step idx of first overflow in
+        self.last_of_step = None     #   the window / of the last one — lets
+        #                              the window replay reconstruct apex's
+        #                              exact backoff/growth sequencing
         self.async_steps = 0         # backward count since last scaler sync
         self.async_mode = False
 
@@ -73,6 +77,10 @@ def _device_buffers(device):
         _state.found_inf = torch.zeros(1, dtype=torch.float32, device=device)
         _state.overflow_count = torch.zeros(1, dtype=torch.float32,
                                             device=device)
+        _state.first_of_step = torch.full((1,), -1.0, dtype=torch.float32,
+                                          device=device)
+        _state.last_of_step = torch.zeros(1, dtype=torch.float32,
+                                          device=device)
     return _state.found_inf
 
 
@@ -82,20 +90,39 @@ def pending_found_inf():
 
 
 def maybe_sync_scaler() -> None:
-    """Epoch-level host sync of the deferred overflow count (async mode):
-    folds any overflows seen this epoch into the dynamic scaler.  With bf16
-    compute overflows are not expected; this keeps the apex policy live
-    without a per-step device->host sync."""
+    """Window-level host sync of the deferred overflow tracking (async mode):
+    folds overflows seen since the last sync into the dynamic scaler with
+    apex's exact sequencing.  Replay = (first_overflow_step - 1) good
+    updates, then n overflow updates, then (steps - last_overflow_step)
+    good updates.  Good steps *between* overflows are dropped: with the
+    window length capped at growth_interval (scale_loss syncs whenever
+    async_steps reaches it) an inter-overflow streak can never reach the
+    growth threshold, so dropping them cannot change the scaler state —
+    the replay is exactly apex-equivalent.
+
+    Remaining (documented) divergence from apex's synchronous policy: the
+    *unscale multiplier* within the window uses the window-entry scale, and
+    Lookahead's k-counter still advances on device-skipped overflow steps.
+    Both are moot for bf16 compute (same exponent range as fp32: overflow
+    means genuinely non-finite loss)."""
     if not _state.async_mode or _state.overflow_count is None:
         return
     n = int(_state.overflow_count.item())
-    good = max(0, _state.async_steps - n)
-    for _ in range(n):
-        _state.scaler.update(found_inf=True)
-    for _ in range(good):          # keep the x2-after-growth_interval policy
-        _state.scaler.update(found_inf=False)
-    if n > 0:
+    if n == 0:
+        for _ in range(_state.async_steps):
+            _state.scaler.update(found_inf=False)
+    else:
+        first = int(_state.first_of_step.item())
+        last = int(_state.last_of_step.item())
+        for _ in range(max(0, first - 1)):
+            _state.scaler.update(found_inf=False)
+        for _ in range(n):
+            _state.scaler.update(found_inf=True)
+        for _ in range(max(0, _state.async_steps - last)):
+            _state.scaler.update(found_inf=False)
         _state.overflow_count.zero_()
+        _state.first_of_step.fill_(-1.0)
+        _state.last_of_step.zero_()
     _state.async_steps = 0
 
 
@@ -223,6 +250,16 @@ def scale_loss(loss, optimizer, delay_unscale: bool = False):
             fi.zero_()  # stream-ordered: prior step's kernels already read it
             ext.multi_tensor_unscale(tensors, fi, inv)
             _state.overflow_count += fi
+            # device-side (no host sync) record of first/last overflow step
+            of = fi > 0
+            step = float(_state.async_steps)
+            _state.first_of_step.masked_fill_(
+                of & (_state.first_of_step < 0), step)
+            _state.last_of_step.masked_fill_(of, step)
+            # cap the window at growth_interval so the replay in
+            # maybe_sync_scaler stays exactly apex-equivalent
+            if _state.async_steps >= _state.scaler.growth_interval:
+                maybe_sync_scaler()
             return
 
     found_inf = _unscale_and_check(tensors, inv)
@@ -258,9 +295,10 @@ def _register_cast(mod, name: str, dtype) -> None:
 
     def wrapped(*args, **kwargs):
         if _state.enabled:
-            args = tuple(a.to(dtype)
-                         if torch.is_tensor(a) and a.is_floating_point()
-                         else a for a in args)
+            cast = (lambda a: a.to(dtype)
+                    if torch.is_tensor(a) and a.is_floating_point() else a)
+            args = tuple(cast(a) for a in args)
+            kwargs = {k: cast(v) for k, v in kwargs.items()}  # apex casts kwargs too
         return fn(*args, **kwargs)
 
     wrapped._amp_registered = True

@@ -84,11 +84,35 @@ class DistributedDataParallel(nn.Module):
             from ..ops import load_extension
             ext = load_extension()
             if ext is not None and hasattr(ext, "rccl_unique_id"):
-                obj = [ext.rccl_unique_id() if self.rank == 0 else None]
-                dist.broadcast_object_list(obj, src=0, group=self.process_group)
-                self._rccl = ext.rccl_comm_init(self.world_size, self.rank,
-                                                obj[0])
-                self._ext = ext
+                # Defensive: if RCCL refuses the topology (e.g. two ranks
+                # sharing one device in a test rig), fall back to the torch
+                # process-group path rather than dying — every rank must
+                # take the SAME branch, so the failure flag is all-reduced
+                # before committing to the native path.
+                err = None
+                try:
+                    obj = [ext.rccl_unique_id() if self.rank == 0 else None]
+                    dist.broadcast_object_list(obj, src=0,
+                                               group=self.process_group)
+                    rccl = ext.rccl_comm_init(self.world_size, self.rank,
+                                              obj[0])
+                except Exception as e:  # noqa: BLE001 — any init failure
+                    err = e
+                    rccl = None
+                flags = [None] * self.world_size
+                dist.all_gather_object(flags, 0 if err is None else 1,
+                                       group=self.process_group)
+                if sum(flags) == 0:
+                    self._rccl = rccl
+                    self._ext = ext
+                else:
+                    if rccl is not None:
+                        ext.rccl_comm_destroy(rccl)
+                    if self.rank == 0:
+                        import warnings
+                        warnings.warn(
+                            f"native RCCL init failed ({err}); falling back "
+                            "to the torch process-group collectives")
 
         if self.world_size > 1:
             self._broadcast_params()

@@ -304,6 +304,14 @@ class FastBatchLoader:
                 yield (torch.index_select(images, 0, sel),
                        torch.index_select(labels, 0, sel))
             return
+        if not self.pin:
+            # pin_memory=False honored: plain pageable H2D,
+            # no staging buffers, synchronous copy semantics.
+            for s0 in range(0, idx.numel(), B):
+                sel = idx[s0:s0 + B]
+                yield (torch.index_select(images, 0, sel).to(self.device),
+                       torch.index_select(labels, 0, sel).to(self.device))
+            return
         # GPU path: own the pinned staging AND the H2D (one-ahead on a copy
         # stream).  Slot reuse is guarded by a host event-sync: the host can
         # run many steps ahead of the device, so "old enough" is not enough.

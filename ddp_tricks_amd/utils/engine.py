@@ -43,8 +43,13 @@ def iterate_loader(
 
     if timers is None and os.environ.get("DDPX_PHASE_TIMERS", "0") == "1":
         from .timers import PhaseTimers
-        timers = iterate_loader._timers = getattr(
-            iterate_loader, "_timers", None) or PhaseTimers(device)
+        # cache keyed by device: events/streams are device-bound
+        cache = getattr(iterate_loader, "_timers", None)
+        if cache is None:
+            cache = iterate_loader._timers = {}
+        timers = cache.get(device)
+        if timers is None:
+            timers = cache[device] = PhaseTimers(device)
     from contextlib import nullcontext
     ph = timers.phase if timers is not None else (lambda name: nullcontext())
 

