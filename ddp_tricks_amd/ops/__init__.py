@@ -54,6 +54,11 @@ class _SyncDebugProxy:
 def load_extension(required: bool = False):
     """Load the in-tree _hip_ops extension; cache the module object."""
     global _EXT, _EXT_TRIED
+    if os.environ.get("DDPX_FORCE_TORCH", "0") == "1":
+        # A/B benchmarking: pretend the HIP extension is absent so the ops
+        # take the torch-ROCm (MIOpen/hipBLASLt) bring-up path.  Pair with
+        # DDPX_ALLOW_TORCH_FALLBACK=1 (VERDICT r01 next-round #6).
+        return None
     if _EXT is not None:
         return _EXT
     if _EXT_TRIED and not required:
@@ -83,7 +88,8 @@ def have_extension() -> bool:
 
 
 def allow_torch_fallback() -> bool:
-    return os.environ.get("DDPX_ALLOW_TORCH_FALLBACK", "0") == "1"
+    return (os.environ.get("DDPX_ALLOW_TORCH_FALLBACK", "0") == "1"
+            or os.environ.get("DDPX_FORCE_TORCH", "0") == "1")
 
 
 def require_ext_for(tensor):

@@ -46,6 +46,45 @@ def _find_idx(root: str, base: str) -> Optional[str]:
     return None
 
 
+_MNIST_MIRRORS = (
+    "https://ossci-datasets.s3.amazonaws.com/mnist/",
+    "http://yann.lecun.com/exdb/mnist/",
+)
+
+
+def _download_mnist(root: str, bases) -> None:
+    """Best-effort fetch of the raw IDX .gz files (reference parity:
+    torchvision download=True at reference utils/train.py:25).  In the
+    offline build/test environment this fails fast and the caller falls
+    back to local files / synthetic data."""
+    import urllib.error
+    import urllib.request
+    dest = os.path.join(root, "MNIST", "raw")
+    os.makedirs(dest, exist_ok=True)
+    for base in bases:
+        fname = base + ".gz"
+        out = os.path.join(dest, fname)
+        if os.path.exists(out) or os.path.exists(out[:-3]):
+            continue
+        for mirror in _MNIST_MIRRORS:
+            try:
+                with urllib.request.urlopen(mirror + fname, timeout=15) as r, \
+                        open(out + ".part", "wb") as f:
+                    while True:
+                        chunk = r.read(1 << 20)
+                        if not chunk:
+                            break
+                        f.write(chunk)
+                os.replace(out + ".part", out)
+                break
+            except (urllib.error.URLError, OSError, ValueError):
+                continue
+        else:
+            import warnings
+            warnings.warn(f"MNIST download failed for {fname} "
+                          "(no network?); falling back to local/synthetic")
+
+
 class MNIST(Dataset):
     """MNIST (or a deterministic synthetic stand-in when files are absent).
 
@@ -60,6 +99,10 @@ class MNIST(Dataset):
         base_lbl = "train-labels-idx1-ubyte" if train else "t10k-labels-idx1-ubyte"
         img_path = _find_idx(root, base_img)
         lbl_path = _find_idx(root, base_lbl)
+        if download and not (img_path and lbl_path) and synthetic is not True:
+            _download_mnist(root, (base_img, base_lbl))
+            img_path = _find_idx(root, base_img)
+            lbl_path = _find_idx(root, base_lbl)
         n_default = 60000 if train else 10000
         self.synthetic = synthetic if synthetic is not None else not (img_path and lbl_path)
         if not self.synthetic:
