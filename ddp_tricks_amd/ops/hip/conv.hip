@@ -151,62 +151,87 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
         }
     }
 
-    auto issue_tile = [&](int kt, int buf) {
-        // ---- A tile: chunks round-robined over the 4 waves ----
-        #pragma unroll
-        for (int ci = 0; ci < A_PER; ++ci) {
-            const int ch = wid + ci * 4;
-            int row = ch * 8 + pl_row;
-            int seg = pl_segp ^ (row & 7);     // logical k-segment
-            int gk = kt + seg * 8;
-            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
-            if (ri_u[ci] != INT_MIN && gk < Kgemm) {
-                if (MODE == 0) {
-                    unsigned rs = fd_div(gk, cs.fdC);
-                    int c = fd_mod(gk, cs.fdC, rs);
-                    int r = fd_div(rs, cs.fdS);
-                    int sx = fd_mod(rs, cs.fdS, r);
-                    int h = ri_u[ci] + r;
-                    int wcol = ri_v[ci] + sx;
-                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                        src = &Asrc[((ri_nb[ci] + h) * cs.W + wcol) * cs.C + c];
-                } else if (MODE == 2) {
-                    unsigned rs2 = fd_div(gk, cs.fdKo);
-                    int ko = fd_mod(gk, cs.fdKo, rs2);
-                    int ti = fd_div(rs2, cs.fdNs);
-                    int si = fd_mod(rs2, cs.fdNs, ti);
-                    int p = ri_u[ci] + cs.off_r[ti];
-                    int q = ri_v[ci] + cs.off_s[si];
-                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
-                } else if (STRIDE1) {
-                    unsigned rs = fd_div(gk, cs.fdKo);
-                    int ko = fd_mod(gk, cs.fdKo, rs);
-                    int r = fd_div(rs, cs.fdS);
-                    int sx = fd_mod(rs, cs.fdS, r);
-                    int p = ri_u[ci] - r;
-                    int q = ri_v[ci] - sx;
-                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
-                } else {
-                    unsigned rs = fd_div(gk, cs.fdKo);
-                    int ko = fd_mod(gk, cs.fdKo, rs);
-                    int r = fd_div(rs, cs.fdS);
-                    int sx = fd_mod(rs, cs.fdS, r);
-                    int pn = ri_u[ci] - r;
-                    int qn = ri_v[ci] - sx;
-                    int p = pn / cs.stride, q = qn / cs.stride;
-                    if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
-                        qn == q * cs.stride && p < cs.P && q < cs.Q)
-                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
-                }
-            }
-            __builtin_amdgcn_global_load_lds(
-                (const __attribute__((address_space(1))) unsigned short*)src,
-                (__attribute__((address_space(3))) unsigned short*)&lds_a[buf][ch * 8][0],
-                16, 0, 0);
+    // Seg-constant k decode (round-2): a thread's logical k-segment
+    // seg = pl_segp ^ pl_row is FIXED (row & 7 == pl_row for every chunk),
+    // so its gk = kt + seg*8 advances by exactly CBK per tile.  One FastDiv
+    // decode at gk0 seeds (tap, fastdim) coordinates; afterwards they
+    // advance with constant increments (every supported shape has the fast
+    // dim — C or Ko, a multiple of 8 — dividing CBK or divided by it).
+    // Odd shapes fall back to the per-tile FastDiv decode.
+    const int kfast = (MODE == 0) ? cs.C : cs.Ko;          // fastest k dim
+    const int ksec = (MODE == 2) ? cs.ns : cs.S;           // second k dim
+    const int gk0 = (pl_segp ^ pl_row) * 8;
+    int kc_r, kc_s, kc_c;                                   // decoded coords
+    {
+        unsigned rs = (MODE == 0) ? fd_div(gk0, cs.fdC) : fd_div(gk0, cs.fdKo);
+        kc_c = (MODE == 0) ? fd_mod(gk0, cs.fdC, rs) : fd_mod(gk0, cs.fdKo, rs);
+        if (MODE == 2) {
+            kc_r = fd_div(rs, cs.fdNs);
+            kc_s = fd_mod(rs, cs.fdNs, kc_r);
+        } else {
+            kc_r = fd_div(rs, cs.fdS);
+            kc_s = fd_mod(rs, cs.fdS, kc_r);
         }
-        // ---- B tile (row-contiguous source) ----
+    }
+    const bool kinc_big = (kfast % CBK) == 0;       // c walks, rare carries
+    const int kdrs = (kfast < CBK && (CBK % kfast) == 0) ? CBK / kfast : 0;
+    auto k_advance = [&](int gk_next) {
+        if (kinc_big) {
+            kc_c += CBK;
+            if (kc_c >= kfast) {
+                kc_c = 0;
+                if (++kc_s >= ksec) { kc_s = 0; ++kc_r; }
+            }
+        } else if (kdrs) {                           // c fixed, taps walk
+            kc_s += kdrs;
+            while (kc_s >= ksec) { kc_s -= ksec; ++kc_r; }
+        } else {
+            unsigned rs = (MODE == 0) ? fd_div(gk_next, cs.fdC)
+                                      : fd_div(gk_next, cs.fdKo);
+            kc_c = (MODE == 0) ? fd_mod(gk_next, cs.fdC, rs)
+                               : fd_mod(gk_next, cs.fdKo, rs);
+            if (MODE == 2) {
+                kc_r = fd_div(rs, cs.fdNs);
+                kc_s = fd_mod(rs, cs.fdNs, kc_r);
+            } else {
+                kc_r = fd_div(rs, cs.fdS);
+                kc_s = fd_mod(rs, cs.fdS, kc_r);
+            }
+        }
+    };
+
+    // A-side gather address for chunk ci at decoded coords (kc_r=tap/ti,
+    // kc_s=tap2/si, kc_c=c/ko); out-of-range lanes read the zero page.
+    auto a_addr = [&](int ci, int gk) -> const bf16* {
+        const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
+        if (ri_u[ci] == INT_MIN || gk >= Kgemm) return src;
+        if (MODE == 0) {
+            int h = ri_u[ci] + kc_r;
+            int wcol = ri_v[ci] + kc_s;
+            if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                src = &Asrc[((ri_nb[ci] + h) * cs.W + wcol) * cs.C + kc_c];
+        } else if (MODE == 2) {
+            int p = ri_u[ci] + cs.off_r[kc_r];
+            int q = ri_v[ci] + cs.off_s[kc_s];
+            if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
+        } else if (STRIDE1) {
+            int p = ri_u[ci] - kc_r;
+            int q = ri_v[ci] - kc_s;
+            if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
+        } else {
+            int pn = ri_u[ci] - kc_r;
+            int qn = ri_v[ci] - kc_s;
+            int p = pn / cs.stride, q = qn / cs.stride;
+            if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
+                qn == q * cs.stride && p < cs.P && q < cs.Q)
+                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
+        }
+        return src;
+    };
+
+    auto issue_b_tile = [&](int kt) {
         for (int ch = wid; ch < B_CHUNKS; ch += 4) {
             int row = ch * 8 + pl_row;
             int seg = pl_segp ^ (row & 7);
@@ -227,15 +252,44 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                 src = &Bsrc[(long)gn * Kgemm + gk];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) unsigned short*)src,
-                (__attribute__((address_space(3))) unsigned short*)&lds_b[buf][ch * 8][0],
+                (__attribute__((address_space(3))) unsigned short*)&lds_b[0][ch * 8][0],
                 16, 0, 0);
         }
     };
 
+    // Prologue: stage tile 0 — A via glds (no VGPR round trip), B via glds.
+    #pragma unroll
+    for (int ci = 0; ci < A_PER; ++ci) {
+        const int ch = wid + ci * 4;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned short*)
+                a_addr(ci, gk0),
+            (__attribute__((address_space(3))) unsigned short*)&lds_a[0][ch * 8][0],
+            16, 0, 0);
+    }
+    issue_b_tile(0);
+
+    // k-loop with register-staged A prefetch (round-2): the NEXT tile's A
+    // gathers are issued as plain b128 loads right after the barrier, so
+    // their HBM latency hides under this tile's MFMA block; after the
+    // tiles are dead they are ds_written to the SAME swizzled LDS image
+    // the glds path produced (+16 VGPR, LDS unchanged — stays at 3
+    // waves/SIMD; the 2-buffer LDS ring alternative was measured 10-30%
+    // slower, see note above).
     constexpr int buf = 0;
+    int gk_cur = gk0;
     for (int kt = 0; kt < Kgemm; kt += CBK) {
-        issue_tile(kt, 0);
-        __syncthreads();   // vmcnt(0) for the in-flight glds + barrier
+        const bool has_next = kt + CBK < Kgemm;
+        __syncthreads();   // tile kt staged (glds vmcnt / ds_write lgkmcnt)
+        bf16x8_t pre[A_PER];
+        if (has_next) {
+            gk_cur += CBK;
+            k_advance(gk_cur);
+            #pragma unroll
+            for (int ci = 0; ci < A_PER; ++ci)
+                pre[ci] = *reinterpret_cast<const bf16x8_t*>(
+                    a_addr(ci, gk_cur));
+        }
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
             bf16x8_t af[MI], bfr[NI];
@@ -259,7 +313,16 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
         }
-        __syncthreads();
+        __syncthreads();   // every wave done reading tile kt
+        if (has_next) {
+            #pragma unroll
+            for (int ci = 0; ci < A_PER; ++ci) {
+                const int ch = wid + ci * 4;
+                *reinterpret_cast<bf16x8_t*>(
+                    &lds_a[0][ch * 8 + pl_row][pl_segp * 8]) = pre[ci];
+            }
+            issue_b_tile(kt + CBK);
+        }
     }
 
     float psum[NI] = {}, psq[NI] = {};
