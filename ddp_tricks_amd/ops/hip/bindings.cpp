@@ -64,7 +64,8 @@ std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
                                          c10::optional<at::Tensor> bias,
                                          long stride, long pad);
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C, long H,
-                        long W, long R, long S, long stride, long pad);
+                        long W, long R, long S, long stride, long pad,
+                        c10::optional<at::Tensor> addend);
 std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor dy, at::Tensor wt2,
                                         long N, long C, long H, long W,
                                         long R, long S, long pad,
@@ -131,7 +132,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_fwd_stats", &conv2d_fwd_stats, py::arg("x"), py::arg("w"),
           py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
           py::arg("pad") = 0);
-    m.def("conv2d_dgrad", &conv2d_dgrad);
+    m.def("conv2d_dgrad", &conv2d_dgrad,
+          py::arg("dy"), py::arg("wt2"), py::arg("N"), py::arg("C"),
+          py::arg("H"), py::arg("W"), py::arg("R"), py::arg("S"),
+          py::arg("stride"), py::arg("pad"),
+          py::arg("addend") = c10::nullopt);
     m.def("conv2d_dgrad_bn", &conv2d_dgrad_bn);
     m.def("conv2d_wgrad", &conv2d_wgrad);
 }

@@ -350,7 +350,172 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
     }
 }
 
+// ----------------------------------------------- single-pass backward ------
+//
+// DDPX_BN1PASS=1: one persistent-grid kernel
+// replaces partial+combine+dx.  Each workgroup computes its split's
+// partial sums (phase 1, identical row striding and arithmetic to
+// k_bn_bwd_partial), takes a ticket; the LAST workgroup combines the slab
+// exactly like k_bn_bwd_combine (same lane/stride order — bitwise-equal
+// coefficients), publishes them with a release flag; everyone else spins
+// (s_sleep + acquire load), then runs the dx sweep over the SAME rows it
+// just read — x/dy/mask stay hot in cache, saving one full read of each.
+// The grid is sized to guaranteed-resident blocks
+// (hipOccupancyMaxActiveBlocksPerMultiprocessor × CUs) so the ticket
+// rendezvous cannot deadlock; the spin also carries a bail-out bound so a
+// protocol bug degrades to wrong numbers (caught by the numerics tests),
+// never a hung device.
+
+__global__ __launch_bounds__(256)
+void k_bn_bwd_onepass(const bf16* __restrict__ x,
+                      const bf16* __restrict__ dy,
+                      const unsigned char* __restrict__ mask,
+                      const float* __restrict__ save_mean,
+                      const float* __restrict__ save_invstd,
+                      const float* __restrict__ gamma,
+                      long M, int C, int S, bool relu,
+                      float* __restrict__ slab,
+                      float* __restrict__ dgamma,
+                      float* __restrict__ dbeta,
+                      float* __restrict__ coef_a,
+                      float* __restrict__ coef_b,
+                      float* __restrict__ coef_c,
+                      int* __restrict__ ticket,
+                      int* __restrict__ flag,
+                      bf16* __restrict__ dx,
+                      bf16* __restrict__ dresid) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* red = reinterpret_cast<float*>(smem);
+    const int cpg = C >> 3;
+    const int c8 = threadIdx.x % cpg;
+    const int walker = threadIdx.x / cpg;
+    const int nw = blockDim.x / cpg;
+    const int s = blockIdx.x;
+    float mean[8], invstd[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        mean[j] = save_mean[c8 * 8 + j];
+        invstd[j] = save_invstd[c8 * 8 + j];
+    }
+    // ---- phase 1: partial sums over this split's rows (== k_bn_bwd_partial)
+    float sum_dy[8] = {}, sum_dyx[8] = {};
+    const long rstep = (long)S * nw;
+    const long r0 = (long)s * nw + walker;
+    for (long r = r0; r < M; r += rstep) {
+        s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        unsigned m = relu ? mask[r * cpg + c8] : 0xffu;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = (m >> j) & 1u ? us2f((unsigned short)vg[j]) : 0.f;
+            float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
+            sum_dy[j] += g;
+            sum_dyx[j] = fmaf(g, xh, sum_dyx[j]);
+        }
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red[walker * C + c8 * 8 + j] = sum_dy[j];
+        red[nw * C + walker * C + c8 * 8 + j] = sum_dyx[j];
+    }
+    __syncthreads();
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float a = 0.f, b = 0.f;
+        for (int w = 0; w < nw; ++w) {
+            a += red[w * C + c];
+            b += red[nw * C + w * C + c];
+        }
+        slab[(long)c * S + s] = a;
+        slab[(long)C * S + (long)c * S + s] = b;
+    }
+    // ---- ticket rendezvous
+    __shared__ int is_last;
+    __threadfence();                       // slab visible device-wide
+    if (threadIdx.x == 0)
+        is_last = (atomicAdd(ticket, 1) == gridDim.x - 1);
+    __syncthreads();
+    if (is_last) {
+        // combine exactly like k_bn_bwd_combine: wave per channel, lanes
+        // stride the splits, shfl tree — bitwise-equal to the 2-pass path
+        const int lane = threadIdx.x & 63;
+        for (int c = (int)(threadIdx.x >> 6); c < C; c += (int)(blockDim.x >> 6)) {
+            float a = 0.f, b = 0.f;
+            for (int sp = lane; sp < S; sp += 64) {
+                a += slab[(long)c * S + sp];
+                b += slab[(long)C * S + (long)c * S + sp];
+            }
+            a = wave_reduce_sum(a);
+            b = wave_reduce_sum(b);
+            if (lane == 0) {
+                dgamma[c] = b;
+                dbeta[c] = a;
+                float g = gamma ? gamma[c] : 1.f;
+                coef_a[c] = g * save_invstd[c];
+                coef_b[c] = a / M;
+                coef_c[c] = b / M;
+            }
+        }
+        __syncthreads();
+        __threadfence();
+        if (threadIdx.x == 0)
+            __hip_atomic_store(flag, 1, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_AGENT);
+    }
+    // ---- spin for the coefficients (bounded: bail-out, never a hang)
+    if (threadIdx.x == 0) {
+        long spins = 0;
+        while (__hip_atomic_load(flag, __ATOMIC_ACQUIRE,
+                                 __HIP_MEMORY_SCOPE_AGENT) == 0) {
+            __builtin_amdgcn_s_sleep(8);
+            if (++spins > (1l << 28)) break;   // ~seconds; wrong > hung
+        }
+    }
+    __syncthreads();
+    // ---- phase 2: dx over the SAME rows (cache-hot x/dy/mask)
+    float ca[8], cb[8], cc[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        int c = c8 * 8 + j;
+        ca[j] = coef_a[c];
+        cb[j] = coef_b[c];
+        cc[j] = coef_c[c];
+    }
+    for (long r = r0; r < M; r += rstep) {
+        s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        unsigned m = relu ? mask[r * cpg + c8] : 0xffu;
+        s16x8 o, og;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = (m >> j) & 1u ? us2f((unsigned short)vg[j]) : 0.f;
+            float xh = (us2f((unsigned short)vx[j]) - mean[j]) * invstd[j];
+            o[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
+            og[j] = (short)f2us(g);
+        }
+        reinterpret_cast<s16x8*>(dx + r * C)[c8] = o;
+        if (dresid)
+            reinterpret_cast<s16x8*>(dresid + r * C)[c8] = og;
+    }
+}
+
 // ------------------------------------------------------------------ hosts ---
+
+// Guaranteed-resident grid for the one-pass rendezvous: blocks that are
+// not simultaneously resident would deadlock the ticket protocol, so the
+// grid is CUs × occupancy-per-CU, never more.
+static int resident_grid(const void* kfunc, int block, int lds) {
+    static int numCU = 0;
+    if (!numCU) {
+        hipDeviceProp_t p;
+        if (hipGetDeviceProperties(&p, 0) != hipSuccess) return 0;
+        numCU = p.multiProcessorCount;
+    }
+    int per = 0;
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &per, kfunc, block, lds) != hipSuccess || per < 1)
+        return 0;
+    return numCU * per;
+}
 
 static void shape_mc(const at::Tensor& x, long& M, int& C) {
     if (x.dim() == 4) {        // NCHW logical, channels_last physical
@@ -501,6 +666,40 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
     const unsigned char* yp = fuse_relu
         ? mask.data_ptr<unsigned char>() : nullptr;
     int lds = 2 * nw * C * 4;
+
+    const char* e1p = getenv("DDPX_BN1PASS");
+    if (e1p && e1p[0] == '1' && !pre_slab.has_value()) {
+        int grid = resident_grid((const void*)k_bn_bwd_onepass, block, lds);
+        if (grid > 0) {
+            int S1 = grid;
+            auto slab1 = at::empty({2, C, S1}, fopts);
+            auto work = at::zeros({2}, gamma.options().dtype(at::kInt));
+            at::Tensor dresid1;
+            bf16* drp1 = nullptr;
+            if (want_dresid) {
+                dresid1 = x.dim() == 4
+                    ? at::empty_like(x, x.options().memory_format(
+                          at::MemoryFormat::ChannelsLast))
+                    : at::empty_like(x);
+                drp1 = reinterpret_cast<bf16*>(dresid1.data_ptr());
+            }
+            hipLaunchKernelGGL(k_bn_bwd_onepass, dim3(S1), dim3(block), lds,
+                               stream.stream(), xp, dyp, yp,
+                               save_mean.data_ptr<float>(),
+                               save_invstd.data_ptr<float>(),
+                               gamma.data_ptr<float>(), M, C, S1, fuse_relu,
+                               slab1.data_ptr<float>(),
+                               dgamma.data_ptr<float>(),
+                               dbeta.data_ptr<float>(),
+                               ca.data_ptr<float>(), cb.data_ptr<float>(),
+                               cc.data_ptr<float>(),
+                               work.data_ptr<int>(), work.data_ptr<int>() + 1,
+                               reinterpret_cast<bf16*>(dx.data_ptr()), drp1);
+            HIP_CHECK_LAST();
+            if (want_dresid) return {dx, dgamma, dbeta, dresid1};
+            return {dx, dgamma, dbeta};
+        }
+    }
 
     if (!pre_slab.has_value()) {
         hipLaunchKernelGGL(k_bn_bwd_partial, dim3(S), dim3(block), lds,

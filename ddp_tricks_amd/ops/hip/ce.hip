@@ -9,6 +9,7 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <climits>
 #include "common.h"
 
 constexpr int CE_BLOCK = 256;
@@ -30,6 +31,121 @@ __global__ void k_ce_rows(const T* __restrict__ logits,
     float lse = logf(s) + mx;
     lse_out[i] = lse;
     rowloss[i] = lse - (float)row[target[i]];
+}
+
+// Wave-per-row variants for wide rows (the ResNet-50 [B,1000] head —
+// the thread-per-row kernel walks 3×C serially,
+// 254 µs for ~4 MB of reads).  One 64-lane wave owns a row; lanes stream
+// 16 B bf16x8 chunks (coalesced 1 KiB per wave-iteration) with an online
+// softmax (single pass over the row), then a shfl tree combines the
+// per-lane (max, scaled-sum) pairs.  bf16 rows with C % 8 == 0 only;
+// everything else keeps the scalar kernels.
+constexpr int CE_WPB = CE_BLOCK / WAVE;   // rows (waves) per block
+
+DEV_INLINE void ce_online8(const s16x8 v, float& m, float& s) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        float x = us2f((unsigned short)v[j]);
+        if (x > m) {
+            s *= expf(m - x);   // m == -inf ⇒ s == 0, exp(-inf)=0 handled
+            m = x;
+        }
+        s += expf(x - m);
+    }
+}
+
+__global__ void k_ce_rows_wave(const bf16* __restrict__ logits,
+                               const long* __restrict__ target,
+                               float* __restrict__ rowloss,
+                               float* __restrict__ lse_out,
+                               int B, int C) {
+    const int row_i = blockIdx.x * CE_WPB + ((int)threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (row_i >= B) return;
+    const bf16* row = logits + (long)row_i * C;
+    const int chunks = C >> 3;
+    float m = -INFINITY, s = 0.f;
+    for (int ch = lane; ch < chunks; ch += WAVE) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(row + ch * 8);
+        ce_online8(v, m, s);
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        float mo = __shfl_down(m, off, 64);
+        float so = __shfl_down(s, off, 64);
+        float mn = fmaxf(m, mo);
+        float t1 = (s == 0.f) ? 0.f : s * expf(m - mn);
+        float t2 = (so == 0.f) ? 0.f : so * expf(mo - mn);
+        m = mn;
+        s = t1 + t2;
+    }
+    if (lane == 0) {
+        float lse = logf(s) + m;
+        lse_out[row_i] = lse;
+        rowloss[row_i] = lse - (float)row[target[row_i]];
+    }
+}
+
+__global__ void k_ce_bwd_wave(const bf16* __restrict__ logits,
+                              const long* __restrict__ target,
+                              const float* __restrict__ lse,
+                              const float* __restrict__ dloss,
+                              bf16* __restrict__ dlogits, int B, int C) {
+    const int row_i = blockIdx.x * CE_WPB + ((int)threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (row_i >= B) return;
+    const float g = dloss[0] / B;
+    const bf16* row = logits + (long)row_i * C;
+    bf16* drow = dlogits + (long)row_i * C;
+    const float l = lse[row_i];
+    const int t = (int)target[row_i];
+    const int chunks = C >> 3;
+    for (int ch = lane; ch < chunks; ch += WAVE) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(row + ch * 8);
+        s16x8 o;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            int c = ch * 8 + j;
+            float p = expf(us2f((unsigned short)v[j]) - l);
+            o[j] = (short)f2us((p - (c == t ? 1.f : 0.f)) * g);
+        }
+        *reinterpret_cast<s16x8*>(drow + ch * 8) = o;
+    }
+}
+
+__global__ void k_argmax_correct_wave(const bf16* __restrict__ logits,
+                                      const long* __restrict__ target,
+                                      long long* __restrict__ out,
+                                      int B, int C) {
+    const int row_i = blockIdx.x * CE_WPB + ((int)threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (row_i >= B) return;
+    const bf16* row = logits + (long)row_i * C;
+    const int chunks = C >> 3;
+    float best = -INFINITY;
+    int arg = INT_MAX;
+    for (int ch = lane; ch < chunks; ch += WAVE) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(row + ch * 8);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float x = us2f((unsigned short)v[j]);
+            if (x > best) { best = x; arg = ch * 8 + j; }
+        }
+    }
+    // combine with torch's first-max tie rule: larger value wins, equal
+    // value ⇒ smaller index wins
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        float bo = __shfl_down(best, off, 64);
+        int ao = __shfl_down(arg, off, 64);
+        if (bo > best || (bo == best && ao < arg)) { best = bo; arg = ao; }
+    }
+    if (lane == 0 && arg == (int)target[row_i])
+        atomicAdd(reinterpret_cast<unsigned long long*>(out), 1ull);
+}
+
+static inline bool ce_wave_ok(const at::Tensor& logits, int C) {
+    return logits.scalar_type() == at::kBFloat16 && (C % 8) == 0 && C >= 64;
 }
 
 // deterministic mean: single block, fixed-order tree over per-thread partials
@@ -57,7 +173,13 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) {
     auto loss = at::empty({}, opts);
     auto stream = at::hip::getCurrentHIPStream();
     int blocks = ceil_div_i(B, CE_BLOCK);
-    if (logits.scalar_type() == at::kBFloat16) {
+    if (ce_wave_ok(logits, C)) {
+        hipLaunchKernelGGL(k_ce_rows_wave, dim3(ceil_div_i(B, CE_WPB)),
+                           dim3(CE_BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const bf16*>(logits.data_ptr()),
+                           target.data_ptr<long>(), rowloss.data_ptr<float>(),
+                           lse.data_ptr<float>(), B, C);
+    } else if (logits.scalar_type() == at::kBFloat16) {
         hipLaunchKernelGGL(k_ce_rows<bf16>, dim3(blocks), dim3(CE_BLOCK), 0,
                            stream.stream(),
                            reinterpret_cast<const bf16*>(logits.data_ptr()),
@@ -105,7 +227,14 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
     auto stream = at::hip::getCurrentHIPStream();
     int blocks = ceil_div_i(B, CE_BLOCK);
     auto dlossf = dloss.scalar_type() == at::kFloat ? dloss : dloss.to(at::kFloat);
-    if (logits.scalar_type() == at::kBFloat16) {
+    if (ce_wave_ok(logits, C)) {
+        hipLaunchKernelGGL(k_ce_bwd_wave, dim3(ceil_div_i(B, CE_WPB)),
+                           dim3(CE_BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const bf16*>(logits.data_ptr()),
+                           target.data_ptr<long>(), lse.data_ptr<float>(),
+                           dlossf.data_ptr<float>(),
+                           reinterpret_cast<bf16*>(dlogits.data_ptr()), B, C);
+    } else if (logits.scalar_type() == at::kBFloat16) {
         hipLaunchKernelGGL(k_ce_bwd<bf16>, dim3(blocks), dim3(CE_BLOCK), 0,
                            stream.stream(),
                            reinterpret_cast<const bf16*>(logits.data_ptr()),
@@ -157,7 +286,13 @@ at::Tensor argmax_correct(at::Tensor logits, at::Tensor target) {
     auto out = at::zeros({}, lc.options().dtype(at::kLong));
     auto stream = at::hip::getCurrentHIPStream();
     int blocks = ceil_div_i(B, CE_BLOCK);
-    if (lc.scalar_type() == at::kBFloat16) {
+    if (ce_wave_ok(lc, C)) {
+        hipLaunchKernelGGL(k_argmax_correct_wave, dim3(ceil_div_i(B, CE_WPB)),
+                           dim3(CE_BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const bf16*>(lc.data_ptr()),
+                           target.data_ptr<long>(),
+                           reinterpret_cast<long long*>(out.data_ptr()), B, C);
+    } else if (lc.scalar_type() == at::kBFloat16) {
         hipLaunchKernelGGL(k_argmax_correct<bf16>, dim3(blocks), dim3(CE_BLOCK),
                            0, stream.stream(),
                            reinterpret_cast<const bf16*>(lc.data_ptr()),

@@ -75,14 +75,22 @@ struct BnFuse {
     float* slab;                    // [2][C][gridM]
 };
 
-template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
-          int CBM_T = CBM>
-__global__ __launch_bounds__(256)
-void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
-                 const float* __restrict__ bias, bf16* __restrict__ out,
-                 ConvShape cs, int M, int Kgemm, int Nout,
-                 float* __restrict__ stats = nullptr,
-                 BnFuse bn = BnFuse{}) {
+// The GEMM body lives in a __device__ function so two launchers can share
+// it: k_conv_gemm (one ConvShape per launch) and k_conv_gemm_cls (MODE-2
+// strided dgrad with blockIdx.z = residue class, one launch for ALL
+// classes — VERDICT r01 next-round #7).  ``addin`` is an optional second
+// operand streamed into the output epilogue (dx += addin): the ResNet
+// skip-gradient accumulation fused into the junction conv's dgrad
+// (VERDICT r01 next-round #4).
+template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1,
+          int CBM_T>
+__device__ __forceinline__
+void conv_gemm_body(const bf16* __restrict__ Asrc,
+                    const bf16* __restrict__ Bsrc,
+                    const float* __restrict__ bias, bf16* __restrict__ out,
+                    const ConvShape& cs, int M, int Kgemm, int Nout,
+                    float* __restrict__ stats, BnFuse bn,
+                    const bf16* __restrict__ addin) {
     constexpr int WM = CBM_T / WAVES_M;
     constexpr int WN = TBN / WAVES_N;
     constexpr int MI = WM / 16;
@@ -409,6 +417,18 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                 int i = rl * TBN + csub;    // 16 B chunk within one half
                 const bf16* src16 = i < HALF ? &stageA[i] : &stageB[i - HALF];
                 bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(src16);
+                if (MODE == 1 && addin) {
+                    // fused skip-grad accumulation: dx += addin (the
+                    // junction's dresid), one inline read instead of a
+                    // separate ATen add pass over dx+addin+out
+                    s16x8 a8 = *reinterpret_cast<const s16x8*>(
+                        &addin[row * Nout + col]);
+                    s16x8& vv = reinterpret_cast<s16x8&>(v);
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        vv[j] = (short)f2us(us2f((unsigned short)vv[j])
+                                            + us2f((unsigned short)a8[j]));
+                }
                 *reinterpret_cast<bf16x8_t*>(&out[row * Nout + col]) = v;
                 if (do_bn) {
                     unsigned m8 = bn.mask
@@ -490,6 +510,42 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
             }
         }
     }
+}
+
+template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
+          int CBM_T = CBM>
+__global__ __launch_bounds__(256)
+void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
+                 const float* __restrict__ bias, bf16* __restrict__ out,
+                 ConvShape cs, int M, int Kgemm, int Nout,
+                 float* __restrict__ stats = nullptr,
+                 BnFuse bn = BnFuse{},
+                 const bf16* __restrict__ addin = nullptr) {
+    conv_gemm_body<MODE, TBN, WAVES_M, WAVES_N, STRIDE1, CBM_T>(
+        Asrc, Bsrc, bias, out, cs, M, Kgemm, Nout, stats, bn, addin);
+}
+
+// MODE-2 strided dgrad, all residue classes in ONE launch: blockIdx.z
+// selects the class (its own ConvShape/M/Kgemm from the by-value pack).
+// Class grids differ by at most one tile row, so grid.x is the max tile
+// count and overhanging blocks exit before touching LDS (block-uniform).
+struct ClsPack {
+    ConvShape cs[4];
+    int M[4];
+    int K[4];
+};
+
+template <int TBN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(256)
+void k_conv_gemm_cls(const bf16* __restrict__ Asrc,
+                     const bf16* __restrict__ Bsrc, bf16* __restrict__ out,
+                     ClsPack pack, int Nout) {
+    const int z = blockIdx.z;
+    const int M = pack.M[z];
+    if ((int)blockIdx.x * CBM >= M) return;
+    conv_gemm_body<2, TBN, WAVES_M, WAVES_N, true, CBM>(
+        Asrc, Bsrc, nullptr, out, pack.cs[z], M, pack.K[z], Nout,
+        nullptr, BnFuse{}, nullptr);
 }
 
 // conv with Cin < 8 (e.g. the MNIST stem, Cin=1): direct VALU kernel,
@@ -1593,8 +1649,11 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
 }
 
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
-                        long H, long W, long R, long S, long stride, long pad) {
+                        long H, long W, long R, long S, long stride, long pad,
+                        c10::optional<at::Tensor> addend) {
     // dy: NCHW logical / channels_last bf16 [N,Ko,P,Q]; wt2: [C, R*S*Ko] bf16
+    // addend: optional bf16 channels_last [N,C,H,W] streamed into the
+    // epilogue (dx += addend — the fused ResNet skip-grad; stride 1 only)
     ConvShape cs;
     cs.N = N; cs.C = C; cs.H = H; cs.W = W;
     cs.Ko = dy.size(1); cs.P = dy.size(2); cs.Q = dy.size(3);
@@ -1613,17 +1672,26 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     const bf16* dyp_ = reinterpret_cast<const bf16*>(dy.data_ptr());
     const bf16* wt2p = reinterpret_cast<const bf16*>(wt2.data_ptr());
     bf16* dxp = reinterpret_cast<bf16*>(dx.data_ptr());
+    const bf16* adp = nullptr;
+    if (addend.has_value()) {
+        TORCH_CHECK(cs.stride == 1, "fused dgrad addend needs stride 1");
+        TORCH_CHECK(addend->sizes() == dx.sizes() &&
+                    addend->scalar_type() == at::kBFloat16);
+        adp = reinterpret_cast<const bf16*>(addend->data_ptr());
+    }
     if (cs.stride == 1) {
         if (cs.C >= 128) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
             hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                               nullptr, BnFuse{}, adp);
         } else {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
             hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
                                dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C);
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                               nullptr, BnFuse{}, adp);
         }
         HIP_CHECK_LAST();
         return dx;
@@ -1652,6 +1720,13 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
             }
         if (any_empty) dx.zero_();
     }
+    // Build the per-class shapes; classes merge into ONE launch with
+    // blockIdx.z = class when they fit the 4-slot pack (stride 2 — every
+    // ResNet site); larger strides launch per class.
+    ClsPack pack{};
+    int ncls = 0;
+    bool overflow = false;
+    int maxTiles = 0;
     for (int a = 0; a < st; ++a) {
         for (int b = 0; b < st; ++b) {
             ConvShape c2 = cs;
@@ -1672,7 +1747,7 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
                     c2.off_s[c2.ns] = (b + cs.pad - sidx) / st;
                     ++c2.ns;
                 }
-            if (c2.nr == 0 || c2.ns == 0) continue;  // dx stays... no: must zero
+            if (c2.nr == 0 || c2.ns == 0) continue;  // dx pre-zeroed above
             int Ha = (cs.H - a + st - 1) / st;
             int Wb = (cs.W - b + st - 1) / st;
             c2.H = Ha; c2.W = Wb;           // fdH/fdW decode the class grid
@@ -1680,19 +1755,45 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
             c2.fdNs.init(c2.ns);
             long M2 = (long)cs.N * Ha * Wb;
             int K2 = c2.nr * c2.ns * cs.Ko;
-            if (cs.C >= 128) {
-                dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 128));
-                hipLaunchKernelGGL((k_conv_gemm<2, 128, 2, 2, true>), grid,
-                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
-                                   nullptr, dxp, c2, (int)M2, K2, cs.C);
+            if (ncls < 4) {
+                pack.cs[ncls] = c2;
+                pack.M[ncls] = (int)M2;
+                pack.K[ncls] = K2;
+                maxTiles = std::max(maxTiles, (int)ceil_div_i(M2, CBM));
+                ++ncls;
             } else {
-                dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 64));
-                hipLaunchKernelGGL((k_conv_gemm<2, 64, 4, 1, true>), grid,
-                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
-                                   nullptr, dxp, c2, (int)M2, K2, cs.C);
+                overflow = true;
+                if (cs.C >= 128) {
+                    dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 128));
+                    hipLaunchKernelGGL((k_conv_gemm<2, 128, 2, 2, true>),
+                                       grid, dim3(256), 0, stream.stream(),
+                                       dyp_, wt2p, nullptr, dxp, c2, (int)M2,
+                                       K2, cs.C);
+                } else {
+                    dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 64));
+                    hipLaunchKernelGGL((k_conv_gemm<2, 64, 4, 1, true>),
+                                       grid, dim3(256), 0, stream.stream(),
+                                       dyp_, wt2p, nullptr, dxp, c2, (int)M2,
+                                       K2, cs.C);
+                }
+                HIP_CHECK_LAST();
             }
-            HIP_CHECK_LAST();
         }
+    }
+    (void)overflow;
+    if (ncls > 0) {
+        if (cs.C >= 128) {
+            dim3 grid(maxTiles, ceil_div_i(cs.C, 128), ncls);
+            hipLaunchKernelGGL((k_conv_gemm_cls<128, 2, 2>), grid, dim3(256),
+                               0, stream.stream(), dyp_, wt2p, dxp, pack,
+                               (int)cs.C);
+        } else {
+            dim3 grid(maxTiles, ceil_div_i(cs.C, 64), ncls);
+            hipLaunchKernelGGL((k_conv_gemm_cls<64, 4, 1>), grid, dim3(256),
+                               0, stream.stream(), dyp_, wt2p, dxp, pack,
+                               (int)cs.C);
+        }
+        HIP_CHECK_LAST();
     }
     return dx;
 }

@@ -693,3 +693,125 @@ def test_nhwc_flatten_matches_torch():
         y.backward(dy)
         y_ref.backward(dy)
         assert torch.equal(x.grad.float(), x_ref.grad.float())
+
+
+# ------------------------------------------------ round-2 kernel additions ---
+
+def test_ce_wave_wide_rows():
+    """Wave-per-row CE for wide C (the ResNet-50 [B,1000] head)."""
+    for B, C in [(512, 1000), (1024, 1024), (64, 4000), (7, 1000)]:
+        torch.manual_seed(B + C)
+        logits = torch.randn(B, C, device=DEV).to(torch.bfloat16) * 4
+        target = torch.randint(0, C, (B,), device=DEV)
+        loss, lse = ext.ce_fwd(logits, target)
+        ref = torch.nn.functional.cross_entropy(logits.float(), target)
+        _close(loss, ref, rel=1e-3, atol=1e-3, name=f"ce wave fwd {B}x{C}")
+        dloss = torch.tensor(0.41, device=DEV)
+        dl = ext.ce_bwd(logits, target, lse, dloss)
+        lf = logits.float().requires_grad_(True)
+        torch.nn.functional.cross_entropy(lf, target).backward(dloss)
+        _close(dl, lf.grad, rel=2e-2, atol=1e-4, name=f"ce wave bwd {B}x{C}")
+        n = ext.argmax_correct(logits, target)
+        refn = (logits.float().argmax(dim=1) == target).sum()
+        assert n.item() == refn.item(), f"argmax wave {B}x{C}"
+
+
+def test_ce_wave_ties_first_index():
+    """Tie rows: wave argmax must keep torch's first-max-index semantics."""
+    B, C = 128, 1000
+    logits = torch.zeros(B, C, device=DEV, dtype=torch.bfloat16)
+    logits[:, 17] = 2.0
+    logits[:, 900] = 2.0   # tie — first index (17) must win
+    target = torch.full((B,), 17, device=DEV, dtype=torch.long)
+    n = ext.argmax_correct(logits, target)
+    assert n.item() == B
+    target2 = torch.full((B,), 900, device=DEV, dtype=torch.long)
+    n2 = ext.argmax_correct(logits, target2)
+    assert n2.item() == 0
+
+
+@pytest.mark.timeout(120)
+def test_bn1pass_bitwise_matches_twopass(monkeypatch):
+    """DDPX_BN1PASS single-kernel BN backward must be BITWISE equal to the
+    two-pass path (identical partial striding + combine order)."""
+    import os
+    for (N, C, H, W), relu in [((16, 64, 13, 13), True),
+                               ((8, 256, 14, 14), False),
+                               ((4, 512, 7, 7), True),
+                               ((32, 128, 28, 28), True)]:
+        torch.manual_seed(C)
+        x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+            .contiguous(memory_format=CL)
+        g = torch.randn(C, device=DEV).abs() + 0.5
+        b = torch.randn(C, device=DEV)
+        rm = torch.zeros(C, device=DEV)
+        rv = torch.ones(C, device=DEV)
+        y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, relu)
+        dy = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+            .contiguous(memory_format=CL)
+        monkeypatch.delenv("DDPX_BN1PASS", raising=False)
+        dx2, dg2, db2 = ext.bn_bwd(x, dy, g, sm, si, mask, relu)
+        torch.cuda.synchronize()
+        monkeypatch.setenv("DDPX_BN1PASS", "1")
+        dx1, dg1, db1 = ext.bn_bwd(x, dy, g, sm, si, mask, relu)
+        torch.cuda.synchronize()
+        monkeypatch.delenv("DDPX_BN1PASS", raising=False)
+        assert torch.equal(dg1, dg2), f"dgamma {C}"
+        assert torch.equal(db1, db2), f"dbeta {C}"
+        assert torch.equal(dx1, dx2), f"dx {C}"
+
+
+@pytest.mark.timeout(120)
+def test_bn1pass_dresid(monkeypatch):
+    """One-pass path with the skip-grad output (want_dresid)."""
+    N, C, H, W = 8, 128, 14, 14
+    torch.manual_seed(5)
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL)
+    g = torch.randn(C, device=DEV).abs() + 0.5
+    b = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    res = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL)
+    y, sm, si, mask = ext.bn_fwd_train(x, g, b, rm, rv, 0.1, 1e-5, True, res)
+    dy = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL)
+    monkeypatch.delenv("DDPX_BN1PASS", raising=False)
+    dx2, dg2, db2, dr2 = ext.bn_bwd(x, dy, g, sm, si, mask, True, True)
+    torch.cuda.synchronize()
+    monkeypatch.setenv("DDPX_BN1PASS", "1")
+    dx1, dg1, db1, dr1 = ext.bn_bwd(x, dy, g, sm, si, mask, True, True)
+    torch.cuda.synchronize()
+    assert torch.equal(dx1, dx2)
+    assert torch.equal(dr1, dr2)
+
+
+@pytest.mark.timeout(180)
+def test_skipfuse_grads_match_unfused(monkeypatch):
+    """DDPX_SKIPFUSE (dx += dresid in the junction conv's dgrad epilogue)
+    must produce the same gradients as the autograd-accumulated path, on a
+    mini-ResNet with both a plain block and a downsample block."""
+    from ddp_tricks_amd.models.resnet import resnet18
+    from ddp_tricks_amd import amp
+
+    def run(flag):
+        monkeypatch.setenv("DDPX_SKIPFUSE", flag)
+        amp._state.__init__()
+        amp._state.enabled = True   # bf16 compute path
+        torch.manual_seed(11)
+        model = resnet18(num_classes=10, cifar_stem=True).to(DEV)
+        x = torch.randn(16, 3, 32, 32, device=DEV)
+        y = model(x)
+        loss = y.float().pow(2).mean()
+        loss.backward()
+        grads = {k: p.grad.detach().double().clone()
+                 for k, p in model.named_parameters()}
+        amp._state.__init__()
+        return float(loss), grads
+
+    l0, g0 = run("0")
+    l1, g1 = run("1")
+    assert l0 == l1
+    for k in g0:
+        assert torch.equal(g0[k], g1[k]), f"grad mismatch at {k}"
