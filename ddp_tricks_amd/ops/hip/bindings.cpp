@@ -42,11 +42,18 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
                        at::Tensor running_mean, at::Tensor running_var,
                        double eps, bool fuse_relu,
                        c10::optional<at::Tensor> residual);
+std::vector<at::Tensor> bn_fwd_eval_mask(at::Tensor x, at::Tensor gamma,
+                                         at::Tensor beta,
+                                         at::Tensor running_mean,
+                                         at::Tensor running_var, double eps,
+                                         bool fuse_relu,
+                                         c10::optional<at::Tensor> residual);
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
                                at::Tensor save_mean, at::Tensor save_invstd,
                                at::Tensor mask, bool fuse_relu,
                                bool want_dresid,
-                               c10::optional<at::Tensor> pre_slab);
+                               c10::optional<at::Tensor> pre_slab,
+                               bool eval_stats);
 
 // gemm.hip
 at::Tensor gemm_tn(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> bias,
@@ -117,7 +124,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_bwd", &bn_bwd,
           py::arg("x"), py::arg("dy"), py::arg("gamma"), py::arg("save_mean"),
           py::arg("save_invstd"), py::arg("mask"), py::arg("fuse_relu") = false,
-          py::arg("want_dresid") = false, py::arg("pre_slab") = c10::nullopt);
+          py::arg("want_dresid") = false, py::arg("pre_slab") = c10::nullopt,
+          py::arg("eval_stats") = false);
+    m.def("bn_fwd_eval_mask", &bn_fwd_eval_mask,
+          py::arg("x"), py::arg("gamma"), py::arg("beta"),
+          py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
+          py::arg("fuse_relu") = false, py::arg("residual") = c10::nullopt);
     m.def("gemm_tn", &gemm_tn, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
     m.def("transpose_bf16", &transpose_bf16);

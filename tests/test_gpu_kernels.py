@@ -815,3 +815,32 @@ def test_skipfuse_grads_match_unfused(monkeypatch):
     assert l0 == l1
     for k in g0:
         assert torch.equal(g0[k], g1[k]), f"grad mismatch at {k}"
+
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_bn_eval_backward(relu):
+    """Frozen-stats (eval-mode) BN backward — fine-tuning support."""
+    N, C, H, W = 8, 64, 10, 10
+    torch.manual_seed(21)
+    from ddp_tricks_amd.ops.functional import batch_norm
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL).requires_grad_(True)
+    g = (torch.randn(C, device=DEV).abs() + 0.5).requires_grad_(True)
+    b = torch.randn(C, device=DEV).requires_grad_(True)
+    rm = torch.randn(C, device=DEV) * 0.2
+    rv = torch.rand(C, device=DEV) + 0.5
+    y = batch_norm(x, rm, rv, g, b, False, 0.1, 1e-5, fuse_relu=relu)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().requires_grad_(True)
+    g2 = g.detach().requires_grad_(True)
+    b2 = b.detach().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(xf, rm.clone(), rv.clone(), g2, b2,
+                                         False, 0.1, 1e-5)
+    if relu:
+        ref = ref.relu()
+    ref.backward(dy.float())
+    _close(x.grad, xf.grad, rel=3e-2, atol=2e-2, name="bn eval dx")
+    _close(g.grad, g2.grad, rel=2e-2, atol=0.1, name="bn eval dgamma")
+    _close(b.grad, b2.grad, rel=2e-2, atol=0.1, name="bn eval dbeta")
