@@ -11,6 +11,8 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <mutex>
+#include <unordered_map>
 #include "common.h"
 
 static inline int bn_splits(long M, int nw) {
@@ -435,6 +437,10 @@ void k_bn_bwd_onepass(const bf16* __restrict__ x,
         is_last = (atomicAdd(ticket, 1) == gridDim.x - 1);
     __syncthreads();
     if (is_last) {
+        // acquire: the slab lines written by blocks on OTHER XCDs must
+        // not be served from stale L2 (the allocator reuses this buffer
+        // every step)
+        __threadfence();
         // combine exactly like k_bn_bwd_combine: wave per channel, lanes
         // stride the splits, shfl tree — bitwise-equal to the 2-pass path
         const int lane = threadIdx.x & 63;
@@ -461,16 +467,21 @@ void k_bn_bwd_onepass(const bf16* __restrict__ x,
             __hip_atomic_store(flag, 1, __ATOMIC_RELEASE,
                                __HIP_MEMORY_SCOPE_AGENT);
     }
-    // ---- spin for the coefficients (bounded: bail-out, never a hang)
+    // ---- spin for the coefficients (bounded: bail-out, never a hang).
+    // Exponential backoff: hundreds of blocks polling one line at full
+    // rate congest the memory path the combiner needs.
     if (threadIdx.x == 0) {
         long spins = 0;
         while (__hip_atomic_load(flag, __ATOMIC_ACQUIRE,
                                  __HIP_MEMORY_SCOPE_AGENT) == 0) {
-            __builtin_amdgcn_s_sleep(8);
-            if (++spins > (1l << 28)) break;   // ~seconds; wrong > hung
+            // backoff: short sleeps for the first polls, long after
+            if (spins < 4) __builtin_amdgcn_s_sleep(2);
+            else __builtin_amdgcn_s_sleep(64);
+            if (++spins > (1l << 26)) break;   // ~seconds; wrong > hung
         }
     }
     __syncthreads();
+    __threadfence();   // acquire for ALL threads before reading the coefs
     // ---- phase 2: dx over the SAME rows (cache-hot x/dy/mask)
     float ca[8], cb[8], cc[8];
     #pragma unroll
@@ -510,11 +521,24 @@ static int resident_grid(const void* kfunc, int block, int lds) {
         if (hipGetDeviceProperties(&p, 0) != hipSuccess) return 0;
         numCU = p.multiProcessorCount;
     }
+    // cache per (block, lds): the occupancy query does a kernel lookup
+    // every call and BN backward runs ~50x per ResNet-50 step
+    static std::unordered_map<long long, int> cache;
+    static std::mutex mu;
+    long long key = ((long long)block << 32) | (unsigned)lds;
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        auto it = cache.find(key);
+        if (it != cache.end()) return it->second;
+    }
     int per = 0;
+    int grid = 0;
     if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &per, kfunc, block, lds) != hipSuccess || per < 1)
-        return 0;
-    return numCU * per;
+            &per, kfunc, block, lds) == hipSuccess && per >= 1)
+        grid = numCU * per;
+    std::lock_guard<std::mutex> lk(mu);
+    cache[key] = grid;
+    return grid;
 }
 
 static void shape_mc(const at::Tensor& x, long& M, int& C) {
@@ -719,7 +743,9 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
     if (e1p && e1p[0] == '1' && !pre_slab.has_value() && !eval_stats) {
         int grid = resident_grid((const void*)k_bn_bwd_onepass, block, lds);
         if (grid > 0) {
-            int S1 = grid;
+            // no empty splits: blocks with no rows still pay the full
+            // rendezvous+spin, so cap the grid at the rows available
+            int S1 = (int)std::min<long>(grid, ceil_div_i(M, nw));
             auto slab1 = at::empty({2, C, S1}, fopts);
             auto work = at::zeros({2}, gamma.options().dtype(at::kInt));
             at::Tensor dresid1;

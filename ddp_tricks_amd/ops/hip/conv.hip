@@ -83,7 +83,7 @@ struct BnFuse {
 // skip-gradient accumulation fused into the junction conv's dgrad
 // (VERDICT r01 next-round #4).
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1,
-          int CBM_T>
+          int CBM_T, bool PF>
 __device__ __forceinline__
 void conv_gemm_body(const bf16* __restrict__ Asrc,
                     const bf16* __restrict__ Bsrc,
@@ -187,7 +187,7 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
         if (kinc_big) {
             kc_c += CBK;
             if (kc_c >= kfast) {
-                kc_c = 0;
+                kc_c -= kfast;   // back to this thread's channel offset
                 if (++kc_s >= ksec) { kc_s = 0; ++kc_r; }
             }
         } else if (kdrs) {                           // c fixed, taps walk
@@ -289,14 +289,16 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
     for (int kt = 0; kt < Kgemm; kt += CBK) {
         const bool has_next = kt + CBK < Kgemm;
         __syncthreads();   // tile kt staged (glds vmcnt / ds_write lgkmcnt)
-        bf16x8_t pre[A_PER];
+        bf16x8_t pre[PF ? A_PER : 1];
         if (has_next) {
             gk_cur += CBK;
             k_advance(gk_cur);
-            #pragma unroll
-            for (int ci = 0; ci < A_PER; ++ci)
-                pre[ci] = *reinterpret_cast<const bf16x8_t*>(
-                    a_addr(ci, gk_cur));
+            if constexpr (PF) {
+                #pragma unroll
+                for (int ci = 0; ci < A_PER; ++ci)
+                    pre[ci] = *reinterpret_cast<const bf16x8_t*>(
+                        a_addr(ci, gk_cur));
+            }
         }
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
@@ -323,11 +325,24 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
         }
         __syncthreads();   // every wave done reading tile kt
         if (has_next) {
-            #pragma unroll
-            for (int ci = 0; ci < A_PER; ++ci) {
-                const int ch = wid + ci * 4;
-                *reinterpret_cast<bf16x8_t*>(
-                    &lds_a[0][ch * 8 + pl_row][pl_segp * 8]) = pre[ci];
+            if constexpr (PF) {
+                #pragma unroll
+                for (int ci = 0; ci < A_PER; ++ci) {
+                    const int ch = wid + ci * 4;
+                    *reinterpret_cast<bf16x8_t*>(
+                        &lds_a[0][ch * 8 + pl_row][pl_segp * 8]) = pre[ci];
+                }
+            } else {
+                #pragma unroll
+                for (int ci = 0; ci < A_PER; ++ci) {
+                    const int ch = wid + ci * 4;
+                    __builtin_amdgcn_global_load_lds(
+                        (const __attribute__((address_space(1))) unsigned short*)
+                            a_addr(ci, gk_cur),
+                        (__attribute__((address_space(3))) unsigned short*)
+                            &lds_a[0][ch * 8][0],
+                        16, 0, 0);
+                }
             }
             issue_b_tile(kt + CBK);
         }
@@ -513,7 +528,7 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
 }
 
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
-          int CBM_T = CBM>
+          int CBM_T = CBM, bool PF = true>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
@@ -521,8 +536,19 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  float* __restrict__ stats = nullptr,
                  BnFuse bn = BnFuse{},
                  const bf16* __restrict__ addin = nullptr) {
-    conv_gemm_body<MODE, TBN, WAVES_M, WAVES_N, STRIDE1, CBM_T>(
+    conv_gemm_body<MODE, TBN, WAVES_M, WAVES_N, STRIDE1, CBM_T, PF>(
         Asrc, Bsrc, bias, out, cs, M, Kgemm, Nout, stats, bn, addin);
+}
+
+// DDPX_CONV_PF=0 disables the register-staged A prefetch (A/B lever; the
+// glds path is otherwise identical — same decode, same LDS image).
+static bool conv_pf() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("DDPX_CONV_PF");
+        v = (!e || e[0] != '0') ? 1 : 0;
+    }
+    return v == 1;
 }
 
 // MODE-2 strided dgrad, all residue classes in ONE launch: blockIdx.z
@@ -543,7 +569,7 @@ void k_conv_gemm_cls(const bf16* __restrict__ Asrc,
     const int z = blockIdx.z;
     const int M = pack.M[z];
     if ((int)blockIdx.x * CBM >= M) return;
-    conv_gemm_body<2, TBN, WAVES_M, WAVES_N, true, CBM>(
+    conv_gemm_body<2, TBN, WAVES_M, WAVES_N, true, CBM, true>(
         Asrc, Bsrc, nullptr, out, pack.cs[z], M, pack.K[z], Nout,
         nullptr, BnFuse{}, nullptr);
 }
@@ -1630,18 +1656,30 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
         static const char* tbn_env = getenv("DDPX_CONV_TBN64");
         if (cs.Ko >= 128 && !tbn_env) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 128));
-            hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid, dim3(256), 0,
-                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
-                               Kgemm, cs.Ko);
+            if (conv_pf())
+                hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid,
+                                   dim3(256), 0, stream.stream(), xp, wp, bp,
+                                   yp, cs, (int)M, Kgemm, cs.Ko);
+            else
+                hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2, true, CBM,
+                                    false>), grid, dim3(256), 0,
+                                   stream.stream(), xp, wp, bp, yp, cs,
+                                   (int)M, Kgemm, cs.Ko);
         } else {
             // NOTE: a 256-row tile for the narrow (Ko<128) case was tried
             // and REVERTED — its 40 KB LDS drops residency 6 -> 4 blocks/CU
             // and measured 20-45% slower (same lesson as the 2-buf glds
             // ring: this kernel lives on block-level parallelism).
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
-            hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
-                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
-                               Kgemm, cs.Ko);
+            if (conv_pf())
+                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid,
+                                   dim3(256), 0, stream.stream(), xp, wp, bp,
+                                   yp, cs, (int)M, Kgemm, cs.Ko);
+            else
+                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1, true, CBM,
+                                    false>), grid, dim3(256), 0,
+                                   stream.stream(), xp, wp, bp, yp, cs,
+                                   (int)M, Kgemm, cs.Ko);
         }
     }
     HIP_CHECK_LAST();
@@ -1682,16 +1720,30 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     if (cs.stride == 1) {
         if (cs.C >= 128) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
-            hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
-                               dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
-                               nullptr, BnFuse{}, adp);
+            if (conv_pf())
+                hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
+                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
+                                   nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                                   nullptr, BnFuse{}, adp);
+            else
+                hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true, CBM,
+                                    false>), grid, dim3(256), 0,
+                                   stream.stream(), dyp_, wt2p, nullptr, dxp,
+                                   cs, (int)M, Kgemm, cs.C, nullptr,
+                                   BnFuse{}, adp);
         } else {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
-            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
-                               dim3(256), 0, stream.stream(), dyp_, wt2p,
-                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
-                               nullptr, BnFuse{}, adp);
+            if (conv_pf())
+                hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
+                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
+                                   nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                                   nullptr, BnFuse{}, adp);
+            else
+                hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true, CBM,
+                                    false>), grid, dim3(256), 0,
+                                   stream.stream(), dyp_, wt2p, nullptr, dxp,
+                                   cs, (int)M, Kgemm, cs.C, nullptr,
+                                   BnFuse{}, adp);
         }
         HIP_CHECK_LAST();
         return dx;
