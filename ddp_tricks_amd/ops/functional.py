@@ -363,8 +363,11 @@ class _HIPBatchNorm(torch.autograd.Function):
             ctx.bnbwd_box = {"x": xb, "mask": mask, "mean": save_mean,
                              "invstd": save_invstd, "slab": None,
                              "dx_ref": None}
-        elif torch.is_grad_enabled() and (x.requires_grad
-                                          or weight.requires_grad):
+        elif (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]
+              or ctx.needs_input_grad[2]):
+            # (grad mode is force-disabled inside Function.forward, so the
+            # needs_input_grad flags — not torch.is_grad_enabled() — tell
+            # whether a backward will run)
             # eval-mode BN on the grad path (fine-tuning with frozen stats;
             # VERDICT r01 weak #8): keep the relu mask + frozen stats for
             # backward.  The C-length stat vectors are cloned — the DDP

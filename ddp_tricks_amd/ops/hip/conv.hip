@@ -83,7 +83,7 @@ struct BnFuse {
 // skip-gradient accumulation fused into the junction conv's dgrad
 // (VERDICT r01 next-round #4).
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1,
-          int CBM_T, bool PF>
+          int CBM_T>
 __device__ __forceinline__
 void conv_gemm_body(const bf16* __restrict__ Asrc,
                     const bf16* __restrict__ Bsrc,
@@ -159,87 +159,62 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
         }
     }
 
-    // Seg-constant k decode (round-2): a thread's logical k-segment
-    // seg = pl_segp ^ pl_row is FIXED (row & 7 == pl_row for every chunk),
-    // so its gk = kt + seg*8 advances by exactly CBK per tile.  One FastDiv
-    // decode at gk0 seeds (tap, fastdim) coordinates; afterwards they
-    // advance with constant increments (every supported shape has the fast
-    // dim — C or Ko, a multiple of 8 — dividing CBK or divided by it).
-    // Odd shapes fall back to the per-tile FastDiv decode.
-    const int kfast = (MODE == 0) ? cs.C : cs.Ko;          // fastest k dim
-    const int ksec = (MODE == 2) ? cs.ns : cs.S;           // second k dim
-    const int gk0 = (pl_segp ^ pl_row) * 8;
-    int kc_r, kc_s, kc_c;                                   // decoded coords
-    {
-        unsigned rs = (MODE == 0) ? fd_div(gk0, cs.fdC) : fd_div(gk0, cs.fdKo);
-        kc_c = (MODE == 0) ? fd_mod(gk0, cs.fdC, rs) : fd_mod(gk0, cs.fdKo, rs);
-        if (MODE == 2) {
-            kc_r = fd_div(rs, cs.fdNs);
-            kc_s = fd_mod(rs, cs.fdNs, kc_r);
-        } else {
-            kc_r = fd_div(rs, cs.fdS);
-            kc_s = fd_mod(rs, cs.fdS, kc_r);
-        }
-    }
-    const bool kinc_big = (kfast % CBK) == 0;       // c walks, rare carries
-    const int kdrs = (kfast < CBK && (CBK % kfast) == 0) ? CBK / kfast : 0;
-    auto k_advance = [&](int gk_next) {
-        if (kinc_big) {
-            kc_c += CBK;
-            if (kc_c >= kfast) {
-                kc_c -= kfast;   // back to this thread's channel offset
-                if (++kc_s >= ksec) { kc_s = 0; ++kc_r; }
+    auto issue_tile = [&](int kt, int buf) {
+        // ---- A tile: chunks round-robined over the 4 waves ----
+        #pragma unroll
+        for (int ci = 0; ci < A_PER; ++ci) {
+            const int ch = wid + ci * 4;
+            int row = ch * 8 + pl_row;
+            int seg = pl_segp ^ (row & 7);     // logical k-segment
+            int gk = kt + seg * 8;
+            const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
+            if (ri_u[ci] != INT_MIN && gk < Kgemm) {
+                if (MODE == 0) {
+                    unsigned rs = fd_div(gk, cs.fdC);
+                    int c = fd_mod(gk, cs.fdC, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int h = ri_u[ci] + r;
+                    int wcol = ri_v[ci] + sx;
+                    if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
+                        src = &Asrc[((ri_nb[ci] + h) * cs.W + wcol) * cs.C + c];
+                } else if (MODE == 2) {
+                    unsigned rs2 = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs2);
+                    int ti = fd_div(rs2, cs.fdNs);
+                    int si = fd_mod(rs2, cs.fdNs, ti);
+                    int p = ri_u[ci] + cs.off_r[ti];
+                    int q = ri_v[ci] + cs.off_s[si];
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
+                } else if (STRIDE1) {
+                    unsigned rs = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int p = ri_u[ci] - r;
+                    int q = ri_v[ci] - sx;
+                    if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
+                } else {
+                    unsigned rs = fd_div(gk, cs.fdKo);
+                    int ko = fd_mod(gk, cs.fdKo, rs);
+                    int r = fd_div(rs, cs.fdS);
+                    int sx = fd_mod(rs, cs.fdS, r);
+                    int pn = ri_u[ci] - r;
+                    int qn = ri_v[ci] - sx;
+                    int p = pn / cs.stride, q = qn / cs.stride;
+                    if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
+                        qn == q * cs.stride && p < cs.P && q < cs.Q)
+                        src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + ko];
+                }
             }
-        } else if (kdrs) {                           // c fixed, taps walk
-            kc_s += kdrs;
-            while (kc_s >= ksec) { kc_s -= ksec; ++kc_r; }
-        } else {
-            unsigned rs = (MODE == 0) ? fd_div(gk_next, cs.fdC)
-                                      : fd_div(gk_next, cs.fdKo);
-            kc_c = (MODE == 0) ? fd_mod(gk_next, cs.fdC, rs)
-                               : fd_mod(gk_next, cs.fdKo, rs);
-            if (MODE == 2) {
-                kc_r = fd_div(rs, cs.fdNs);
-                kc_s = fd_mod(rs, cs.fdNs, kc_r);
-            } else {
-                kc_r = fd_div(rs, cs.fdS);
-                kc_s = fd_mod(rs, cs.fdS, kc_r);
-            }
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned short*)src,
+                (__attribute__((address_space(3))) unsigned short*)&lds_a[buf][ch * 8][0],
+                16, 0, 0);
         }
-    };
-
-    // A-side gather address for chunk ci at decoded coords (kc_r=tap/ti,
-    // kc_s=tap2/si, kc_c=c/ko); out-of-range lanes read the zero page.
-    auto a_addr = [&](int ci, int gk) -> const bf16* {
-        const bf16* src = reinterpret_cast<const bf16*>(g_zero16);
-        if (ri_u[ci] == INT_MIN || gk >= Kgemm) return src;
-        if (MODE == 0) {
-            int h = ri_u[ci] + kc_r;
-            int wcol = ri_v[ci] + kc_s;
-            if (h >= 0 && h < cs.H && wcol >= 0 && wcol < cs.W)
-                src = &Asrc[((ri_nb[ci] + h) * cs.W + wcol) * cs.C + kc_c];
-        } else if (MODE == 2) {
-            int p = ri_u[ci] + cs.off_r[kc_r];
-            int q = ri_v[ci] + cs.off_s[kc_s];
-            if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
-        } else if (STRIDE1) {
-            int p = ri_u[ci] - kc_r;
-            int q = ri_v[ci] - kc_s;
-            if (p >= 0 && p < cs.P && q >= 0 && q < cs.Q)
-                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
-        } else {
-            int pn = ri_u[ci] - kc_r;
-            int qn = ri_v[ci] - kc_s;
-            int p = pn / cs.stride, q = qn / cs.stride;
-            if (pn >= 0 && qn >= 0 && pn == p * cs.stride &&
-                qn == q * cs.stride && p < cs.P && q < cs.Q)
-                src = &Asrc[((ri_nb[ci] + p) * cs.Q + q) * cs.Ko + kc_c];
-        }
-        return src;
-    };
-
-    auto issue_b_tile = [&](int kt) {
+        // ---- B tile (row-contiguous source) ----
         for (int ch = wid; ch < B_CHUNKS; ch += 4) {
             int row = ch * 8 + pl_row;
             int seg = pl_segp ^ (row & 7);
@@ -260,46 +235,15 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
                 src = &Bsrc[(long)gn * Kgemm + gk];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) unsigned short*)src,
-                (__attribute__((address_space(3))) unsigned short*)&lds_b[0][ch * 8][0],
+                (__attribute__((address_space(3))) unsigned short*)&lds_b[buf][ch * 8][0],
                 16, 0, 0);
         }
     };
 
-    // Prologue: stage tile 0 — A via glds (no VGPR round trip), B via glds.
-    #pragma unroll
-    for (int ci = 0; ci < A_PER; ++ci) {
-        const int ch = wid + ci * 4;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned short*)
-                a_addr(ci, gk0),
-            (__attribute__((address_space(3))) unsigned short*)&lds_a[0][ch * 8][0],
-            16, 0, 0);
-    }
-    issue_b_tile(0);
-
-    // k-loop with register-staged A prefetch (round-2): the NEXT tile's A
-    // gathers are issued as plain b128 loads right after the barrier, so
-    // their HBM latency hides under this tile's MFMA block; after the
-    // tiles are dead they are ds_written to the SAME swizzled LDS image
-    // the glds path produced (+16 VGPR, LDS unchanged — stays at 3
-    // waves/SIMD; the 2-buffer LDS ring alternative was measured 10-30%
-    // slower, see note above).
     constexpr int buf = 0;
-    int gk_cur = gk0;
     for (int kt = 0; kt < Kgemm; kt += CBK) {
-        const bool has_next = kt + CBK < Kgemm;
-        __syncthreads();   // tile kt staged (glds vmcnt / ds_write lgkmcnt)
-        bf16x8_t pre[PF ? A_PER : 1];
-        if (has_next) {
-            gk_cur += CBK;
-            k_advance(gk_cur);
-            if constexpr (PF) {
-                #pragma unroll
-                for (int ci = 0; ci < A_PER; ++ci)
-                    pre[ci] = *reinterpret_cast<const bf16x8_t*>(
-                        a_addr(ci, gk_cur));
-            }
-        }
+        issue_tile(kt, 0);
+        __syncthreads();   // vmcnt(0) for the in-flight glds + barrier
         #pragma unroll
         for (int ks = 0; ks < CBK; ks += 32) {
             bf16x8_t af[MI], bfr[NI];
@@ -323,29 +267,7 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
         }
-        __syncthreads();   // every wave done reading tile kt
-        if (has_next) {
-            if constexpr (PF) {
-                #pragma unroll
-                for (int ci = 0; ci < A_PER; ++ci) {
-                    const int ch = wid + ci * 4;
-                    *reinterpret_cast<bf16x8_t*>(
-                        &lds_a[0][ch * 8 + pl_row][pl_segp * 8]) = pre[ci];
-                }
-            } else {
-                #pragma unroll
-                for (int ci = 0; ci < A_PER; ++ci) {
-                    const int ch = wid + ci * 4;
-                    __builtin_amdgcn_global_load_lds(
-                        (const __attribute__((address_space(1))) unsigned short*)
-                            a_addr(ci, gk_cur),
-                        (__attribute__((address_space(3))) unsigned short*)
-                            &lds_a[0][ch * 8][0],
-                        16, 0, 0);
-                }
-            }
-            issue_b_tile(kt + CBK);
-        }
+        __syncthreads();
     }
 
     float psum[NI] = {}, psq[NI] = {};
@@ -528,7 +450,7 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
 }
 
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
-          int CBM_T = CBM, bool PF = true>
+          int CBM_T = CBM>
 __global__ __launch_bounds__(256)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
@@ -536,20 +458,10 @@ void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  float* __restrict__ stats = nullptr,
                  BnFuse bn = BnFuse{},
                  const bf16* __restrict__ addin = nullptr) {
-    conv_gemm_body<MODE, TBN, WAVES_M, WAVES_N, STRIDE1, CBM_T, PF>(
+    conv_gemm_body<MODE, TBN, WAVES_M, WAVES_N, STRIDE1, CBM_T>(
         Asrc, Bsrc, bias, out, cs, M, Kgemm, Nout, stats, bn, addin);
 }
 
-// DDPX_CONV_PF=0 disables the register-staged A prefetch (A/B lever; the
-// glds path is otherwise identical — same decode, same LDS image).
-static bool conv_pf() {
-    static int v = -1;
-    if (v < 0) {
-        const char* e = getenv("DDPX_CONV_PF");
-        v = (!e || e[0] != '0') ? 1 : 0;
-    }
-    return v == 1;
-}
 
 // MODE-2 strided dgrad, all residue classes in ONE launch: blockIdx.z
 // selects the class (its own ConvShape/M/Kgemm from the by-value pack).
@@ -569,7 +481,7 @@ void k_conv_gemm_cls(const bf16* __restrict__ Asrc,
     const int z = blockIdx.z;
     const int M = pack.M[z];
     if ((int)blockIdx.x * CBM >= M) return;
-    conv_gemm_body<2, TBN, WAVES_M, WAVES_N, true, CBM, true>(
+    conv_gemm_body<2, TBN, WAVES_M, WAVES_N, true, CBM>(
         Asrc, Bsrc, nullptr, out, pack.cs[z], M, pack.K[z], Nout,
         nullptr, BnFuse{}, nullptr);
 }
@@ -1656,30 +1568,18 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
         static const char* tbn_env = getenv("DDPX_CONV_TBN64");
         if (cs.Ko >= 128 && !tbn_env) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 128));
-            if (conv_pf())
-                hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid,
-                                   dim3(256), 0, stream.stream(), xp, wp, bp,
-                                   yp, cs, (int)M, Kgemm, cs.Ko);
-            else
-                hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2, true, CBM,
-                                    false>), grid, dim3(256), 0,
-                                   stream.stream(), xp, wp, bp, yp, cs,
-                                   (int)M, Kgemm, cs.Ko);
+            hipLaunchKernelGGL((k_conv_gemm<0, 128, 2, 2>), grid, dim3(256), 0,
+                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                               Kgemm, cs.Ko);
         } else {
             // NOTE: a 256-row tile for the narrow (Ko<128) case was tried
             // and REVERTED — its 40 KB LDS drops residency 6 -> 4 blocks/CU
             // and measured 20-45% slower (same lesson as the 2-buf glds
             // ring: this kernel lives on block-level parallelism).
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.Ko, 64));
-            if (conv_pf())
-                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid,
-                                   dim3(256), 0, stream.stream(), xp, wp, bp,
-                                   yp, cs, (int)M, Kgemm, cs.Ko);
-            else
-                hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1, true, CBM,
-                                    false>), grid, dim3(256), 0,
-                                   stream.stream(), xp, wp, bp, yp, cs,
-                                   (int)M, Kgemm, cs.Ko);
+            hipLaunchKernelGGL((k_conv_gemm<0, 64, 4, 1>), grid, dim3(256), 0,
+                               stream.stream(), xp, wp, bp, yp, cs, (int)M,
+                               Kgemm, cs.Ko);
         }
     }
     HIP_CHECK_LAST();
@@ -1720,30 +1620,16 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     if (cs.stride == 1) {
         if (cs.C >= 128) {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 128));
-            if (conv_pf())
-                hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
-                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
-                                   nullptr, dxp, cs, (int)M, Kgemm, cs.C,
-                                   nullptr, BnFuse{}, adp);
-            else
-                hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true, CBM,
-                                    false>), grid, dim3(256), 0,
-                                   stream.stream(), dyp_, wt2p, nullptr, dxp,
-                                   cs, (int)M, Kgemm, cs.C, nullptr,
-                                   BnFuse{}, adp);
+            hipLaunchKernelGGL((k_conv_gemm<1, 128, 2, 2, true>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                               nullptr, BnFuse{}, adp);
         } else {
             dim3 grid(ceil_div_i(M, CBM), ceil_div_i(cs.C, 64));
-            if (conv_pf())
-                hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
-                                   dim3(256), 0, stream.stream(), dyp_, wt2p,
-                                   nullptr, dxp, cs, (int)M, Kgemm, cs.C,
-                                   nullptr, BnFuse{}, adp);
-            else
-                hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true, CBM,
-                                    false>), grid, dim3(256), 0,
-                                   stream.stream(), dyp_, wt2p, nullptr, dxp,
-                                   cs, (int)M, Kgemm, cs.C, nullptr,
-                                   BnFuse{}, adp);
+            hipLaunchKernelGGL((k_conv_gemm<1, 64, 4, 1, true>), grid,
+                               dim3(256), 0, stream.stream(), dyp_, wt2p,
+                               nullptr, dxp, cs, (int)M, Kgemm, cs.C,
+                               nullptr, BnFuse{}, adp);
         }
         HIP_CHECK_LAST();
         return dx;
