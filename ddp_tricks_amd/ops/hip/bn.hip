@@ -437,18 +437,21 @@ void k_bn_bwd_onepass(const bf16* __restrict__ x,
         is_last = (atomicAdd(ticket, 1) == gridDim.x - 1);
     __syncthreads();
     if (is_last) {
-        // acquire: the slab lines written by blocks on OTHER XCDs must
-        // not be served from stale L2 (the allocator reuses this buffer
-        // every step)
-        __threadfence();
-        // combine exactly like k_bn_bwd_combine: wave per channel, lanes
-        // stride the splits, shfl tree — bitwise-equal to the 2-pass path
+        // combine like k_bn_bwd_combine (wave per channel, lanes stride
+        // the splits).  The slab lines were written by blocks on OTHER
+        // XCDs: read them with cache-bypassing (relaxed agent-scope
+        // atomic) loads — an acquire FENCE here would invalidate this
+        // XCD's whole L2 and destroy the phase-2 cache-reuse premise.
         const int lane = threadIdx.x & 63;
         for (int c = (int)(threadIdx.x >> 6); c < C; c += (int)(blockDim.x >> 6)) {
             float a = 0.f, b = 0.f;
             for (int sp = lane; sp < S; sp += 64) {
-                a += slab[(long)c * S + sp];
-                b += slab[(long)C * S + (long)c * S + sp];
+                a += __hip_atomic_load(&slab[(long)c * S + sp],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                b += __hip_atomic_load(&slab[(long)C * S + (long)c * S + sp],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
             }
             a = wave_reduce_sum(a);
             b = wave_reduce_sum(b);
@@ -468,28 +471,33 @@ void k_bn_bwd_onepass(const bf16* __restrict__ x,
                                __HIP_MEMORY_SCOPE_AGENT);
     }
     // ---- spin for the coefficients (bounded: bail-out, never a hang).
-    // Exponential backoff: hundreds of blocks polling one line at full
-    // rate congest the memory path the combiner needs.
+    // RELAXED polls: an acquire load per poll emits an L2 invalidate —
+    // hundreds of blocks polling would storm every XCD's L2 (measured:
+    // ~20 ms per call).  A relaxed agent-scope atomic load bypasses the
+    // stale cache without invalidating anything.
     if (threadIdx.x == 0) {
         long spins = 0;
-        while (__hip_atomic_load(flag, __ATOMIC_ACQUIRE,
+        while (__hip_atomic_load(flag, __ATOMIC_RELAXED,
                                  __HIP_MEMORY_SCOPE_AGENT) == 0) {
-            // backoff: short sleeps for the first polls, long after
             if (spins < 4) __builtin_amdgcn_s_sleep(2);
             else __builtin_amdgcn_s_sleep(64);
             if (++spins > (1l << 26)) break;   // ~seconds; wrong > hung
         }
     }
     __syncthreads();
-    __threadfence();   // acquire for ALL threads before reading the coefs
-    // ---- phase 2: dx over the SAME rows (cache-hot x/dy/mask)
+    // ---- phase 2: dx over the SAME rows (cache-hot x/dy/mask).  The
+    // coefficients come from the last block (another XCD): bypassing
+    // loads again, NOT a fence — x/dy/mask must stay in cache.
     float ca[8], cb[8], cc[8];
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
         int c = c8 * 8 + j;
-        ca[j] = coef_a[c];
-        cb[j] = coef_b[c];
-        cc[j] = coef_c[c];
+        ca[j] = __hip_atomic_load(&coef_a[c], __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+        cb[j] = __hip_atomic_load(&coef_b[c], __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+        cc[j] = __hip_atomic_load(&coef_c[c], __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
     }
     for (long r = r0; r < M; r += rstep) {
         s16x8 vx = reinterpret_cast<const s16x8*>(x + r * C)[c8];
