@@ -451,7 +451,11 @@ void conv_gemm_body(const bf16* __restrict__ Asrc,
 
 template <int MODE, int TBN, int WAVES_M, int WAVES_N, bool STRIDE1 = true,
           int CBM_T = CBM>
-__global__ __launch_bounds__(256)
+// min-4-waves/SIMD bound: the allocator packs the body into 122 arch
+// VGPRs (no AGPRs, zero spills) instead of 84V+64A, lifting residency
+// 3 -> 4 blocks/CU on the 128-wide tiles (and 4 -> 5 on the 64-wide) —
+// these gather-fed GEMMs live on block-level parallelism.
+__global__ __launch_bounds__(256, 4)
 void k_conv_gemm(const bf16* __restrict__ Asrc, const bf16* __restrict__ Bsrc,
                  const float* __restrict__ bias, bf16* __restrict__ out,
                  ConvShape cs, int M, int Kgemm, int Nout,
@@ -474,7 +478,7 @@ struct ClsPack {
 };
 
 template <int TBN, int WAVES_M, int WAVES_N>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256, 4)
 void k_conv_gemm_cls(const bf16* __restrict__ Asrc,
                      const bf16* __restrict__ Bsrc, bf16* __restrict__ out,
                      ClsPack pack, int Nout) {
