@@ -1384,42 +1384,32 @@ __global__ void k_wgrad_c1_r3(const bf16* __restrict__ dy,
                                        q * cs.stride + s - cs.pad);
         };
         fill();
-        // dy loads batched 8-deep: the old single-row lookahead kept ONE
-        // load in flight per wave, so the row chain ran at HBM latency
-        // (137 us for 11 us of traffic).  Accumulation order is unchanged
-        // (sequential rows) — results stay bitwise identical.
-        long r = gm;
-        while (r < gm_end) {
-            const int batch = (int)min((long)8, gm_end - r);
-            bf16 gbuf[8];
+        // NOTE: an 8-deep dy load batch was tried and REVERTED (161.7 vs
+        // 142.2 us measured): the per-row tail branches inside the
+        // unrolled batch cost more than the extra loads-in-flight win —
+        // the sequential 128 B/row stream already rides the L2 prefetch.
+        bf16 gnext = dy[gm * cs.Ko + ko];
+        for (;;) {
+            float g = bf2f(gnext);
+            if (gm + 1 < gm_end)               // prefetch next row's dy
+                gnext = dy[(gm + 1) * cs.Ko + ko];
             #pragma unroll
-            for (int i = 0; i < 8; ++i)
-                if (i < batch) gbuf[i] = dy[(r + i) * cs.Ko + ko];
-            #pragma unroll
-            for (int i = 0; i < 8; ++i) {
-                if (i >= batch) break;
-                float g = bf2f(gbuf[i]);
+            for (int k = 0; k < 9; ++k) acc[k] = fmaf(g, xw[k], acc[k]);
+            if (++gm >= gm_end) break;
+            if (++q == cs.Q) {
+                q = 0;
+                if (++p == cs.P) { p = 0; ++n; }
+                fill();
+            } else if (cs.stride == 1) {
                 #pragma unroll
-                for (int k = 0; k < 9; ++k) acc[k] = fmaf(g, xw[k], acc[k]);
-                if (r + i + 1 < gm_end) {       // window for the NEXT row
-                    if (++q == cs.Q) {
-                        q = 0;
-                        if (++p == cs.P) { p = 0; ++n; }
-                        fill();
-                    } else if (cs.stride == 1) {
-                        #pragma unroll
-                        for (int rr = 0; rr < 3; ++rr) {
-                            xw[rr * 3] = xw[rr * 3 + 1];
-                            xw[rr * 3 + 1] = xw[rr * 3 + 2];
-                            xw[rr * 3 + 2] = ld(p + rr - cs.pad,
-                                                q + 2 - cs.pad);
-                        }
-                    } else {
-                        fill();
-                    }
+                for (int r = 0; r < 3; ++r) {
+                    xw[r * 3] = xw[r * 3 + 1];
+                    xw[r * 3 + 1] = xw[r * 3 + 2];
+                    xw[r * 3 + 2] = ld(p + r - cs.pad, q + 2 - cs.pad);
                 }
+            } else {
+                fill();
             }
-            r += batch;
         }
     }
     __shared__ float red[4][64];
