@@ -802,7 +802,7 @@ void k_conv_wgrad_sb(const bf16* __restrict__ dy, const bf16* __restrict__ x,
 // per-thread staging cost per element (transposed bf16x8 stores).
 // Requires Ko >= 128 and Kgemm >= 128 (host falls back to sb otherwise).
 template <int DEPTH>
-__global__ __launch_bounds__(256, 4)
+__global__ __launch_bounds__(256)
 void k_conv_wgrad_wide(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                        float* __restrict__ slab, ConvShape cs, long M,
                        int Kgemm, int S) {
@@ -1213,7 +1213,11 @@ void k_conv_wgrad_wide_tr(const bf16* __restrict__ dy,
 // LDS writes are packed b32 (16 stores/operand/iter vs 32 conflicted b16) —
 // the stage phase of the wide kernel is store-issue bound.
 template <int DEPTH = 64>
-__global__ __launch_bounds__(256, 4)
+__global__ // NOTE: a 4-waves/SIMD bound (V 78+64A -> 128+0A, occ 3->4) was tried
+// here like k_conv_gemm's and REVERTED: measured 285.7 vs 263.1 us —
+// this kernel's pair-store stage phase wants the AGPR split, not
+// residency (same lesson as the depth-64 staging dead ends).
+__launch_bounds__(256)
 void k_conv_wgrad_wide_pair(const bf16* __restrict__ dy,
                             const bf16* __restrict__ x,
                             float* __restrict__ slab, ConvShape cs, long M,

@@ -27,7 +27,7 @@ constexpr int GEMM_THREADS = 256;
 // ------------------------------------------------------------ core kernel ---
 
 template <bool OUT_F32, bool SPLITK>
-__global__ __launch_bounds__(GEMM_THREADS, 4)
+__global__ __launch_bounds__(GEMM_THREADS)
 void k_gemm_tn(const bf16* __restrict__ A, const bf16* __restrict__ B,
                const float* __restrict__ bias, void* __restrict__ Cout,
                int M, int N, int K, int k_per_split) {
