@@ -132,9 +132,15 @@ def test_resnet_step_gpu(name, kw, shape):
     assert losses[-1] < losses[0], losses
 
 
-def test_resnet18_matches_cpu_oracle():
-    """One bf16 HIP fwd/bwd of ResNet-18 vs the CPU fp32 oracle, same
-    weights/batch: logits and a sample of grads agree to bf16 tolerance."""
+import pytest as _pytest
+
+
+@_pytest.mark.parametrize("arch", ["resnet18", "resnet50"])
+def test_resnet_matches_cpu_oracle(arch):
+    """One bf16 HIP fwd/bwd of a ResNet vs the CPU fp32 oracle, same
+    weights/batch: logits and a sample of grads agree to bf16 tolerance.
+    resnet50 adds Bottleneck/1x1-junction coverage (incl. the skip-grad
+    fusion path through conv1x1 dgrad)."""
     from ddp_tricks_amd import amp, same_seeds
     from ddp_tricks_amd.models import build_model
     from ddp_tricks_amd.ops.functional import (clear_weight_cache,
@@ -144,7 +150,7 @@ def test_resnet18_matches_cpu_oracle():
         amp._state.__init__()
         clear_weight_cache()
         same_seeds(3)
-        model = build_model("resnet18", num_classes=10, cifar_stem=True).to(dev)
+        model = build_model(arch, num_classes=10, cifar_stem=True).to(dev)
         if dev.type == "cuda":
             model, _ = amp.initialize(model, None, opt_level="O1")
         g = torch.Generator().manual_seed(4)
