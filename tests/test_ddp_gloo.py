@@ -232,3 +232,56 @@ def test_full_stack_ranks_stay_identical():
         g1, p1 = got[1][step]
         assert g0 == pytest.approx(g1, abs=1e-12), f"grads differ at step {step}"
         assert p0 == pytest.approx(p1, abs=1e-12), f"params differ at step {step}"
+
+
+def _worker_w4(rank, port, results):
+    """World-4 full-stack lockstep (closer to the 8-GPU driver shape):
+    bucket count > world size, different data per rank."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    from ddp_tricks_amd import amp
+    from ddp_tricks_amd.ops.optim import FusedSGD
+    from ddp_tricks_amd.parallel.ddp import DistributedDataParallel as DDP
+    from ddp_tricks_amd.utils.lookahead import Lookahead
+    amp._state.__init__()
+    torch.manual_seed(7)
+    model = TinyNet()
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, nesterov=True)
+    la = Lookahead(opt, k=2, alpha=0.5)
+    model, apex_opt = amp.initialize(model, la, "O1")
+    ddp = DDP(model, bucket_cap_mb=0.0001)
+    torch.manual_seed(900 + rank)
+    sigs = []
+    for _ in range(3):
+        apex_opt.zero_grad()
+        x = torch.randn(4, 8)
+        t = torch.randn(4, 4)
+        ddp.train()
+        loss = ((ddp(x) - t) ** 2).mean()
+        with amp.scale_loss(loss, apex_opt) as sl:
+            sl.backward()
+        apex_opt.step()
+        sigs.append(sorted(float(p.detach().double().sum())
+                           for p in model.parameters()))
+    amp._state.__init__()
+    results.put((rank, sigs))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_full_stack_world4_lockstep():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_w4, args=(r, 29619, results))
+             for r in range(4)]
+    [p.start() for p in procs]
+    got = dict(results.get(timeout=170) for _ in range(4))
+    [p.join(timeout=60) for p in procs]
+    assert all(p.exitcode == 0 for p in procs)
+    for step in range(3):
+        base = got[0][step]
+        for r in range(1, 4):
+            assert got[r][step] == pytest.approx(base, abs=1e-12), \
+                f"rank {r} diverged at step {step}"
