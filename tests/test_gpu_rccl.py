@@ -147,3 +147,43 @@ def test_native_rccl_world2_full_stack_lockstep():
         for a, b in zip(pn, pt):
             assert a == pytest.approx(b, rel=1e-6, abs=1e-9), \
                 f"native-vs-torch param mismatch at step {step}"
+
+
+@pytest.mark.timeout(300)
+def test_rccl_reducer_stream_pattern_world1():
+    """Drive the exact enqueue pattern the DDP reducer uses on the native
+    path — compute-stream event → comm-stream wait → grouped
+    rccl_all_reduce of bucket flats → main-stream wait — through a real
+    RCCL communicator (world 1: transport is a no-op, the stream
+    semantics and bindings are not)."""
+    from ddp_tricks_amd.ops import load_extension
+    ext = load_extension()
+    uid = ext.rccl_unique_id()
+    comm = ext.rccl_comm_init(1, 0, uid)
+    dev = torch.device("cuda", 0)
+    torch.cuda.set_device(0)
+    comm_stream = torch.cuda.Stream(device=dev)
+    flats = [torch.randn(1 << k, device=dev) for k in (10, 14, 8)]
+    want = [f.clone() for f in flats]
+    # produce on the compute stream (in-place mul), then reduce on the
+    # comm stream exactly like _on_grad_ready
+    for f, w in zip(flats, want):
+        f.mul_(3.0)
+        w.mul_(3.0)
+        ev = torch.cuda.Event()
+        ev.record(torch.cuda.current_stream(dev))
+        comm_stream.wait_event(ev)
+        with torch.cuda.stream(comm_stream):
+            ext.rccl_all_reduce(f, comm, False)
+    torch.cuda.current_stream(dev).wait_stream(comm_stream)
+    # grouped window over all flats (the broadcast path's shape)
+    with torch.cuda.stream(comm_stream):
+        ext.rccl_group_start()
+        for f in flats:
+            ext.rccl_broadcast(f, 0, comm)
+        ext.rccl_group_end()
+    torch.cuda.current_stream(dev).wait_stream(comm_stream)
+    torch.cuda.synchronize()
+    for f, w in zip(flats, want):
+        assert torch.equal(f, w)
+    ext.rccl_comm_destroy(comm)
