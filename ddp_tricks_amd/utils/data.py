@@ -138,10 +138,38 @@ class CIFAR10(Dataset):
     no network in this environment).  Returns (float32 [3,32,32] in [0,1],
     int64 label); BASELINE.json config 4."""
 
+    @staticmethod
+    def _download(root: str) -> None:
+        """Best-effort fetch+extract of the python-batches tarball
+        (torchvision download=True parity; offline → warn and fall back)."""
+        import tarfile
+        import urllib.error
+        import urllib.request
+        url = ("https://www.cs.toronto.edu/~kriz/cifar-10-python.tar.gz")
+        os.makedirs(root, exist_ok=True)
+        tgz = os.path.join(root, "cifar-10-python.tar.gz")
+        try:
+            with urllib.request.urlopen(url, timeout=20) as r, \
+                    open(tgz + ".part", "wb") as f:
+                while True:
+                    chunk = r.read(1 << 20)
+                    if not chunk:
+                        break
+                    f.write(chunk)
+            os.replace(tgz + ".part", tgz)
+            with tarfile.open(tgz, "r:gz") as tf:
+                tf.extractall(root)
+        except (urllib.error.URLError, OSError, ValueError, tarfile.TarError):
+            import warnings
+            warnings.warn("CIFAR-10 download failed (no network?); "
+                          "falling back to local/synthetic")
+
     def __init__(self, root: str = "./datasets/", train: bool = True,
                  download: bool = False, synthetic: Optional[bool] = None,
                  num_samples: Optional[int] = None):
         batch_dir = os.path.join(root, "cifar-10-batches-py")
+        if download and not os.path.isdir(batch_dir) and synthetic is not True:
+            self._download(root)
         self.synthetic = synthetic if synthetic is not None \
             else not os.path.isdir(batch_dir)
         n_default = 50000 if train else 10000
