@@ -73,6 +73,18 @@ def train(args):
                                   shuffle=False, pin_memory=pin,
                                   sampler=train_sampler)
     valid_set = Dataset(root=args.data_path, train=False, download=True)
+    if getattr(train_set, "synthetic", False) \
+            and not getattr(valid_set, "synthetic", True):
+        # Coherence guard: the train images are absent (synthetic stand-in)
+        # but real valid files exist — validating a synthetic-trained model
+        # on real digits reports chance accuracy and would silently distort
+        # EarlyStopping/plateau/best-checkpoint decisions.  Keep the pair
+        # learnable: synthetic valid drawn from the same class templates.
+        print("[data] train split is synthetic but real valid files exist; "
+              "using the synthetic valid split for a coherent train/valid "
+              "pair")
+        valid_set = Dataset(root=args.data_path, train=False, download=False,
+                            synthetic=True)
     if hasattr(valid_set, "images"):
         valid_loader = FastBatchLoader(valid_set, args.batch_size,
                                        pin_memory=pin, device=dev_arg)

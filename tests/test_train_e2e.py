@@ -120,3 +120,45 @@ def test_two_runs_same_seed_identical(tmp_path, monkeypatch):
                               weights_only=False)["model"])
     for k in sds[0]:
         assert torch.equal(sds[0][k], sds[1][k]), k
+
+
+def test_synthetic_train_never_pairs_with_real_valid(tmp_path, capsys):
+    """Coherence guard: when train images are absent (synthetic) but real
+    t10k files exist, the trainer must fall back to a synthetic valid set
+    (validating synthetic training on real digits reports chance acc and
+    distorts EarlyStopping/plateau/checkpoint gates)."""
+    import gzip
+    import shutil
+    import struct
+
+    import numpy as np
+
+    raw = tmp_path / "MNIST" / "raw"
+    raw.mkdir(parents=True)
+    # forge tiny REAL-format t10k idx files (images + labels), no train files
+    n = 64
+    img = np.random.default_rng(0).integers(0, 255, (n, 28, 28),
+                                            dtype=np.uint8)
+    with gzip.open(raw / "t10k-images-idx3-ubyte.gz", "wb") as f:
+        f.write(struct.pack(">IIII", 0x00000803, n, 28, 28))
+        f.write(img.tobytes())
+    lbl = np.random.default_rng(1).integers(0, 10, (n,), dtype=np.uint8)
+    with gzip.open(raw / "t10k-labels-idx1-ubyte.gz", "wb") as f:
+        f.write(struct.pack(">II", 0x00000801, n))
+        f.write(lbl.tobytes())
+
+    import argparse
+
+    from ddp_tricks_amd.utils.train import train
+    args = argparse.Namespace(
+        exp_name="coherence", learning_rate=0.1, batch_size=32, epochs=1,
+        warmup_epochs=1, warmup_type="linear", seed_num=42,
+        data_path=str(tmp_path), model_path=str(tmp_path / "out"),
+        local_rank=0)
+    os.environ["DDPX_SYNTH_SAMPLES"] = "64"
+    try:
+        train(args)
+    finally:
+        del os.environ["DDPX_SYNTH_SAMPLES"]
+    out = capsys.readouterr().out
+    assert "synthetic valid split" in out
