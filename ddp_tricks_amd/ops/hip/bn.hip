@@ -126,35 +126,33 @@ __global__ void k_bn_apply_v8(const bf16* __restrict__ x, bf16* __restrict__ y,
                               unsigned char* __restrict__ mask) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
-    // 4-deep unroll: four 16 B loads in flight before any is consumed
-    // (2-deep measured ~5 of 8 TB/s; elementwise — depth cannot change
-    // results)
-    for (; i + 3 * stride < total_v; i += 4 * stride) {
-        long ii[4] = {i, i + stride, i + 2 * stride, i + 3 * stride};
-        s16x8 v[4], r[4];
+    // 2-deep unroll: both 16 B loads in flight before either is consumed
+    // (single-load iterations leave HBM latency exposed — ~5 of 8 TB/s)
+    for (; i + stride < total_v; i += 2 * stride) {
+        long i2 = i + stride;
+        s16x8 v0 = reinterpret_cast<const s16x8*>(x)[i];
+        s16x8 v1 = reinterpret_cast<const s16x8*>(x)[i2];
+        s16x8 r0 = resid ? reinterpret_cast<const s16x8*>(resid)[i] : s16x8{};
+        s16x8 r1 = resid ? reinterpret_cast<const s16x8*>(resid)[i2] : s16x8{};
+        int cv0 = (i % Cv) * 8, cv1 = (i2 % Cv) * 8;
+        s16x8 o0, o1;
+        unsigned m0 = 0, m1 = 0;
         #pragma unroll
-        for (int u = 0; u < 4; ++u)
-            v[u] = reinterpret_cast<const s16x8*>(x)[ii[u]];
-        #pragma unroll
-        for (int u = 0; u < 4; ++u)
-            r[u] = resid ? reinterpret_cast<const s16x8*>(resid)[ii[u]]
-                         : s16x8{};
-        #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            int cv = (ii[u] % Cv) * 8;
-            s16x8 o;
-            unsigned m = 0;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                float f = fmaf(us2f((unsigned short)v[u][j]), scale[cv + j],
-                               shift[cv + j]);
-                if (resid) f += us2f((unsigned short)r[u][j]);
-                if (relu) { if (f > 0.f) m |= 1u << j; f = fmaxf(f, 0.f); }
-                o[j] = (short)f2us(f);
-            }
-            reinterpret_cast<s16x8*>(y)[ii[u]] = o;
-            if (mask) mask[ii[u]] = (unsigned char)m;
+        for (int j = 0; j < 8; ++j) {
+            float f = fmaf(us2f((unsigned short)v0[j]), scale[cv0 + j],
+                           shift[cv0 + j]);
+            if (resid) f += us2f((unsigned short)r0[j]);
+            if (relu) { if (f > 0.f) m0 |= 1u << j; f = fmaxf(f, 0.f); }
+            o0[j] = (short)f2us(f);
+            float g = fmaf(us2f((unsigned short)v1[j]), scale[cv1 + j],
+                           shift[cv1 + j]);
+            if (resid) g += us2f((unsigned short)r1[j]);
+            if (relu) { if (g > 0.f) m1 |= 1u << j; g = fmaxf(g, 0.f); }
+            o1[j] = (short)f2us(g);
         }
+        reinterpret_cast<s16x8*>(y)[i] = o0;
+        reinterpret_cast<s16x8*>(y)[i2] = o1;
+        if (mask) { mask[i] = (unsigned char)m0; mask[i2] = (unsigned char)m1; }
     }
     for (; i < total_v; i += stride) {
         int cv = (i % Cv) * 8;
@@ -309,35 +307,31 @@ __global__ void k_bn_bwd_dx(const bf16* __restrict__ x,
     long r0 = (long)blockIdx.x * nw + walker;
     long rstride = (long)gridDim.x * nw;
     long r = r0;
-    // 4-deep: all four row-pairs of loads in flight first (elementwise —
-    // bitwise-identical at any depth)
-    for (; r + 3 * rstride < M; r += 4 * rstride) {
-        long rr[4] = {r, r + rstride, r + 2 * rstride, r + 3 * rstride};
-        s16x8 vx[4], vg[4];
-        unsigned mm[4];
+    for (; r + rstride < M; r += 2 * rstride) {
+        long r2 = r + rstride;
+        s16x8 vx0 = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 vg0 = reinterpret_cast<const s16x8*>(dy + r * C)[c8];
+        s16x8 vx1 = reinterpret_cast<const s16x8*>(x + r2 * C)[c8];
+        s16x8 vg1 = reinterpret_cast<const s16x8*>(dy + r2 * C)[c8];
+        unsigned m0 = relu ? mask[r * cpg + c8] : 0xffu;
+        unsigned m1 = relu ? mask[r2 * cpg + c8] : 0xffu;
+        s16x8 o0, og0, o1, og1;
         #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            vx[u] = reinterpret_cast<const s16x8*>(x + rr[u] * C)[c8];
-            vg[u] = reinterpret_cast<const s16x8*>(dy + rr[u] * C)[c8];
+        for (int j = 0; j < 8; ++j) {
+            float g = (m0 >> j) & 1u ? us2f((unsigned short)vg0[j]) : 0.f;
+            float xh = (us2f((unsigned short)vx0[j]) - mean[j]) * invstd[j];
+            o0[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
+            og0[j] = (short)f2us(g);
+            float g1 = (m1 >> j) & 1u ? us2f((unsigned short)vg1[j]) : 0.f;
+            float xh1 = (us2f((unsigned short)vx1[j]) - mean[j]) * invstd[j];
+            o1[j] = (short)f2us(ca[j] * (g1 - cb[j] - xh1 * cc[j]));
+            og1[j] = (short)f2us(g1);
         }
-        #pragma unroll
-        for (int u = 0; u < 4; ++u)
-            mm[u] = relu ? mask[rr[u] * cpg + c8] : 0xffu;
-        #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            s16x8 o, og;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                float g = (mm[u] >> j) & 1u
-                    ? us2f((unsigned short)vg[u][j]) : 0.f;
-                float xh = (us2f((unsigned short)vx[u][j]) - mean[j])
-                           * invstd[j];
-                o[j] = (short)f2us(ca[j] * (g - cb[j] - xh * cc[j]));
-                og[j] = (short)f2us(g);
-            }
-            reinterpret_cast<s16x8*>(dx + rr[u] * C)[c8] = o;
-            if (dresid)
-                reinterpret_cast<s16x8*>(dresid + rr[u] * C)[c8] = og;
+        reinterpret_cast<s16x8*>(dx + r * C)[c8] = o0;
+        reinterpret_cast<s16x8*>(dx + r2 * C)[c8] = o1;
+        if (dresid) {
+            reinterpret_cast<s16x8*>(dresid + r * C)[c8] = og0;
+            reinterpret_cast<s16x8*>(dresid + r2 * C)[c8] = og1;
         }
     }
     for (; r < M; r += rstride) {
