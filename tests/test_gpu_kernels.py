@@ -846,3 +846,22 @@ def test_bn_eval_backward(relu):
     _close(x.grad, xf.grad, rel=3e-2, atol=2e-2, name="bn eval dx")
     _close(g.grad, g2.grad, rel=2e-2, atol=0.1, name="bn eval dgamma")
     _close(b.grad, b2.grad, rel=2e-2, atol=0.1, name="bn eval dbeta")
+
+
+def test_conv_dgrad_addend_kernel():
+    """Direct kernel-level check of the fused epilogue addend:
+    dgrad(dy, addend=g) == dgrad(dy) + g, bitwise (the add is the same
+    fp32-then-round as ATen's bf16 add)."""
+    from ddp_tricks_amd.ops.functional import weight_variant
+    torch.manual_seed(9)
+    N, C, H, W, K, R = 8, 128, 14, 14, 256, 3
+    w = torch.randn(K, C, R, R, device=DEV)
+    dy = torch.randn(N, K, H - 2, W - 2, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL)
+    g = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16) \
+        .contiguous(memory_format=CL)
+    wt2 = weight_variant(w, "wt2")
+    dx_plain = ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R, 1, 0)
+    ref = (dx_plain.float() + g.float()).to(torch.bfloat16)
+    dx_fused = ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R, 1, 0, g)
+    assert torch.equal(dx_fused, ref)
