@@ -185,10 +185,13 @@ def train(args):
             tb.add_scalars("Acc", {"train": train_acc, "valid": valid_acc}, epoch)
 
         phase_info = ""
-        timers = getattr(iterate_loader, "_timers", None)
-        if timers is not None:  # DDPX_PHASE_TIMERS=1
-            phase_info = f", phases[{timers.format()}]"
-            timers.reset()
+        # DDPX_PHASE_TIMERS=1; the engine caches PhaseTimers
+        # per device (events/streams are device-bound)
+        tcache = getattr(iterate_loader, "_timers", None) or {}
+        tm = tcache.get(device)
+        if tm is not None:
+            phase_info = f", phases[{tm.format()}]"
+            tm.reset()
         print(f"epoch: {epoch:03d}/{args.epochs}, "
               f"time: {time.time() - epoch_start_time:.2f}s, "
               f"learning_rate: {curr_lr}, "

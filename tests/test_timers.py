@@ -25,3 +25,23 @@ def test_format():
     with t.phase("h2d"):
         time.sleep(0.002)
     assert "h2d=" in t.format()
+
+
+def test_phase_timers_epoch_line(tmp_path, capsys, monkeypatch):
+    """DDPX_PHASE_TIMERS=1 end-to-end: the epoch line carries the phase
+    breakdown (regression: the per-device timer cache broke the trainer's
+    consumer in round 2)."""
+    import argparse
+
+    from ddp_tricks_amd.utils.train import train
+    monkeypatch.setenv("DDPX_PHASE_TIMERS", "1")
+    monkeypatch.setenv("DDPX_SYNTH_SAMPLES", "512")
+    monkeypatch.setenv("DDPX_NO_TQDM", "1")
+    args = argparse.Namespace(
+        exp_name="timers", learning_rate=0.1, batch_size=128, epochs=1,
+        warmup_epochs=1, warmup_type="linear", seed_num=42,
+        data_path=str(tmp_path / "d"), model_path=str(tmp_path / "m"),
+        local_rank=0)
+    train(args)
+    out = capsys.readouterr().out
+    assert "phases[" in out and "fwd=" in out and "opt=" in out
