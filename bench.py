@@ -53,9 +53,10 @@ def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     # default steps sized so the timed region is ≳1 s on an MI355X (the
-    # driver's GPU-busy sampler needs to see it)
-    p.add_argument("--steps", type=int, default=200)
-    p.add_argument("--warmup", type=int, default=20)
+    # driver's GPU-busy sampler needs to see it; VERDICT r01 weak #9 —
+    # 300 × 3.3 ms ≈ 1.0 s at the round-2 Toy_Net step time)
+    p.add_argument("--steps", type=int, default=300)
+    p.add_argument("--warmup", type=int, default=30)
     p.add_argument("--batch-size", type=int, default=None,
                    help="per-GPU batch size (default: model-specific)")
     p.add_argument("--model", default="toy_net",
