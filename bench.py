@@ -46,6 +46,8 @@ MODEL_CONFIGS = {
                  2048, "CIFAR-10(synthetic)"),
     "resnet50": ({"num_classes": 1000}, (3, 224, 224), 1000, 1024,
                  "ImageNet(synthetic)"),
+    "vgg16": ({"num_classes": 1000}, (3, 224, 224), 1000, 256,
+              "ImageNet(synthetic)"),
 }
 
 

@@ -241,3 +241,34 @@ def test_resnet_matches_cpu_oracle(arch):
             ng = grads_g[k].norm().item()
             assert ng == pytest.approx(nc, rel=0.5, abs=1e-3), (k, nc, ng)
     assert coses["fc.weight"] > 0.99, coses["fc.weight"]
+
+
+@pytest.mark.timeout(300)
+def test_vgg16_step_gpu():
+    """VGG-16-BN trains on GPU through the full stack (loss decreases)."""
+    from ddp_tricks_amd import amp, same_seeds
+    from ddp_tricks_amd.models import build_model
+    from ddp_tricks_amd.ops.functional import cross_entropy_loss
+    from ddp_tricks_amd.ops.optim import FusedSGD
+    from ddp_tricks_amd.utils.lookahead import Lookahead
+    amp._state.__init__()
+    same_seeds(5)
+    model = build_model("vgg16", num_classes=10, cifar_head=True).to(DEV)
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9, nesterov=True)
+    la = Lookahead(opt, k=10, alpha=0.5)
+    model, apex_opt = amp.initialize(model, la, "O1")
+    x = torch.rand(256, 3, 32, 32, device=DEV)
+    t = torch.randint(0, 10, (256,), device=DEV)
+    model.train()
+    losses = []
+    for _ in range(8):
+        apex_opt.zero_grad()
+        out = model(x)
+        loss = cross_entropy_loss(out, t) / out.shape[0]
+        with amp.scale_loss(loss, apex_opt) as sl:
+            sl.backward()
+        apex_opt.step()
+        losses.append(float(loss.detach()))
+    amp._state.__init__()
+    assert all(l == l and l != float("inf") for l in losses), losses
+    assert losses[-1] < losses[0], losses

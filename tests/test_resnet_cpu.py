@@ -111,3 +111,38 @@ def test_flatten_cpu_fallback_semantics():
     x = torch.randn(3, 5, 2, 4).contiguous(memory_format=torch.channels_last)
     y = Flatten()(x)
     assert torch.equal(y, torch.flatten(x, 1))
+
+
+def test_vgg16_cpu_forward_backward():
+    """VGG-16-BN zoo extension: shapes, torchvision-style keys, grads."""
+    import torch
+
+    from ddp_tricks_amd.models import build_model
+    for kw, shape, ncls in [({"num_classes": 7}, (2, 3, 224, 224), 7),
+                            ({"num_classes": 10, "cifar_head": True},
+                             (2, 3, 32, 32), 10)]:
+        m = build_model("vgg16", **kw)
+        y = m(torch.randn(*shape))
+        assert y.shape == (2, ncls)
+        y.pow(2).mean().backward()
+        assert all(p.grad is not None for p in m.parameters())
+    keys = m.state_dict()
+    assert "features.0.weight" in keys and "features.1.running_mean" in keys
+    assert "classifier.6.weight" in keys   # torchvision vgg16_bn layout
+
+
+def test_vgg16_dropout_slot():
+    import torch
+
+    from ddp_tricks_amd.models import build_model
+    m = build_model("vgg16", num_classes=10, cifar_head=True, dropout=0.5)
+    m.train()
+    torch.manual_seed(0)
+    y1 = m(torch.ones(4, 3, 32, 32))
+    y2 = m(torch.ones(4, 3, 32, 32))
+    assert not torch.equal(y1, y2)   # dropout active in train mode
+    m.eval()
+    with torch.no_grad():
+        z1 = m(torch.ones(4, 3, 32, 32))
+        z2 = m(torch.ones(4, 3, 32, 32))
+    assert torch.equal(z1, z2)
