@@ -251,8 +251,11 @@ __global__ void k_transpose_bf16(const bf16* __restrict__ in,
 // colsum[c] = sum_m in[m][c] (bf16 in, fp32 out), deterministic slab
 // scheme, bf16x8-vectorized; slab layout [C][S] so combine waves read each
 // channel's splits coalesced.
+// C = column-window width (<= 2048); ldx = full row stride, so wide
+// matrices (e.g. the VGG classifier's 4096-wide bias grad) sweep in
+// column windows
 __global__ void k_colsum_partial(const bf16* __restrict__ x, long M, int C,
-                                 int S, float* __restrict__ slab) {
+                                 long ldx, int S, float* __restrict__ slab) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* red = reinterpret_cast<float*>(smem);   // [nw][C]
     const int cpg = C >> 3;
@@ -262,7 +265,7 @@ __global__ void k_colsum_partial(const bf16* __restrict__ x, long M, int C,
     const int s = blockIdx.x;
     float acc[8] = {};
     for (long r = (long)s * nw + walker; r < M; r += (long)S * nw) {
-        s16x8 v = reinterpret_cast<const s16x8*>(x + r * C)[c8];
+        s16x8 v = reinterpret_cast<const s16x8*>(x + r * ldx)[c8];
         #pragma unroll
         for (int j = 0; j < 8; ++j) acc[j] += us2f((unsigned short)v[j]);
     }
@@ -396,21 +399,28 @@ at::Tensor col_sum(at::Tensor x) {
     int C = x.size(1);
     auto stream = at::hip::getCurrentHIPStream();
     auto out = at::empty({C}, x.options().dtype(at::kFloat));
-    if (C % 8 == 0 && C / 8 <= 256) {
-        int cpg = C / 8;
-        int block = (256 / cpg) * cpg;
-        int nw = block / cpg;
-        int S = (int)std::max<long>(1, std::min<long>(1024, M / (nw * 4)));
-        auto slab = at::empty({C, S}, x.options().dtype(at::kFloat));
-        hipLaunchKernelGGL(k_colsum_partial, dim3(S), dim3(block),
-                           nw * C * 4, stream.stream(),
-                           reinterpret_cast<const bf16*>(x.data_ptr()), M, C,
-                           S, slab.data_ptr<float>());
-        HIP_CHECK_LAST();
-        hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(C, 4)), dim3(256),
-                           0, stream.stream(), slab.data_ptr<float>(), S, C,
-                           out.data_ptr<float>());
-        HIP_CHECK_LAST();
+    if (C % 8 == 0) {
+        // column windows of <= 2048 (one window for every model head up
+        // to that width; the VGG 4096-wide classifier takes two)
+        for (int c0 = 0; c0 < C; c0 += 2048) {
+            int Cw = std::min(2048, C - c0);
+            int cpg = Cw / 8;
+            int block = (256 / cpg) * cpg;
+            if (block == 0) block = cpg;     // cpg in (128, 256]: 1 walker
+            int nw = std::max(1, block / cpg);
+            int S = (int)std::max<long>(1, std::min<long>(1024, M / (nw * 4)));
+            auto slab = at::empty({Cw, S}, x.options().dtype(at::kFloat));
+            hipLaunchKernelGGL(k_colsum_partial, dim3(S), dim3(block),
+                               nw * Cw * 4, stream.stream(),
+                               reinterpret_cast<const bf16*>(x.data_ptr()) + c0,
+                               M, Cw, (long)C, S, slab.data_ptr<float>());
+            HIP_CHECK_LAST();
+            hipLaunchKernelGGL(k_colsum_combine, dim3(ceil_div_i(Cw, 4)),
+                               dim3(256), 0, stream.stream(),
+                               slab.data_ptr<float>(), S, Cw,
+                               out.data_ptr<float>() + c0);
+            HIP_CHECK_LAST();
+        }
         return out;
     }
     TORCH_CHECK(C <= 256, "col_sum scalar path needs C<=256");

@@ -868,3 +868,12 @@ def test_conv_dgrad_addend_kernel():
     ref = (dx_plain.float() + g.float()).to(torch.bfloat16)
     dx_fused = ext.conv2d_dgrad(dy, wt2, N, C, H, W, R, R, 1, 0, g)
     assert torch.equal(dx_fused, ref)
+
+
+def test_colsum_wide_c():
+    """Column-windowed col_sum for wide heads (VGG classifier: C=4096)."""
+    torch.manual_seed(3)
+    for M, C in [(256, 4096), (1024, 2048), (64, 1032), (512, 25088)]:
+        x = torch.randn(M, C, device=DEV).to(torch.bfloat16).contiguous()
+        cs = ext.col_sum(x)
+        _close(cs, x.float().sum(0), rel=1e-3, atol=1.0, name=f"colsum {C}")
