@@ -55,6 +55,9 @@ def train(args):
     if model_name.startswith("resnet") and dataset_name == "cifar10":
         model_kwargs.setdefault("num_classes", 10)
         model_kwargs.setdefault("cifar_stem", True)
+    if model_name.startswith("vgg") and dataset_name == "cifar10":
+        model_kwargs.setdefault("num_classes", 10)
+        model_kwargs.setdefault("cifar_head", True)
     Dataset = DATASETS[dataset_name]
 
     # Data pipeline (reference utils/train.py:24-30: sharded train loader,

@@ -40,7 +40,7 @@ def parse_args():
                         help="local rank for DistributedDataParallel")
     # extensions beyond the reference's 10-flag contract
     parser.add_argument("--model", default="toy_net", type=str,
-                        help="model registry name (toy_net, resnet18, "
+                        help="model registry name (toy_net, vgg16, resnet18, "
                              "resnet34, resnet50)")
     parser.add_argument("--dataset", default="mnist", type=str,
                         help="dataset registry name (mnist, cifar10, "
