@@ -311,5 +311,11 @@ def test_vgg16_matches_cpu_oracle():
     med = sorted(weight_cos)[len(weight_cos) // 2]
     assert med > 0.6, (med, sorted(weight_cos)[:5])
     for k in gc:
+        if k.startswith("features") and k.endswith("bias"):
+            # conv bias under BN is mathematically dead (the mean
+            # subtraction cancels any bias shift) — its grad is pure
+            # rounding noise at each precision's floor (fp32 ~1e-6,
+            # bf16 ~6e-3); kept for torchvision key compat only
+            continue
         assert gg[k].norm().item() == pytest.approx(
             gc[k].norm().item(), rel=0.5, abs=1e-3), k
