@@ -1675,7 +1675,6 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
     // ResNet site); larger strides launch per class.
     ClsPack pack{};
     int ncls = 0;
-    bool overflow = false;
     int maxTiles = 0;
     for (int a = 0; a < st; ++a) {
         for (int b = 0; b < st; ++b) {
@@ -1712,7 +1711,7 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
                 maxTiles = std::max(maxTiles, (int)ceil_div_i(M2, CBM));
                 ++ncls;
             } else {
-                overflow = true;
+                // >4 contributing classes (stride > 2): per-class launch
                 if (cs.C >= 128) {
                     dim3 grid(ceil_div_i(M2, CBM), ceil_div_i(cs.C, 128));
                     hipLaunchKernelGGL((k_conv_gemm<2, 128, 2, 2, true>),
@@ -1730,7 +1729,6 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt2, long N, long C,
             }
         }
     }
-    (void)overflow;
     if (ncls > 0) {
         if (cs.C >= 128) {
             dim3 grid(maxTiles, ceil_div_i(cs.C, 128), ncls);
